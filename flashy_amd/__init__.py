@@ -1,0 +1,27 @@
+# Copyright (c) Flashy-AMD authors.
+"""flashy_amd — an MI355X-native minimal solver framework for deep learning.
+
+A from-scratch framework with the capabilities of facebookresearch/flashy
+(metric logging to multiple backends, automatic stateful checkpointing, and
+DDP-alternative distributed utilities), re-designed for AMD Instinct MI355X:
+one process per GPU over RCCL/xGMI, bucketed overlapped gradient sync on a
+side HIP stream, HIP-graph step capture, hand-written CDNA4 (gfx950) HIP
+kernels for the hot ops, and pinned-host streamed checkpoints.
+
+Public surface (parity: /root/reference/flashy/__init__.py:11-15):
+``distrib``, ``adversarial`` modules; ``Formatter``, ``ResultLogger``,
+``LogProgressBar``, ``bold``, ``setup_logging``, ``BaseSolver``, ``averager``.
+Extras beyond the reference: ``xp`` (experiment runtime), ``graph``
+(HIP-graph capture), ``models``, ``ops`` (CDNA4 kernels).
+"""
+
+__version__ = "0.1.0a1"
+
+from . import adversarial  # noqa: F401
+from . import distrib  # noqa: F401
+from . import graph  # noqa: F401
+from . import xp  # noqa: F401
+from .formatter import Formatter  # noqa: F401
+from .logging import LogProgressBar, ResultLogger, bold, setup_logging  # noqa: F401
+from .solver import BaseSolver  # noqa: F401
+from .utils import averager  # noqa: F401

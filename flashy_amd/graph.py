@@ -1,0 +1,81 @@
+# Copyright (c) Flashy-AMD authors.
+"""HIP-graph step capture: replace hundreds of per-step kernel launches with
+one graph replay.
+
+The MI355X-native answer to launch-bound small-batch training (a CIFAR-shaped
+ResNet step at batch 64 is dominated by launch overhead on a 256-CU chip):
+run the whole training step — forward, loss, backward, optimizer — once into
+a ``hipGraph`` (``torch.cuda.CUDAGraph`` on ROCm) over static buffers, then
+replay it each step.  Fresh data is copied into the static input buffers
+before each replay.
+
+Usage::
+
+    static_x = torch.empty(bs, 3, 32, 32, device="cuda")
+    static_y = torch.empty(bs, dtype=torch.long, device="cuda")
+
+    def step():
+        optimizer.zero_grad(set_to_none=False)
+        loss = F.cross_entropy(model(static_x), static_y)
+        loss.backward()
+        optimizer.step()
+        return loss
+
+    graphed = CapturedStep(step).capture()
+    for x, y in data:
+        static_x.copy_(x, non_blocking=True)
+        static_y.copy_(y, non_blocking=True)
+        loss = graphed()          # one hipGraphLaunch
+
+Notes:
+ * warmup iterations run on a side stream so allocations (grads, optimizer
+   state) exist before capture;
+ * ``zero_grad(set_to_none=False)`` keeps gradient storage stable across
+   replays;
+ * on CPU (tests/CI) the same object degrades to calling ``fn`` eagerly.
+"""
+from __future__ import annotations
+
+import typing as tp
+
+import torch
+
+
+class CapturedStep:
+    """Capture a closure over static tensors into a replayable HIP graph."""
+
+    def __init__(self, fn: tp.Callable[[], tp.Any], warmup: int = 3,
+                 pool: tp.Optional[tp.Any] = None):
+        self.fn = fn
+        self.warmup = warmup
+        self.pool = pool
+        self.graph: tp.Optional[torch.cuda.CUDAGraph] = None
+        self.output: tp.Any = None
+
+    @property
+    def enabled(self) -> bool:
+        return torch.cuda.is_available()
+
+    def capture(self) -> "CapturedStep":
+        if not self.enabled:
+            return self
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(self.warmup):
+                self.fn()
+        torch.cuda.current_stream().wait_stream(side)
+        torch.cuda.synchronize()
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph, pool=self.pool):
+            self.output = self.fn()
+        return self
+
+    def replay(self) -> tp.Any:
+        if self.graph is None:
+            # eager fallback (CPU tests, or capture explicitly disabled)
+            return self.fn()
+        self.graph.replay()
+        return self.output
+
+    __call__ = replay

@@ -1,0 +1,49 @@
+# Copyright (c) Flashy-AMD authors.
+"""End-to-end integration via the real CLI (subprocess), mirroring the
+reference's resume oracle (reference tests/test_integ.py): run the dummy
+workload to epoch 2, re-run without --clear and assert the history grows to
+4 with a bit-identical 2-epoch prefix, then exercise 2-worker local DDP."""
+import json
+import os
+import subprocess
+import sys
+from pathlib import Path
+
+REPO = Path(__file__).resolve().parent.parent
+
+
+def _run(tmp: Path, *args: str) -> None:
+    env = dict(os.environ)
+    env["_FLASHY_AMD_DIR"] = str(tmp)
+    env["PYTHONPATH"] = str(REPO) + os.pathsep + env.get("PYTHONPATH", "")
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    subprocess.run([sys.executable, "-m", "flashy_amd.run", "tests.dummy", *args],
+                   check=True, cwd=REPO, env=env, timeout=300)
+
+
+def _history(tmp: Path) -> list:
+    xps = list((tmp / "xps").iterdir())
+    assert len(xps) == 1, xps
+    with open(xps[0] / "history.json") as fh:
+        return json.load(fh)
+
+
+def test_resume_prefix_equality(tmp_path):
+    # stop_at is excluded from the signature via run.exclude so partial and
+    # full runs share one XP
+    excl = "run.exclude=[stop_at]"
+    _run(tmp_path, "--clear", excl, "stop_at=2")
+    hist2 = _history(tmp_path)
+    assert len(hist2) == 2
+
+    _run(tmp_path, excl)  # stop_at back to null -> runs to epochs=4
+    hist4 = _history(tmp_path)
+    assert len(hist4) == 4
+    assert hist4[:2] == hist2  # bit-identical persisted prefix
+
+
+def test_local_ddp_two_workers(tmp_path):
+    _run(tmp_path, "--clear", "-d", "--workers", "2", "epochs=2")
+    hist = _history(tmp_path)
+    assert len(hist) == 2
