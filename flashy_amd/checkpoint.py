@@ -30,13 +30,21 @@ _copy_stream: tp.Optional["torch.cuda.Stream"] = None
 _pinned_pool: tp.Dict[torch.dtype, torch.Tensor] = {}
 
 
+def _compact_cpu(t: torch.Tensor) -> torch.Tensor:
+    """Clone CPU tensors that view a larger storage (e.g. flat-optimizer
+    parameter views): ``torch.save`` serializes the WHOLE underlying storage
+    of a view, which would bloat every param in a flat group to the full
+    flat-buffer size."""
+    t = t.detach()
+    if t.untyped_storage().nbytes() != t.numel() * t.element_size() \
+            or not t.is_contiguous():
+        return t.clone()
+    return t
+
+
 def _stage_to_host(state: tp.Any) -> tp.Any:
     """Deep-copy ``state`` with every CUDA tensor replaced by an async pinned
     host copy; all copies are in flight before the final sync."""
-    global _copy_stream
-    if _copy_stream is None:
-        _copy_stream = torch.cuda.Stream()
-
     pending: tp.List[tp.Tuple[torch.Tensor, torch.Tensor]] = []
 
     def _walk(obj: tp.Any) -> tp.Any:
@@ -45,7 +53,7 @@ def _stage_to_host(state: tp.Any) -> tp.Any:
                 host = torch.empty(obj.shape, dtype=obj.dtype, pin_memory=True)
                 pending.append((host, obj))
                 return host
-            return obj.detach().clone() if obj.requires_grad else obj
+            return _compact_cpu(obj)
         if isinstance(obj, dict):
             return {k: _walk(v) for k, v in obj.items()}
         if isinstance(obj, (list, tuple)):
@@ -54,6 +62,9 @@ def _stage_to_host(state: tp.Any) -> tp.Any:
 
     out = _walk(state)
     if pending:
+        global _copy_stream
+        if _copy_stream is None:
+            _copy_stream = torch.cuda.Stream()
         _copy_stream.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(_copy_stream):
             for host, dev in pending:
@@ -68,8 +79,7 @@ def save_state(state: tp.Any, path: tp.Union[str, Path]) -> None:
     CUDA tensors are staged through pinned host buffers on a copy stream
     first; the pickle then writes from host memory only.
     """
-    if torch.cuda.is_available():
-        state = _stage_to_host(state)
+    state = _stage_to_host(state)
     with write_and_rename(path) as fh:
         torch.save(state, fh)
 

@@ -308,6 +308,22 @@ def sync_model(model: torch.nn.Module, sync_buffers: tp.Union[bool, str] = True,
             broadcast_tensors(buffers, src=0, bucket_bytes=bucket_bytes)
 
 
+def sync_flat_gradients(optimizer) -> None:
+    """DP gradient sync for a :class:`flashy_amd.optim.FlatOptimizer`: the
+    gradients already live in one contiguous flat buffer per (device, dtype)
+    group, so the sync is a single RCCL all-reduce per group — the maximal
+    bucket for the per-link-bound xGMI ring — followed by the exact
+    sum/world-size division."""
+    if not is_distributed():
+        return
+    ws = world_size()
+    handles = [dist.all_reduce(g, op=dist.ReduceOp.SUM, async_op=True)
+               for g in optimizer.grad_buffers]
+    for g, h in zip(optimizer.grad_buffers, handles):
+        h.wait()
+        g.div_(ws)
+
+
 # ---------------------------------------------------------------------------
 # Eager (overlapped) gradient sync — comm on a side HIP stream during backward
 # ---------------------------------------------------------------------------
