@@ -43,7 +43,10 @@ def build_step(model, optim, static_x, static_y, autocast: bool, distributed: bo
         optim.zero_grad(set_to_none=False)
         with torch.autocast("cuda", torch.bfloat16, enabled=autocast):
             logits = model(static_x)
-        loss = cross_entropy(logits, static_y)  # fused fwd+grad, bf16-aware
+        if fused:
+            loss = cross_entropy(logits, static_y)  # fused fwd+grad, bf16-aware
+        else:
+            loss = torch.nn.functional.cross_entropy(logits, static_y)
         loss.backward()
         if distributed:
             if fused:
@@ -87,6 +90,10 @@ def main():
     parser.add_argument("--img", type=int, default=32)
     parser.add_argument("--classes", type=int, default=10)
     parser.add_argument("--no-graph", action="store_true")
+    parser.add_argument("--ref", action="store_true",
+                        help="reference mode: stock torch ops (torch.optim.SGD"
+                             " + F.cross_entropy), eager — what the reference"
+                             " framework executes on torch-ROCm")
     parser.add_argument("--no-ckpt", action="store_true",
                         help="skip the checkpoint save/restore measurement")
     parser.add_argument("--channels-last", action="store_true")
@@ -106,7 +113,7 @@ def main():
     if args.channels_last:
         model = model.to(memory_format=torch.channels_last)
     distrib.broadcast_model(model)
-    if use_cuda:
+    if use_cuda and not args.ref:
         optim = FusedSGD(model.parameters(), lr=0.1, momentum=0.9,
                          weight_decay=5e-4)
     else:
@@ -128,7 +135,7 @@ def main():
     autocast = use_cuda
     step = build_step(model, optim, static_x, static_y, autocast, ws > 1)
 
-    use_graph = use_cuda and ws == 1 and not args.no_graph
+    use_graph = use_cuda and ws == 1 and not args.no_graph and not args.ref
     runner = CapturedStep(step, warmup=3).capture() if use_graph else step
 
     def one_step(i: int):
@@ -182,6 +189,7 @@ def main():
                 "img_size": args.img,
                 "num_classes": args.classes,
                 "parallelism": f"dp{ws}",
+                "mode": "reference-torch-ops" if args.ref else "native",
                 "graph": use_graph,
                 "channels_last": args.channels_last,
                 "checkpoint_save_s": ckpt_save_s,
