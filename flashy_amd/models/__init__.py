@@ -8,4 +8,6 @@ in-house: ResNet-18/50 (CIFAR and ImageNet stems) and the DCGAN-style 64x64
 generator/discriminator pair used by the adversarial workload.
 """
 from .resnet import ResNet, resnet18, resnet50  # noqa: F401
+from .resnet_native import (NativeResNet, native_resnet18,  # noqa: F401
+                            native_resnet50)
 from .dcgan import DCGANGenerator, DCGANDiscriminator  # noqa: F401

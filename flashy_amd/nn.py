@@ -1,0 +1,161 @@
+# Copyright (c) Flashy-AMD authors.
+"""Native NHWC modules backed by the gfx950 kernels.
+
+Layout convention: activations are logical-NHWC tensors ``[N, H, W, C]``
+(plain contiguous — the same bytes as torch channels_last), conv weights are
+``[K, R, S, C]``.  These modules are GPU-only (bf16 activations, fp32 master
+params) and pair with :class:`flashy_amd.optim.FlatOptimizer`:
+
+* weight gradients are fp32 and written (accumulated) directly into
+  ``param.grad`` when it exists — with a flat optimizer that is the flat
+  gradient buffer, so conv/BN backward needs no extra accumulate kernels;
+* when the optimizer keeps a bf16 mirror of the flat params
+  (``bf16_mirror=True``), forward reads the mirror view stashed on the
+  parameter (``param._bf16_mirror``) — zero weight-cast kernels per step.
+
+Replaces the ResNet conv+bn+relu torch chains of the reference workload
+ with MFMA implicit-GEMM conv and fused NHWC BatchNorm
+(+residual +ReLU) kernels.
+"""
+from __future__ import annotations
+
+import math
+import typing as tp
+
+import torch
+from torch import nn
+
+from . import ops
+
+
+def _weight_bf16(w: torch.Tensor) -> torch.Tensor:
+    mirror = getattr(w, "_bf16_mirror", None)
+    if mirror is not None:
+        return mirror
+    return w.detach().to(torch.bfloat16)
+
+
+def _grad_target(p: torch.Tensor) -> tp.Tuple[torch.Tensor, bool]:
+    """(fp32 accumulation buffer, direct) — direct=True writes into
+    ``p.grad`` in place (flat-optimizer fast path), else a temp returned to
+    autograd for accumulation."""
+    if p.grad is not None and p.grad.is_contiguous():
+        return p.grad, True
+    return torch.zeros_like(p, memory_format=torch.contiguous_format), False
+
+
+class _ConvFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, w: torch.Tensor, stride: int, pad: int,
+                input_grad: bool):
+        w16 = _weight_bf16(w)
+        d = ops.ConvDims.infer(x, w16, stride, pad)
+        y = x.new_empty((d.N, d.Ho, d.Wo, d.K))
+        ops.conv_fwd(x, w16, y, d)
+        ctx.save_for_backward(x, w16)
+        ctx.dims = d
+        ctx.input_grad = input_grad
+        ctx.w_ref = w
+        return y
+
+    @staticmethod
+    def backward(ctx, dy: torch.Tensor):
+        x, w16 = ctx.saved_tensors
+        d: ops.ConvDims = ctx.dims
+        dy = dy.contiguous()
+        dw_buf, direct = _grad_target(ctx.w_ref)
+        ops.conv_wgrad(x, dy, dw_buf, d)
+        dx = None
+        if ctx.input_grad:
+            wt = w16.new_empty((d.R, d.S, d.C, d.K))
+            ops.weight_transpose(w16, wt)
+            dx = x.new_empty(x.shape)
+            ops.conv_dgrad(dy, wt, dx, d)
+        return dx, None if direct else dw_buf, None, None, None
+
+
+class Conv2d(nn.Module):
+    """NHWC bf16 conv (no bias, as in ResNet).  Weight: [K, R, S, C] fp32."""
+
+    def __init__(self, in_channels: int, out_channels: int, kernel_size: int,
+                 stride: int = 1, padding: int = 0, input_grad: bool = True):
+        super().__init__()
+        self.stride = stride
+        self.padding = padding
+        self.input_grad = input_grad
+        k = kernel_size
+        self.weight = nn.Parameter(
+            torch.empty(out_channels, k, k, in_channels))
+        fan_out = out_channels * k * k
+        nn.init.normal_(self.weight, std=math.sqrt(2.0 / fan_out))
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return _ConvFn.apply(x, self.weight, self.stride, self.padding,
+                             self.input_grad and x.requires_grad)
+
+
+class _BnFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
+                res: tp.Optional[torch.Tensor], relu: bool, module):
+        N, H, W, C = x.shape
+        M = N * H * W
+        y = torch.empty_like(x)
+        if module.training:
+            sums = torch.zeros(2 * C, dtype=torch.float32, device=x.device)
+            work = torch.empty(4 * C, dtype=torch.float32, device=x.device)
+            ops.bn_stats(x, sums, M, C)
+            ops.bn_finalize(sums, gamma, beta, module.running_mean,
+                            module.running_var, work, M, C, module.eps,
+                            module.momentum, update_running=True)
+        else:
+            invstd = torch.rsqrt(module.running_var + module.eps)
+            scale = gamma.detach() * invstd
+            shift = beta.detach() - module.running_mean * scale
+            work = torch.cat([module.running_mean, invstd, scale, shift])
+        ops.bn_apply(x, res, y, work, M, C, relu)
+        ctx.save_for_backward(x, y, work)
+        ctx.relu = relu
+        ctx.has_res = res is not None
+        ctx.refs = (gamma, beta)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy: torch.Tensor):
+        x, y, work = ctx.saved_tensors
+        gamma, beta = ctx.refs
+        N, H, W, C = x.shape
+        M = N * H * W
+        dy = dy.contiguous()
+        bsums = torch.zeros(2 * C, dtype=torch.float32, device=x.device)
+        dz = torch.empty_like(dy)
+        ops.bn_bwd_reduce(dy, y, x, work, dz, bsums, M, C, ctx.relu)
+        dgamma, g_direct = _grad_target(gamma)
+        dbeta, b_direct = _grad_target(beta)
+        ops.bn_bwd_grads(bsums, dgamma, dbeta, C)
+        dx = torch.empty_like(x)
+        ops.bn_bwd_apply(dz, x, work, bsums, dx, M, C)
+        return (dx,
+                None if g_direct else dgamma,
+                None if b_direct else dbeta,
+                dz if ctx.has_res else None,
+                None, None)
+
+
+class BatchNorm2d(nn.Module):
+    """NHWC training BatchNorm with optional fused residual-add + ReLU."""
+
+    def __init__(self, num_features: int, eps: float = 1e-5, momentum: float = 0.1):
+        super().__init__()
+        self.num_features = num_features
+        self.eps = eps
+        self.momentum = momentum
+        self.weight = nn.Parameter(torch.ones(num_features))
+        self.bias = nn.Parameter(torch.zeros(num_features))
+        self.register_buffer("running_mean", torch.zeros(num_features))
+        self.register_buffer("running_var", torch.ones(num_features))
+
+    def forward(self, x: torch.Tensor,
+                res: tp.Optional[torch.Tensor] = None,
+                relu: bool = False) -> torch.Tensor:
+        return _BnFn.apply(x, self.weight, self.bias, res, relu, self)

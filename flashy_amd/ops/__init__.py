@@ -105,6 +105,109 @@ def fused_adam(p: torch.Tensor, g: torch.Tensor, m: torch.Tensor,
 
 
 # ---------------------------------------------------------------------------
+# convolution kernels (NHWC bf16 implicit GEMM; see flashy_amd/nn.py)
+# All activation tensors are logical-NHWC [N, H, W, C]; weights [K, R, S, C].
+# ---------------------------------------------------------------------------
+
+class ConvDims(tp.NamedTuple):
+    N: int; H: int; W: int; C: int
+    K: int; R: int; S: int
+    Ho: int; Wo: int
+    stride: int; pad: int
+
+    @staticmethod
+    def infer(x: torch.Tensor, w: torch.Tensor, stride: int, pad: int) -> "ConvDims":
+        N, H, W, C = x.shape
+        K, R, S, Cw = w.shape
+        assert C == Cw, (x.shape, w.shape)
+        Ho = (H + 2 * pad - R) // stride + 1
+        Wo = (W + 2 * pad - S) // stride + 1
+        return ConvDims(N, H, W, C, K, R, S, Ho, Wo, stride, pad)
+
+
+def conv_fwd(x: torch.Tensor, w: torch.Tensor, y: torch.Tensor, d: ConvDims,
+             relu: bool = False) -> None:
+    ext = require()
+    if d.C % 8 == 0:
+        assert d.K % 64 == 0 and (d.R * d.S * d.C) % 32 == 0, d
+        ext.conv_fwd(x.data_ptr(), w.data_ptr(), y.data_ptr(), *d, relu, _stream())
+    else:
+        assert not relu
+        ext.conv_stem_fwd(x.data_ptr(), w.data_ptr(), y.data_ptr(), *d, _stream())
+
+
+def conv_dgrad(dout: torch.Tensor, w_rsck: torch.Tensor, dx: torch.Tensor,
+               d: ConvDims) -> None:
+    ext = require()
+    assert d.C % 64 == 0 and d.K % 32 == 0, d
+    ext.conv_dgrad(dout.data_ptr(), w_rsck.data_ptr(), dx.data_ptr(), *d, _stream())
+
+
+def weight_transpose(w: torch.Tensor, wt: torch.Tensor) -> None:
+    ext = require()
+    K = w.shape[0]
+    rsc = w.numel() // K
+    ext.weight_transpose(w.data_ptr(), wt.data_ptr(), K, rsc, _stream())
+
+
+def conv_wgrad(x: torch.Tensor, dout: torch.Tensor, dw: torch.Tensor,
+               d: ConvDims, n_splits: tp.Optional[int] = None) -> None:
+    """Accumulates (+=) fp32 weight grads; dw must be zeroed or hold the
+    running gradient (matches autograd accumulate semantics)."""
+    ext = require()
+    rsc = d.R * d.S * d.C
+    if d.C % 8 == 0:
+        assert d.K % 64 == 0 and rsc % 64 == 0, d
+        if n_splits is None:
+            tiles = (d.K // 64) * (rsc // 64)
+            n_splits = max(1, min(512 // tiles if tiles else 1, 64))
+            M = d.N * d.Ho * d.Wo
+            n_splits = max(1, min(n_splits, M // 32 or 1))
+        ext.conv_wgrad(x.data_ptr(), dout.data_ptr(), dw.data_ptr(), *d,
+                       n_splits, _stream())
+    else:
+        assert rsc <= 32, d
+        ext.conv_stem_wgrad(x.data_ptr(), dout.data_ptr(), dw.data_ptr(), *d,
+                            _stream())
+
+
+# ---------------------------------------------------------------------------
+# batchnorm kernels (NHWC training BN; see flashy_amd/nn.py)
+# ---------------------------------------------------------------------------
+
+def bn_stats(x: torch.Tensor, sums: torch.Tensor, M: int, C: int) -> None:
+    require().bn_stats(x.data_ptr(), sums.data_ptr(), M, C, _stream())
+
+
+def bn_finalize(sums, gamma, beta, rmean, rvar, work, M: int, C: int,
+                eps: float, momentum: float, update_running: bool) -> None:
+    require().bn_finalize(sums.data_ptr(), gamma.data_ptr(), beta.data_ptr(),
+                          rmean.data_ptr(), rvar.data_ptr(), work.data_ptr(),
+                          M, C, eps, momentum, update_running, _stream())
+
+
+def bn_apply(x, res, y, work, M: int, C: int, relu: bool) -> None:
+    require().bn_apply(x.data_ptr(), res.data_ptr() if res is not None else 0,
+                       y.data_ptr(), work.data_ptr(), M, C, relu, _stream())
+
+
+def bn_bwd_reduce(dy, y, x, work, dz_out, bsums, M: int, C: int, relu: bool) -> None:
+    require().bn_bwd_reduce(dy.data_ptr(), y.data_ptr(), x.data_ptr(),
+                            work.data_ptr(), dz_out.data_ptr(),
+                            bsums.data_ptr(), M, C, relu, _stream())
+
+
+def bn_bwd_grads(bsums, dgamma, dbeta, C: int) -> None:
+    require().bn_bwd_grads(bsums.data_ptr(), dgamma.data_ptr(),
+                           dbeta.data_ptr(), C, _stream())
+
+
+def bn_bwd_apply(dz, x, work, bsums, dx, M: int, C: int) -> None:
+    require().bn_bwd_apply(dz.data_ptr(), x.data_ptr(), work.data_ptr(),
+                           bsums.data_ptr(), dx.data_ptr(), M, C, _stream())
+
+
+# ---------------------------------------------------------------------------
 # loss kernels (fused forward+input-grad; see flashy_amd/functional.py)
 # ---------------------------------------------------------------------------
 

@@ -9,6 +9,9 @@
 
 #define WAVE_SIZE 64
 
+using short8 = __attribute__((ext_vector_type(8))) short;
+using floatx4 = __attribute__((ext_vector_type(4))) float;
+
 // Grid sizing for memory-bound elementwise kernels (guideline 11: cap the
 // grid near 256 CU x 8 blocks and grid-stride the rest).
 static inline int ew_grid(int64_t n_items, int block, int per_thread) {

@@ -1,0 +1,147 @@
+// Copyright (c) Flashy-AMD authors.
+// NHWC bf16 implicit-GEMM convolution BACKWARD-DATA (dgrad) for gfx950.
+//
+// GEMM view:  dX[M][C] = A[M][rsk] * B[rsk][C]
+//   M = N*H*W (input pixels), rsk = R*S*K (k innermost),
+//   A[m][(r,s,k)] = dout[n, (hi+pad-r)/stride, (wi+pad-s)/stride, k] when the
+//   division is exact and in range, else 0 (zero-fill handles stride>1),
+//   B[(r,s,k)][c] = w[k][r][s][c] read from the RSCK transposed copy
+//   (k_weight_transpose below) so each MFMA B-fragment lane reads 8
+//   consecutive k — contiguous 16 B.
+// Same 128x64x32 block structure as the forward kernel.
+// Requires: K % 8 == 0 and K % 32 == 0 for clean chunks, C % 64 == 0.
+
+#include "conv_common.h"
+
+__global__ void __launch_bounds__(CONV_THREADS)
+k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_rsck,
+             uint16_t* __restrict__ dx, ConvDims d) {
+    const int rsk = d.R * d.S * d.K;
+    const int64_t M = (int64_t)d.N * d.H * d.W;
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6;
+    const int wave_m = wid >> 1;
+    const int wave_n = wid & 1;
+    const int64_t m0 = (int64_t)blockIdx.x * CONV_BM;
+    const int col0 = blockIdx.y * CONV_BN;   // input-channel tile
+
+    __shared__ uint16_t A_lds[CONV_BM * CONV_APITCH];
+
+    int st_hi[2], st_wi[2];
+    int64_t st_n[2];
+    for (int t = 0; t < 2; ++t) {
+        const int chunk = tid + t * CONV_THREADS;
+        const int row = chunk >> 2;
+        const int64_t m = m0 + row;
+        if (m < M) {
+            const int hw = d.H * d.W;
+            st_n[t] = m / hw;
+            const int rem = (int)(m % hw);
+            st_hi[t] = rem / d.W + d.pad;    // hi + pad  (subtract r later)
+            st_wi[t] = rem % d.W + d.pad;
+        } else {
+            st_n[t] = -1;
+        }
+    }
+
+    floatx4 acc[4][2] = {};
+    const int a_row = wave_m * 64 + (lane & 15);
+    const int a_koff = (lane >> 4) * 8;
+    const int b_col = col0 + wave_n * 32 + (lane & 15);
+
+    for (int kc = 0; kc < rsk; kc += CONV_BK) {
+        for (int t = 0; t < 2; ++t) {
+            const int chunk = tid + t * CONV_THREADS;
+            const int row = chunk >> 2;
+            const int c8 = (chunk & 3) * 8;
+            const int kk = kc + c8;
+            const int r = kk / (d.S * d.K);
+            const int sk = kk - r * d.S * d.K;
+            const int s = sk / d.K;
+            const int k = sk - s * d.K;
+            short8 v = {};
+            if (st_n[t] >= 0) {
+                const int hnum = st_hi[t] - r;     // = ho * stride
+                const int wnum = st_wi[t] - s;
+                const int ho = hnum / d.stride;
+                const int wo = wnum / d.stride;
+                if (hnum >= 0 && wnum >= 0 && ho * d.stride == hnum &&
+                    wo * d.stride == wnum && ho < d.Ho && wo < d.Wo) {
+                    const int64_t off =
+                        (((st_n[t] * d.Ho + ho) * d.Wo + wo) * (int64_t)d.K + k);
+                    v = *reinterpret_cast<const short8*>(dout + off);
+                }
+            }
+            *reinterpret_cast<short8*>(&A_lds[row * CONV_APITCH + c8]) = v;
+        }
+        __syncthreads();
+
+        // B fragments: w_rsck[((r*S+s)*C + c) * K + k] — contiguous in k.
+        // kc..kc+31 stays within one (r,s) because K % 32 == 0.
+        const int r = (kc + a_koff) / (d.S * d.K);
+        const int sk = (kc + a_koff) - r * d.S * d.K;
+        const int s = sk / d.K;
+        const int k = sk - s * d.K;
+        short8 b[2];
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf)
+            b[nf] = *reinterpret_cast<const short8*>(
+                w_rsck + ((int64_t)(r * d.S + s) * d.C + b_col + nf * 16) * d.K + k);
+#pragma unroll
+        for (int mf = 0; mf < 4; ++mf) {
+            const short8 a = *reinterpret_cast<const short8*>(
+                &A_lds[(a_row + mf * 16) * CONV_APITCH + a_koff]);
+#pragma unroll
+            for (int nf = 0; nf < 2; ++nf)
+                acc[mf][nf] = MFMA_BF16(a, b[nf], acc[mf][nf]);
+        }
+        __syncthreads();
+    }
+
+    const int64_t out_row0 = m0 + wave_m * 64 + (lane >> 4) * 4;
+    const int out_col0 = col0 + wave_n * 32 + (lane & 15);
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf)
+#pragma unroll
+            for (int rr = 0; rr < 4; ++rr) {
+                const int64_t row = out_row0 + mf * 16 + rr;
+                if (row < M)
+                    dx[row * d.C + out_col0 + nf * 16] =
+                        f32_to_bf16(acc[mf][nf][rr]);
+            }
+}
+
+extern "C" void launch_conv_dgrad(const void* dout, const void* w_rsck,
+                                  void* dx, ConvDims d, hipStream_t stream) {
+    const int64_t M = (int64_t)d.N * d.H * d.W;
+    dim3 grid((unsigned)((M + CONV_BM - 1) / CONV_BM), (unsigned)(d.C / CONV_BN));
+    k_conv_dgrad<<<grid, CONV_THREADS, 0, stream>>>(
+        (const uint16_t*)dout, (const uint16_t*)w_rsck, (uint16_t*)dx, d);
+}
+
+// ---------------------------------------------------------------------------
+// Weight transpose  [K][R*S*C] (torch channels_last) -> [R*S*C][K]  (bf16).
+// Tiny tensors (<= a few MB): simple coalesced-read elementwise kernel.
+// ---------------------------------------------------------------------------
+
+__global__ void __launch_bounds__(256)
+k_weight_transpose(const uint16_t* __restrict__ w, uint16_t* __restrict__ wt,
+                   int K, int rsc) {
+    const int64_t total = (int64_t)K * rsc;
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+         i += stride) {
+        const int k = (int)(i / rsc);
+        const int j = (int)(i - (int64_t)k * rsc);
+        wt[(int64_t)j * K + k] = w[i];
+    }
+}
+
+extern "C" void launch_weight_transpose(const void* w, void* wt, int K, int rsc,
+                                        hipStream_t stream) {
+    k_weight_transpose<<<ew_grid((int64_t)K * rsc, 256, 1), 256, 0, stream>>>(
+        (const uint16_t*)w, (uint16_t*)wt, K, rsc);
+}

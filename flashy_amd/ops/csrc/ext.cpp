@@ -13,7 +13,42 @@
 
 namespace py = pybind11;
 
+struct ConvDims {
+    int N, H, W, C;
+    int K, R, S;
+    int Ho, Wo;
+    int stride, pad;
+};
+
 extern "C" {
+void launch_conv_fwd(const void* x, const void* w, void* y, ConvDims d,
+                     int relu, hipStream_t stream);
+void launch_conv_stem_fwd(const void* x, const void* w, void* y, ConvDims d,
+                          hipStream_t stream);
+void launch_conv_stem_wgrad(const void* x, const void* dout, void* dw,
+                            ConvDims d, hipStream_t stream);
+void launch_conv_dgrad(const void* dout, const void* w_rsck, void* dx,
+                       ConvDims d, hipStream_t stream);
+void launch_weight_transpose(const void* w, void* wt, int K, int rsc,
+                             hipStream_t stream);
+void launch_conv_wgrad(const void* x, const void* dout, void* dw, ConvDims d,
+                       int n_splits, hipStream_t stream);
+void launch_bn_stats(const void* x, void* sums, int64_t M, int C,
+                     hipStream_t stream);
+void launch_bn_finalize(const void* sums, const void* gamma, const void* beta,
+                        void* running_mean, void* running_var, void* work,
+                        int64_t M, int C, float eps, float momentum,
+                        int update_running, hipStream_t stream);
+void launch_bn_apply(const void* x, const void* res, void* y, const void* work,
+                     int64_t M, int C, int relu, hipStream_t stream);
+void launch_bn_bwd_reduce(const void* dy, const void* y, const void* x,
+                          const void* work, void* dz_out, void* bsums,
+                          int64_t M, int C, int relu, hipStream_t stream);
+void launch_bn_bwd_grads(const void* bsums, void* dgamma, void* dbeta, int C,
+                         hipStream_t stream);
+void launch_bn_bwd_apply(const void* dz, const void* x, const void* work,
+                         const void* bsums, void* dx, int64_t M, int C,
+                         hipStream_t stream);
 void launch_fused_sgd(void* p, const void* g, void* m, void* p_bf16, int64_t n,
                       float lr, float momentum, float wd, float grad_scale,
                       int nesterov, hipStream_t stream);
@@ -39,9 +74,116 @@ static void check_last() {
 
 static hipStream_t as_stream(uintptr_t s) { return reinterpret_cast<hipStream_t>(s); }
 
+static ConvDims make_dims(int N, int H, int W, int C, int K, int R, int S,
+                          int Ho, int Wo, int stride, int pad) {
+    return ConvDims{N, H, W, C, K, R, S, Ho, Wo, stride, pad};
+}
+
 PYBIND11_MODULE(_hip_ops, m) {
     m.doc() = "flashy_amd gfx950 kernels";
     m.attr("ARCH") = "gfx950";
+
+    m.def("conv_fwd",
+          [](uintptr_t x, uintptr_t w, uintptr_t y, int N, int H, int W, int C,
+             int K, int R, int S, int Ho, int Wo, int stride, int pad,
+             bool relu, uintptr_t stream) {
+              launch_conv_fwd((const void*)x, (const void*)w, (void*)y,
+                              make_dims(N, H, W, C, K, R, S, Ho, Wo, stride, pad),
+                              relu ? 1 : 0, as_stream(stream));
+              check_last();
+          });
+    m.def("conv_stem_fwd",
+          [](uintptr_t x, uintptr_t w, uintptr_t y, int N, int H, int W, int C,
+             int K, int R, int S, int Ho, int Wo, int stride, int pad,
+             uintptr_t stream) {
+              launch_conv_stem_fwd((const void*)x, (const void*)w, (void*)y,
+                                   make_dims(N, H, W, C, K, R, S, Ho, Wo, stride, pad),
+                                   as_stream(stream));
+              check_last();
+          });
+    m.def("conv_stem_wgrad",
+          [](uintptr_t x, uintptr_t dout, uintptr_t dw, int N, int H, int W,
+             int C, int K, int R, int S, int Ho, int Wo, int stride, int pad,
+             uintptr_t stream) {
+              launch_conv_stem_wgrad((const void*)x, (const void*)dout,
+                                     (void*)dw,
+                                     make_dims(N, H, W, C, K, R, S, Ho, Wo, stride, pad),
+                                     as_stream(stream));
+              check_last();
+          });
+    m.def("conv_dgrad",
+          [](uintptr_t dout, uintptr_t w_rsck, uintptr_t dx, int N, int H,
+             int W, int C, int K, int R, int S, int Ho, int Wo, int stride,
+             int pad, uintptr_t stream) {
+              launch_conv_dgrad((const void*)dout, (const void*)w_rsck,
+                                (void*)dx,
+                                make_dims(N, H, W, C, K, R, S, Ho, Wo, stride, pad),
+                                as_stream(stream));
+              check_last();
+          });
+    m.def("weight_transpose",
+          [](uintptr_t w, uintptr_t wt, int K, int rsc, uintptr_t stream) {
+              launch_weight_transpose((const void*)w, (void*)wt, K, rsc,
+                                      as_stream(stream));
+              check_last();
+          });
+    m.def("conv_wgrad",
+          [](uintptr_t x, uintptr_t dout, uintptr_t dw, int N, int H, int W,
+             int C, int K, int R, int S, int Ho, int Wo, int stride, int pad,
+             int n_splits, uintptr_t stream) {
+              launch_conv_wgrad((const void*)x, (const void*)dout, (void*)dw,
+                                make_dims(N, H, W, C, K, R, S, Ho, Wo, stride, pad),
+                                n_splits, as_stream(stream));
+              check_last();
+          });
+    m.def("bn_stats",
+          [](uintptr_t x, uintptr_t sums, int64_t M, int C, uintptr_t stream) {
+              launch_bn_stats((const void*)x, (void*)sums, M, C, as_stream(stream));
+              check_last();
+          });
+    m.def("bn_finalize",
+          [](uintptr_t sums, uintptr_t gamma, uintptr_t beta, uintptr_t rmean,
+             uintptr_t rvar, uintptr_t work, int64_t M, int C, float eps,
+             float momentum, bool update_running, uintptr_t stream) {
+              launch_bn_finalize((const void*)sums, (const void*)gamma,
+                                 (const void*)beta, (void*)rmean, (void*)rvar,
+                                 (void*)work, M, C, eps, momentum,
+                                 update_running ? 1 : 0, as_stream(stream));
+              check_last();
+          });
+    m.def("bn_apply",
+          [](uintptr_t x, uintptr_t res, uintptr_t y, uintptr_t work, int64_t M,
+             int C, bool relu, uintptr_t stream) {
+              launch_bn_apply((const void*)x, (const void*)res, (void*)y,
+                              (const void*)work, M, C, relu ? 1 : 0,
+                              as_stream(stream));
+              check_last();
+          });
+    m.def("bn_bwd_reduce",
+          [](uintptr_t dy, uintptr_t y, uintptr_t x, uintptr_t work,
+             uintptr_t dz_out, uintptr_t bsums, int64_t M, int C, bool relu,
+             uintptr_t stream) {
+              launch_bn_bwd_reduce((const void*)dy, (const void*)y,
+                                   (const void*)x, (const void*)work,
+                                   (void*)dz_out, (void*)bsums, M, C,
+                                   relu ? 1 : 0, as_stream(stream));
+              check_last();
+          });
+    m.def("bn_bwd_grads",
+          [](uintptr_t bsums, uintptr_t dgamma, uintptr_t dbeta, int C,
+             uintptr_t stream) {
+              launch_bn_bwd_grads((const void*)bsums, (void*)dgamma,
+                                  (void*)dbeta, C, as_stream(stream));
+              check_last();
+          });
+    m.def("bn_bwd_apply",
+          [](uintptr_t dz, uintptr_t x, uintptr_t work, uintptr_t bsums,
+             uintptr_t dx, int64_t M, int C, uintptr_t stream) {
+              launch_bn_bwd_apply((const void*)dz, (const void*)x,
+                                  (const void*)work, (const void*)bsums,
+                                  (void*)dx, M, C, as_stream(stream));
+              check_last();
+          });
 
     m.def("fused_sgd",
           [](uintptr_t p, uintptr_t g, uintptr_t mom, uintptr_t p_bf16,
