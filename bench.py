@@ -30,10 +30,12 @@ from flashy_amd import distrib  # noqa: E402
 from flashy_amd import checkpoint as fckpt  # noqa: E402
 from flashy_amd.functional import cross_entropy  # noqa: E402
 from flashy_amd.graph import CapturedStep  # noqa: E402
-from flashy_amd.models import resnet18, resnet50  # noqa: E402
+from flashy_amd.models import (native_resnet18, native_resnet50,  # noqa: E402
+                               resnet18, resnet50)
 from flashy_amd.optim import FusedSGD  # noqa: E402
 
 MODELS = {"resnet18": resnet18, "resnet50": resnet50}
+NATIVE_MODELS = {"resnet18": native_resnet18, "resnet50": native_resnet50}
 
 
 def build_step(model, optim, static_x, static_y, autocast: bool, distributed: bool):
@@ -94,6 +96,10 @@ def main():
                         help="reference mode: stock torch ops (torch.optim.SGD"
                              " + F.cross_entropy), eager — what the reference"
                              " framework executes on torch-ROCm")
+    parser.add_argument("--torch-model", action="store_true",
+                        help="use the torch-module model (MIOpen convs) with"
+                             " our fused optimizer/loss instead of the native"
+                             " NHWC kernel model")
     parser.add_argument("--no-ckpt", action="store_true",
                         help="skip the checkpoint save/restore measurement")
     parser.add_argument("--channels-last", action="store_true")
@@ -108,10 +114,14 @@ def main():
     torch.backends.cudnn.benchmark = True
     torch.manual_seed(1234 + rank)
 
-    model = MODELS[args.model](num_classes=args.classes,
-                               small_input=args.img <= 64).to(device)
-    if args.channels_last:
-        model = model.to(memory_format=torch.channels_last)
+    native = use_cuda and not args.ref and not args.torch_model
+    if native:
+        model = NATIVE_MODELS[args.model](num_classes=args.classes).to(device)
+    else:
+        model = MODELS[args.model](num_classes=args.classes,
+                                   small_input=args.img <= 64).to(device)
+        if args.channels_last:
+            model = model.to(memory_format=torch.channels_last)
     distrib.broadcast_model(model)
     if use_cuda and not args.ref:
         optim = FusedSGD(model.parameters(), lr=0.1, momentum=0.9,
@@ -132,7 +142,7 @@ def main():
         static_x = static_x.to(memory_format=torch.channels_last)
     static_y = torch.zeros_like(ys[0], device=device)
 
-    autocast = use_cuda
+    autocast = use_cuda and not native  # the native model is bf16 internally
     step = build_step(model, optim, static_x, static_y, autocast, ws > 1)
 
     use_graph = use_cuda and ws == 1 and not args.no_graph and not args.ref
@@ -180,7 +190,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,   # reference publishes no numbers
-            "dtype": "bf16" if autocast else "fp32",
+            "dtype": "bf16" if (autocast or native) else "fp32",
             "data": "synthetic",
             "config": {
                 "model": args.model,
@@ -189,7 +199,8 @@ def main():
                 "img_size": args.img,
                 "num_classes": args.classes,
                 "parallelism": f"dp{ws}",
-                "mode": "reference-torch-ops" if args.ref else "native",
+                "mode": ("reference-torch-ops" if args.ref
+                         else "native-kernels" if native else "torch-model"),
                 "graph": use_graph,
                 "channels_last": args.channels_last,
                 "checkpoint_save_s": ckpt_save_s,
