@@ -125,7 +125,7 @@ def main():
     distrib.broadcast_model(model)
     if use_cuda and not args.ref:
         optim = FusedSGD(model.parameters(), lr=0.1, momentum=0.9,
-                         weight_decay=5e-4)
+                         weight_decay=5e-4, bf16_mirror=native)
     else:
         optim = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9,
                                 weight_decay=5e-4)

@@ -102,10 +102,12 @@ class _BnFn(torch.autograd.Function):
         M = N * H * W
         y = torch.empty_like(x)
         if module.training:
-            sums = torch.zeros(2 * C, dtype=torch.float32, device=x.device)
+            msplit = ops.bn_msplit(M, C)
+            partials = torch.empty(msplit * 2 * C, dtype=torch.float32,
+                                   device=x.device)
             work = torch.empty(4 * C, dtype=torch.float32, device=x.device)
-            ops.bn_stats(x, sums, M, C)
-            ops.bn_finalize(sums, gamma, beta, module.running_mean,
+            ops.bn_stats(x, partials, M, C, msplit)
+            ops.bn_finalize(partials, msplit, gamma, beta, module.running_mean,
                             module.running_var, work, M, C, module.eps,
                             module.momentum, update_running=True)
         else:
@@ -127,12 +129,15 @@ class _BnFn(torch.autograd.Function):
         N, H, W, C = x.shape
         M = N * H * W
         dy = dy.contiguous()
-        bsums = torch.zeros(2 * C, dtype=torch.float32, device=x.device)
+        msplit = ops.bn_msplit(M, C)
+        partials = torch.empty(msplit * 2 * C, dtype=torch.float32,
+                               device=x.device)
+        bsums = torch.empty(2 * C, dtype=torch.float32, device=x.device)
         dz = torch.empty_like(dy)
-        ops.bn_bwd_reduce(dy, y, x, work, dz, bsums, M, C, ctx.relu)
+        ops.bn_bwd_reduce(dy, y, x, work, dz, partials, M, C, msplit, ctx.relu)
         dgamma, g_direct = _grad_target(gamma)
         dbeta, b_direct = _grad_target(beta)
-        ops.bn_bwd_grads(bsums, dgamma, dbeta, C)
+        ops.bn_bwd_grads(partials, msplit, bsums, dgamma, dbeta, C)
         dx = torch.empty_like(x)
         ops.bn_bwd_apply(dz, x, work, bsums, dx, M, C)
         return (dx,

@@ -175,15 +175,25 @@ def conv_wgrad(x: torch.Tensor, dout: torch.Tensor, dw: torch.Tensor,
 # batchnorm kernels (NHWC training BN; see flashy_amd/nn.py)
 # ---------------------------------------------------------------------------
 
-def bn_stats(x: torch.Tensor, sums: torch.Tensor, M: int, C: int) -> None:
-    require().bn_stats(x.data_ptr(), sums.data_ptr(), M, C, _stream())
+def bn_msplit(M: int, C: int) -> int:
+    """Blocks along M for the BN reductions (target ~1024 workgroups)."""
+    cols = max(1, C // 64)
+    msplit = max(1, min(1024 // cols, 1024))
+    return max(1, min(msplit, (M + 3) // 4))
 
 
-def bn_finalize(sums, gamma, beta, rmean, rvar, work, M: int, C: int,
-                eps: float, momentum: float, update_running: bool) -> None:
-    require().bn_finalize(sums.data_ptr(), gamma.data_ptr(), beta.data_ptr(),
-                          rmean.data_ptr(), rvar.data_ptr(), work.data_ptr(),
-                          M, C, eps, momentum, update_running, _stream())
+def bn_stats(x: torch.Tensor, partials: torch.Tensor, M: int, C: int,
+             msplit: int) -> None:
+    require().bn_stats(x.data_ptr(), partials.data_ptr(), M, C, msplit, _stream())
+
+
+def bn_finalize(partials, msplit: int, gamma, beta, rmean, rvar, work,
+                M: int, C: int, eps: float, momentum: float,
+                update_running: bool) -> None:
+    require().bn_finalize(partials.data_ptr(), msplit, gamma.data_ptr(),
+                          beta.data_ptr(), rmean.data_ptr(), rvar.data_ptr(),
+                          work.data_ptr(), M, C, eps, momentum,
+                          update_running, _stream())
 
 
 def bn_apply(x, res, y, work, M: int, C: int, relu: bool) -> None:
@@ -191,15 +201,16 @@ def bn_apply(x, res, y, work, M: int, C: int, relu: bool) -> None:
                        y.data_ptr(), work.data_ptr(), M, C, relu, _stream())
 
 
-def bn_bwd_reduce(dy, y, x, work, dz_out, bsums, M: int, C: int, relu: bool) -> None:
+def bn_bwd_reduce(dy, y, x, work, dz_out, partials, M: int, C: int,
+                  msplit: int, relu: bool) -> None:
     require().bn_bwd_reduce(dy.data_ptr(), y.data_ptr(), x.data_ptr(),
                             work.data_ptr(), dz_out.data_ptr(),
-                            bsums.data_ptr(), M, C, relu, _stream())
+                            partials.data_ptr(), M, C, msplit, relu, _stream())
 
 
-def bn_bwd_grads(bsums, dgamma, dbeta, C: int) -> None:
-    require().bn_bwd_grads(bsums.data_ptr(), dgamma.data_ptr(),
-                           dbeta.data_ptr(), C, _stream())
+def bn_bwd_grads(partials, msplit: int, bsums, dgamma, dbeta, C: int) -> None:
+    require().bn_bwd_grads(partials.data_ptr(), msplit, bsums.data_ptr(),
+                           dgamma.data_ptr(), dbeta.data_ptr(), C, _stream())
 
 
 def bn_bwd_apply(dz, x, work, bsums, dx, M: int, C: int) -> None:

@@ -2,12 +2,15 @@
 // NHWC BatchNorm (training) for gfx950, operating on [M, C] with
 // M = N*H*W, bf16 activations, fp32 stats/params.
 //
-// forward:  stats (per-channel sum/sumsq, atomically accumulated)
-//        -> finalize (tiny: mean/invstd/scale/shift + running stats)
+// forward:  stats (per-channel sum/sumsq partials, one slot per block —
+//                  no atomics, no pre-zeroed buffers)
+//        -> finalize (tiny: combine partials, mean/invstd/scale/shift,
+//                     running stats)
 //        -> apply (y = relu(scale*x + shift [+ res]))  [fused add+ReLU]
-// backward: reduce (dz = dy * relu-mask; per-channel sum_dz, sum_dz*xhat;
+// backward: reduce (dz = dy * relu-mask; per-channel partial sums;
 //                   dz written out — it IS the residual gradient)
-//        -> grads (tiny: dgamma/dbeta accumulated into flat fp32 grads)
+//        -> grads (tiny: combine partials -> bsums; dgamma/dbeta += into
+//                  flat fp32 grads)
 //        -> apply (dx = scale*(dz - (sum_dz + xhat*sum_dzxhat)/M))
 //
 // Replaces the BatchNorm + ReLU + residual-add chains of the reference's
@@ -15,14 +18,10 @@
 
 #include "common.h"
 
-// ---------------------------------------------------------------------------
-// fwd 1: per-channel sum / sumsq.  Block: 256 threads = 64 channels x 4
-// m-lanes; grid (C/64, msplit).  Partial cross-thread reduce via LDS, then
-// one atomicAdd per channel per block.
-// ---------------------------------------------------------------------------
+// partials layout: [msplit][2][C]  (sum, then sumsq/dzxhat)
 
 __global__ void __launch_bounds__(256)
-k_bn_stats(const uint16_t* __restrict__ x, float* __restrict__ sums,
+k_bn_stats(const uint16_t* __restrict__ x, float* __restrict__ partials,
            int64_t M, int C, int m_per_block) {
     const int c = blockIdx.x * 64 + (threadIdx.x & 63);
     const int mlane = threadIdx.x >> 6;  // 0..3
@@ -43,17 +42,14 @@ k_bn_stats(const uint16_t* __restrict__ x, float* __restrict__ sums,
             red[0][2][threadIdx.x] + red[0][3][threadIdx.x];
         s2 = red[1][0][threadIdx.x] + red[1][1][threadIdx.x] +
              red[1][2][threadIdx.x] + red[1][3][threadIdx.x];
-        atomicAdd(&sums[c], s);
-        atomicAdd(&sums[C + c], s2);
+        float* slot = partials + (int64_t)blockIdx.y * 2 * C;
+        slot[c] = s;
+        slot[C + c] = s2;
     }
 }
 
-// ---------------------------------------------------------------------------
-// fwd 2 (tiny): mean/invstd/scale/shift + running-stat update.
 // work[0..C) = mean, [C..2C) = invstd, [2C..3C) = scale, [3C..4C) = shift
-// ---------------------------------------------------------------------------
-
-__global__ void k_bn_finalize(const float* __restrict__ sums,
+__global__ void k_bn_finalize(const float* __restrict__ partials, int msplit,
                               const float* __restrict__ gamma,
                               const float* __restrict__ beta,
                               float* __restrict__ running_mean,
@@ -62,8 +58,13 @@ __global__ void k_bn_finalize(const float* __restrict__ sums,
                               float eps, float momentum, int update_running) {
     const int c = blockIdx.x * blockDim.x + threadIdx.x;
     if (c >= C) return;
-    const float mean = sums[c] / (float)M;
-    float var = sums[C + c] / (float)M - mean * mean;
+    float s = 0.f, s2 = 0.f;
+    for (int i = 0; i < msplit; ++i) {
+        s += partials[(int64_t)i * 2 * C + c];
+        s2 += partials[(int64_t)i * 2 * C + C + c];
+    }
+    const float mean = s / (float)M;
+    float var = s2 / (float)M - mean * mean;
     var = fmaxf(var, 0.f);
     const float invstd = rsqrtf(var + eps);
     const float scale = gamma[c] * invstd;
@@ -77,11 +78,6 @@ __global__ void k_bn_finalize(const float* __restrict__ sums,
         running_var[c] += momentum * (unbiased - running_var[c]);
     }
 }
-
-// ---------------------------------------------------------------------------
-// fwd 3: y = [relu](scale*x + shift [+ res]).  short8-vectorized rows.
-// C % 8 == 0.
-// ---------------------------------------------------------------------------
 
 template <bool RELU, bool RES>
 __global__ void __launch_bounds__(256)
@@ -111,17 +107,11 @@ k_bn_apply(const uint16_t* __restrict__ x, const uint16_t* __restrict__ res,
     }
 }
 
-// ---------------------------------------------------------------------------
-// bwd 1: dz = dy * (y > 0) [if relu]; per-channel sum_dz, sum_dz_xhat.
-// dz is written out (it is also the gradient of the residual input).
-// bsums[0..C) = sum_dz, [C..2C) = sum_dz*xhat.
-// ---------------------------------------------------------------------------
-
 template <bool RELU>
 __global__ void __launch_bounds__(256)
 k_bn_bwd_reduce(const uint16_t* __restrict__ dy, const uint16_t* __restrict__ y,
                 const uint16_t* __restrict__ x, const float* __restrict__ work,
-                uint16_t* __restrict__ dz_out, float* __restrict__ bsums,
+                uint16_t* __restrict__ dz_out, float* __restrict__ partials,
                 int64_t M, int C, int m_per_block) {
     const int c = blockIdx.x * 64 + (threadIdx.x & 63);
     const int mlane = threadIdx.x >> 6;
@@ -147,27 +137,29 @@ k_bn_bwd_reduce(const uint16_t* __restrict__ dy, const uint16_t* __restrict__ y,
             red[0][2][threadIdx.x] + red[0][3][threadIdx.x];
         sx = red[1][0][threadIdx.x] + red[1][1][threadIdx.x] +
              red[1][2][threadIdx.x] + red[1][3][threadIdx.x];
-        atomicAdd(&bsums[c], s);
-        atomicAdd(&bsums[C + c], sx);
+        float* slot = partials + (int64_t)blockIdx.y * 2 * C;
+        slot[c] = s;
+        slot[C + c] = sx;
     }
 }
 
-// ---------------------------------------------------------------------------
-// bwd 2 (tiny): dgamma += sum_dz_xhat ; dbeta += sum_dz  (flat fp32 grads)
-// ---------------------------------------------------------------------------
-
-__global__ void k_bn_bwd_grads(const float* __restrict__ bsums,
+// combine partials -> bsums[2C]; dgamma/dbeta += (flat fp32 grads)
+__global__ void k_bn_bwd_grads(const float* __restrict__ partials, int msplit,
+                               float* __restrict__ bsums,
                                float* __restrict__ dgamma,
                                float* __restrict__ dbeta, int C) {
     const int c = blockIdx.x * blockDim.x + threadIdx.x;
     if (c >= C) return;
-    dbeta[c] += bsums[c];
-    dgamma[c] += bsums[C + c];
+    float s = 0.f, sx = 0.f;
+    for (int i = 0; i < msplit; ++i) {
+        s += partials[(int64_t)i * 2 * C + c];
+        sx += partials[(int64_t)i * 2 * C + C + c];
+    }
+    bsums[c] = s;
+    bsums[C + c] = sx;
+    dbeta[c] += s;
+    dgamma[c] += sx;
 }
-
-// ---------------------------------------------------------------------------
-// bwd 3: dx = scale * (dz - (sum_dz + xhat * sum_dz_xhat) / M)
-// ---------------------------------------------------------------------------
 
 __global__ void __launch_bounds__(256)
 k_bn_bwd_apply(const uint16_t* __restrict__ dz, const uint16_t* __restrict__ x,
@@ -198,37 +190,28 @@ k_bn_bwd_apply(const uint16_t* __restrict__ dz, const uint16_t* __restrict__ x,
 }
 
 // ---------------------------------------------------------------------------
-// launchers
+// launchers (msplit chosen by the Python wrapper, shared by both phases)
 // ---------------------------------------------------------------------------
 
-static int bn_msplit(int64_t M, int C, int* m_per_block) {
-    // target ~1024 blocks total
-    int per = (int)((M * (C / 64) + 1023) / 1024);
-    int blocks_per_col = (int)((M + per - 1) / (per > 0 ? per : 1));
-    if (blocks_per_col < 1) blocks_per_col = 1;
-    if (blocks_per_col > 1024) blocks_per_col = 1024;
-    *m_per_block = (int)((M + blocks_per_col - 1) / blocks_per_col);
-    return blocks_per_col;
-}
-
-extern "C" void launch_bn_stats(const void* x, void* sums, int64_t M, int C,
-                                hipStream_t stream) {
-    int mpb;
-    const int msplit = bn_msplit(M, C, &mpb);
+extern "C" void launch_bn_stats(const void* x, void* partials, int64_t M, int C,
+                                int msplit, hipStream_t stream) {
+    const int mpb = (int)((M + msplit - 1) / msplit);
     dim3 grid((unsigned)(C / 64), (unsigned)msplit);
-    k_bn_stats<<<grid, 256, 0, stream>>>((const uint16_t*)x, (float*)sums, M, C, mpb);
+    k_bn_stats<<<grid, 256, 0, stream>>>((const uint16_t*)x, (float*)partials,
+                                         M, C, mpb);
 }
 
-extern "C" void launch_bn_finalize(const void* sums, const void* gamma,
-                                   const void* beta, void* running_mean,
-                                   void* running_var, void* work, int64_t M,
-                                   int C, float eps, float momentum,
-                                   int update_running, hipStream_t stream) {
+extern "C" void launch_bn_finalize(const void* partials, int msplit,
+                                   const void* gamma, const void* beta,
+                                   void* running_mean, void* running_var,
+                                   void* work, int64_t M, int C, float eps,
+                                   float momentum, int update_running,
+                                   hipStream_t stream) {
     const int block = 256;
     k_bn_finalize<<<(C + block - 1) / block, block, 0, stream>>>(
-        (const float*)sums, (const float*)gamma, (const float*)beta,
-        (float*)running_mean, (float*)running_var, (float*)work, M, C, eps,
-        momentum, update_running);
+        (const float*)partials, msplit, (const float*)gamma,
+        (const float*)beta, (float*)running_mean, (float*)running_var,
+        (float*)work, M, C, eps, momentum, update_running);
 }
 
 extern "C" void launch_bn_apply(const void* x, const void* res, void* y,
@@ -253,26 +236,28 @@ extern "C" void launch_bn_apply(const void* x, const void* res, void* y,
 
 extern "C" void launch_bn_bwd_reduce(const void* dy, const void* y,
                                      const void* x, const void* work,
-                                     void* dz_out, void* bsums, int64_t M,
-                                     int C, int relu, hipStream_t stream) {
-    int mpb;
-    const int msplit = bn_msplit(M, C, &mpb);
+                                     void* dz_out, void* partials, int64_t M,
+                                     int C, int msplit, int relu,
+                                     hipStream_t stream) {
+    const int mpb = (int)((M + msplit - 1) / msplit);
     dim3 grid((unsigned)(C / 64), (unsigned)msplit);
     if (relu)
         k_bn_bwd_reduce<true><<<grid, 256, 0, stream>>>(
             (const uint16_t*)dy, (const uint16_t*)y, (const uint16_t*)x,
-            (const float*)work, (uint16_t*)dz_out, (float*)bsums, M, C, mpb);
+            (const float*)work, (uint16_t*)dz_out, (float*)partials, M, C, mpb);
     else
         k_bn_bwd_reduce<false><<<grid, 256, 0, stream>>>(
             (const uint16_t*)dy, (const uint16_t*)y, (const uint16_t*)x,
-            (const float*)work, (uint16_t*)dz_out, (float*)bsums, M, C, mpb);
+            (const float*)work, (uint16_t*)dz_out, (float*)partials, M, C, mpb);
 }
 
-extern "C" void launch_bn_bwd_grads(const void* bsums, void* dgamma,
-                                    void* dbeta, int C, hipStream_t stream) {
+extern "C" void launch_bn_bwd_grads(const void* partials, int msplit,
+                                    void* bsums, void* dgamma, void* dbeta,
+                                    int C, hipStream_t stream) {
     const int block = 256;
     k_bn_bwd_grads<<<(C + block - 1) / block, block, 0, stream>>>(
-        (const float*)bsums, (float*)dgamma, (float*)dbeta, C);
+        (const float*)partials, msplit, (float*)bsums, (float*)dgamma,
+        (float*)dbeta, C);
 }
 
 extern "C" void launch_bn_bwd_apply(const void* dz, const void* x,

@@ -8,103 +8,122 @@
 //   B[(r,s,k)][c] = w[k][r][s][c] read from the RSCK transposed copy
 //   (k_weight_transpose below) so each MFMA B-fragment lane reads 8
 //   consecutive k — contiguous 16 B.
-// Same 128x64x32 block structure as the forward kernel.
-// Requires: K % 8 == 0 and K % 32 == 0 for clean chunks, C % 64 == 0.
+// Same pipelined BMx64x32 structure as the forward kernel.
+// Requires: K % 32 == 0, C % 64 == 0.
 
 #include "conv_common.h"
 
+template <int BM>
 __global__ void __launch_bounds__(CONV_THREADS)
 k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_rsck,
              uint16_t* __restrict__ dx, ConvDims d) {
+    constexpr int WAVES_M = BM >= 64 ? 2 : 1;
+    constexpr int WAVES_N = 4 / WAVES_M;
+    constexpr int MF = BM / WAVES_M / 16;
+    constexpr int NF = CONV_BN / WAVES_N / 16;
+    constexpr int CHUNKS = BM * (CONV_BK / 8);
+    constexpr int CPT = (CHUNKS + CONV_THREADS - 1) / CONV_THREADS;
+
     const int rsk = d.R * d.S * d.K;
     const int64_t M = (int64_t)d.N * d.H * d.W;
     const int tid = threadIdx.x;
     const int lane = tid & 63;
     const int wid = tid >> 6;
-    const int wave_m = wid >> 1;
-    const int wave_n = wid & 1;
-    const int64_t m0 = (int64_t)blockIdx.x * CONV_BM;
-    const int col0 = blockIdx.y * CONV_BN;   // input-channel tile
+    const int wave_m = WAVES_M == 1 ? 0 : (wid >> 1);
+    const int wave_n = WAVES_M == 1 ? wid : (wid & 1);
+    const int64_t m0 = (int64_t)blockIdx.x * BM;
+    const int col0 = blockIdx.y * CONV_BN;
 
-    __shared__ uint16_t A_lds[CONV_BM * CONV_APITCH];
+    __shared__ uint16_t A_lds[BM * CONV_APITCH];
 
-    int st_hi[2], st_wi[2];
-    int64_t st_n[2];
-    for (int t = 0; t < 2; ++t) {
+    int st_row[CPT], st_hi[CPT], st_wi[CPT];
+    int64_t st_n[CPT];
+#pragma unroll
+    for (int t = 0; t < CPT; ++t) {
         const int chunk = tid + t * CONV_THREADS;
         const int row = chunk >> 2;
+        st_row[t] = row;
         const int64_t m = m0 + row;
-        if (m < M) {
+        if (chunk < CHUNKS && m < M) {
             const int hw = d.H * d.W;
             st_n[t] = m / hw;
             const int rem = (int)(m % hw);
-            st_hi[t] = rem / d.W + d.pad;    // hi + pad  (subtract r later)
+            st_hi[t] = rem / d.W + d.pad;
             st_wi[t] = rem % d.W + d.pad;
         } else {
             st_n[t] = -1;
         }
     }
 
-    floatx4 acc[4][2] = {};
-    const int a_row = wave_m * 64 + (lane & 15);
-    const int a_koff = (lane >> 4) * 8;
-    const int b_col = col0 + wave_n * 32 + (lane & 15);
-
-    for (int kc = 0; kc < rsk; kc += CONV_BK) {
-        for (int t = 0; t < 2; ++t) {
+    auto load_chunk = [&](int t, int kc) -> short8 {
+        short8 v = {};
+        if (st_n[t] >= 0) {
             const int chunk = tid + t * CONV_THREADS;
-            const int row = chunk >> 2;
-            const int c8 = (chunk & 3) * 8;
-            const int kk = kc + c8;
+            const int kk = kc + (chunk & 3) * 8;
             const int r = kk / (d.S * d.K);
             const int sk = kk - r * d.S * d.K;
             const int s = sk / d.K;
             const int k = sk - s * d.K;
-            short8 v = {};
-            if (st_n[t] >= 0) {
-                const int hnum = st_hi[t] - r;     // = ho * stride
-                const int wnum = st_wi[t] - s;
-                const int ho = hnum / d.stride;
-                const int wo = wnum / d.stride;
-                if (hnum >= 0 && wnum >= 0 && ho * d.stride == hnum &&
-                    wo * d.stride == wnum && ho < d.Ho && wo < d.Wo) {
-                    const int64_t off =
-                        (((st_n[t] * d.Ho + ho) * d.Wo + wo) * (int64_t)d.K + k);
-                    v = *reinterpret_cast<const short8*>(dout + off);
-                }
-            }
-            *reinterpret_cast<short8*>(&A_lds[row * CONV_APITCH + c8]) = v;
+            const int hnum = st_hi[t] - r;
+            const int wnum = st_wi[t] - s;
+            const int ho = hnum / d.stride;
+            const int wo = wnum / d.stride;
+            if (hnum >= 0 && wnum >= 0 && ho * d.stride == hnum &&
+                wo * d.stride == wnum && ho < d.Ho && wo < d.Wo)
+                v = *reinterpret_cast<const short8*>(
+                    dout + (((st_n[t] * d.Ho + ho) * d.Wo + wo) * (int64_t)d.K + k));
         }
-        __syncthreads();
+        return v;
+    };
 
-        // B fragments: w_rsck[((r*S+s)*C + c) * K + k] — contiguous in k.
-        // kc..kc+31 stays within one (r,s) because K % 32 == 0.
+    floatx4 acc[MF][NF] = {};
+    const int a_row = wave_m * (BM / WAVES_M) + (lane & 15);
+    const int a_koff = (lane >> 4) * 8;
+    const int b_col = col0 + wave_n * (CONV_BN / WAVES_N) + (lane & 15);
+
+    short8 stage[CPT];
+#pragma unroll
+    for (int t = 0; t < CPT; ++t) stage[t] = load_chunk(t, 0);
+
+    for (int kc = 0; kc < rsk; kc += CONV_BK) {
+        __syncthreads();
+#pragma unroll
+        for (int t = 0; t < CPT; ++t)
+            if (tid + t * CONV_THREADS < CHUNKS)
+                *reinterpret_cast<short8*>(
+                    &A_lds[st_row[t] * CONV_APITCH + ((tid + t * CONV_THREADS) & 3) * 8]) =
+                    stage[t];
+        __syncthreads();
+        if (kc + CONV_BK < rsk) {
+#pragma unroll
+            for (int t = 0; t < CPT; ++t) stage[t] = load_chunk(t, kc + CONV_BK);
+        }
+        // B fragments: (r,s) constant across the 32-chunk since K % 32 == 0
         const int r = (kc + a_koff) / (d.S * d.K);
         const int sk = (kc + a_koff) - r * d.S * d.K;
         const int s = sk / d.K;
         const int k = sk - s * d.K;
-        short8 b[2];
+        short8 b[NF];
 #pragma unroll
-        for (int nf = 0; nf < 2; ++nf)
+        for (int nf = 0; nf < NF; ++nf)
             b[nf] = *reinterpret_cast<const short8*>(
                 w_rsck + ((int64_t)(r * d.S + s) * d.C + b_col + nf * 16) * d.K + k);
 #pragma unroll
-        for (int mf = 0; mf < 4; ++mf) {
+        for (int mf = 0; mf < MF; ++mf) {
             const short8 a = *reinterpret_cast<const short8*>(
                 &A_lds[(a_row + mf * 16) * CONV_APITCH + a_koff]);
 #pragma unroll
-            for (int nf = 0; nf < 2; ++nf)
+            for (int nf = 0; nf < NF; ++nf)
                 acc[mf][nf] = MFMA_BF16(a, b[nf], acc[mf][nf]);
         }
-        __syncthreads();
     }
 
-    const int64_t out_row0 = m0 + wave_m * 64 + (lane >> 4) * 4;
-    const int out_col0 = col0 + wave_n * 32 + (lane & 15);
+    const int64_t out_row0 = m0 + wave_m * (BM / WAVES_M) + (lane >> 4) * 4;
+    const int out_col0 = col0 + wave_n * (CONV_BN / WAVES_N) + (lane & 15);
 #pragma unroll
-    for (int mf = 0; mf < 4; ++mf)
+    for (int mf = 0; mf < MF; ++mf)
 #pragma unroll
-        for (int nf = 0; nf < 2; ++nf)
+        for (int nf = 0; nf < NF; ++nf)
 #pragma unroll
             for (int rr = 0; rr < 4; ++rr) {
                 const int64_t row = out_row0 + mf * 16 + rr;
@@ -117,14 +136,22 @@ k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_r
 extern "C" void launch_conv_dgrad(const void* dout, const void* w_rsck,
                                   void* dx, ConvDims d, hipStream_t stream) {
     const int64_t M = (int64_t)d.N * d.H * d.W;
-    dim3 grid((unsigned)((M + CONV_BM - 1) / CONV_BM), (unsigned)(d.C / CONV_BN));
-    k_conv_dgrad<<<grid, CONV_THREADS, 0, stream>>>(
-        (const uint16_t*)dout, (const uint16_t*)w_rsck, (uint16_t*)dx, d);
+    const int ktiles = d.C / CONV_BN;
+    int bm = 32;
+    for (int cand : {128, 64}) {
+        if ((M + cand - 1) / cand * ktiles >= 208) { bm = cand; break; }
+    }
+    dim3 grid((unsigned)((M + bm - 1) / bm), (unsigned)(d.C / CONV_BN));
+    auto dd = (const uint16_t*)dout;
+    auto ww = (const uint16_t*)w_rsck;
+    auto xx = (uint16_t*)dx;
+    if (bm == 128) k_conv_dgrad<128><<<grid, CONV_THREADS, 0, stream>>>(dd, ww, xx, d);
+    else if (bm == 64) k_conv_dgrad<64><<<grid, CONV_THREADS, 0, stream>>>(dd, ww, xx, d);
+    else k_conv_dgrad<32><<<grid, CONV_THREADS, 0, stream>>>(dd, ww, xx, d);
 }
 
 // ---------------------------------------------------------------------------
-// Weight transpose  [K][R*S*C] (torch channels_last) -> [R*S*C][K]  (bf16).
-// Tiny tensors (<= a few MB): simple coalesced-read elementwise kernel.
+// Weight transpose  [K][R*S*C] -> [R*S*C][K]  (bf16).
 // ---------------------------------------------------------------------------
 
 __global__ void __launch_bounds__(256)

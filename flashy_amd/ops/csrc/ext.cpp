@@ -33,19 +33,21 @@ void launch_weight_transpose(const void* w, void* wt, int K, int rsc,
                              hipStream_t stream);
 void launch_conv_wgrad(const void* x, const void* dout, void* dw, ConvDims d,
                        int n_splits, hipStream_t stream);
-void launch_bn_stats(const void* x, void* sums, int64_t M, int C,
-                     hipStream_t stream);
-void launch_bn_finalize(const void* sums, const void* gamma, const void* beta,
-                        void* running_mean, void* running_var, void* work,
-                        int64_t M, int C, float eps, float momentum,
-                        int update_running, hipStream_t stream);
+void launch_bn_stats(const void* x, void* partials, int64_t M, int C,
+                     int msplit, hipStream_t stream);
+void launch_bn_finalize(const void* partials, int msplit, const void* gamma,
+                        const void* beta, void* running_mean,
+                        void* running_var, void* work, int64_t M, int C,
+                        float eps, float momentum, int update_running,
+                        hipStream_t stream);
 void launch_bn_apply(const void* x, const void* res, void* y, const void* work,
                      int64_t M, int C, int relu, hipStream_t stream);
 void launch_bn_bwd_reduce(const void* dy, const void* y, const void* x,
-                          const void* work, void* dz_out, void* bsums,
-                          int64_t M, int C, int relu, hipStream_t stream);
-void launch_bn_bwd_grads(const void* bsums, void* dgamma, void* dbeta, int C,
-                         hipStream_t stream);
+                          const void* work, void* dz_out, void* partials,
+                          int64_t M, int C, int msplit, int relu,
+                          hipStream_t stream);
+void launch_bn_bwd_grads(const void* partials, int msplit, void* bsums,
+                         void* dgamma, void* dbeta, int C, hipStream_t stream);
 void launch_bn_bwd_apply(const void* dz, const void* x, const void* work,
                          const void* bsums, void* dx, int64_t M, int C,
                          hipStream_t stream);
@@ -137,18 +139,21 @@ PYBIND11_MODULE(_hip_ops, m) {
               check_last();
           });
     m.def("bn_stats",
-          [](uintptr_t x, uintptr_t sums, int64_t M, int C, uintptr_t stream) {
-              launch_bn_stats((const void*)x, (void*)sums, M, C, as_stream(stream));
+          [](uintptr_t x, uintptr_t partials, int64_t M, int C, int msplit,
+             uintptr_t stream) {
+              launch_bn_stats((const void*)x, (void*)partials, M, C, msplit,
+                              as_stream(stream));
               check_last();
           });
     m.def("bn_finalize",
-          [](uintptr_t sums, uintptr_t gamma, uintptr_t beta, uintptr_t rmean,
-             uintptr_t rvar, uintptr_t work, int64_t M, int C, float eps,
-             float momentum, bool update_running, uintptr_t stream) {
-              launch_bn_finalize((const void*)sums, (const void*)gamma,
-                                 (const void*)beta, (void*)rmean, (void*)rvar,
-                                 (void*)work, M, C, eps, momentum,
-                                 update_running ? 1 : 0, as_stream(stream));
+          [](uintptr_t partials, int msplit, uintptr_t gamma, uintptr_t beta,
+             uintptr_t rmean, uintptr_t rvar, uintptr_t work, int64_t M, int C,
+             float eps, float momentum, bool update_running, uintptr_t stream) {
+              launch_bn_finalize((const void*)partials, msplit,
+                                 (const void*)gamma, (const void*)beta,
+                                 (void*)rmean, (void*)rvar, (void*)work, M, C,
+                                 eps, momentum, update_running ? 1 : 0,
+                                 as_stream(stream));
               check_last();
           });
     m.def("bn_apply",
@@ -161,19 +166,20 @@ PYBIND11_MODULE(_hip_ops, m) {
           });
     m.def("bn_bwd_reduce",
           [](uintptr_t dy, uintptr_t y, uintptr_t x, uintptr_t work,
-             uintptr_t dz_out, uintptr_t bsums, int64_t M, int C, bool relu,
-             uintptr_t stream) {
+             uintptr_t dz_out, uintptr_t partials, int64_t M, int C,
+             int msplit, bool relu, uintptr_t stream) {
               launch_bn_bwd_reduce((const void*)dy, (const void*)y,
                                    (const void*)x, (const void*)work,
-                                   (void*)dz_out, (void*)bsums, M, C,
-                                   relu ? 1 : 0, as_stream(stream));
+                                   (void*)dz_out, (void*)partials, M, C,
+                                   msplit, relu ? 1 : 0, as_stream(stream));
               check_last();
           });
     m.def("bn_bwd_grads",
-          [](uintptr_t bsums, uintptr_t dgamma, uintptr_t dbeta, int C,
-             uintptr_t stream) {
-              launch_bn_bwd_grads((const void*)bsums, (void*)dgamma,
-                                  (void*)dbeta, C, as_stream(stream));
+          [](uintptr_t partials, int msplit, uintptr_t bsums, uintptr_t dgamma,
+             uintptr_t dbeta, int C, uintptr_t stream) {
+              launch_bn_bwd_grads((const void*)partials, msplit, (void*)bsums,
+                                  (void*)dgamma, (void*)dbeta, C,
+                                  as_stream(stream));
               check_last();
           });
     m.def("bn_bwd_apply",

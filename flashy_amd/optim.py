@@ -36,7 +36,8 @@ ParamsT = tp.Iterable[torch.nn.Parameter]
 class _Group:
     """One (device, dtype) flat group."""
 
-    def __init__(self, params: tp.List[torch.nn.Parameter]):
+    def __init__(self, params: tp.List[torch.nn.Parameter],
+                 bf16_mirror: bool = False):
         self.params = params
         self.device = params[0].device
         self.dtype = params[0].dtype
@@ -59,6 +60,22 @@ class _Group:
             n = p.numel()
             p.grad = self.flat_g[offset:offset + n].view_as(p)
             offset += n
+        # optional bf16 mirror of the packed params, refreshed by the fused
+        # optimizer kernel in the same pass as the update; modules read the
+        # per-param view via `param._bf16_mirror` (flashy_amd/nn.py) so the
+        # step needs zero weight-cast kernels.
+        self.flat_p16: tp.Optional[torch.Tensor] = None
+        if bf16_mirror:
+            self.flat_p16 = self.flat_p.to(torch.bfloat16)
+            offset = 0
+            for p in params:
+                n = p.numel()
+                p._bf16_mirror = self.flat_p16[offset:offset + n].view_as(p)
+                offset += n
+
+    def refresh_bf16(self) -> None:
+        if self.flat_p16 is not None:
+            self.flat_p16.copy_(self.flat_p)
 
     def buffers_like(self) -> torch.Tensor:
         return torch.zeros_like(self.flat_p)
@@ -67,7 +84,8 @@ class _Group:
 class FlatOptimizer:
     """Base: flat parameter/grad packing + state dict plumbing."""
 
-    def __init__(self, params: ParamsT, defaults: tp.Dict[str, tp.Any]):
+    def __init__(self, params: ParamsT, defaults: tp.Dict[str, tp.Any],
+                 bf16_mirror: bool = False):
         params = [p for p in params if p.requires_grad]
         if not params:
             raise ValueError("no trainable parameters")
@@ -78,7 +96,7 @@ class FlatOptimizer:
         by_key: tp.Dict[tp.Any, tp.List[torch.nn.Parameter]] = {}
         for p in params:
             by_key.setdefault((p.device, p.dtype), []).append(p)
-        self.groups = [_Group(ps) for ps in by_key.values()]
+        self.groups = [_Group(ps, bf16_mirror) for ps in by_key.values()]
         self.defaults = dict(defaults)
         self.step_count = 0
         self._use_hip = any(g.device.type == "cuda" for g in self.groups)
@@ -100,6 +118,12 @@ class FlatOptimizer:
     @property
     def param_buffers(self) -> tp.List[torch.Tensor]:
         return [g.flat_p for g in self.groups]
+
+    def refresh_bf16(self) -> None:
+        """Re-sync the bf16 mirrors after any out-of-band param mutation
+        (checkpoint restore, broadcast_model)."""
+        for g in self.groups:
+            g.refresh_bf16()
 
     def step(self, closure: tp.Optional[tp.Callable] = None) -> None:
         loss = closure() if closure is not None else None
@@ -136,10 +160,11 @@ class FusedSGD(FlatOptimizer):
     fused HIP kernel over the flat buffers."""
 
     def __init__(self, params: ParamsT, lr: float, momentum: float = 0.0,
-                 weight_decay: float = 0.0, nesterov: bool = False):
+                 weight_decay: float = 0.0, nesterov: bool = False,
+                 bf16_mirror: bool = False):
         super().__init__(params, dict(lr=lr, momentum=momentum,
                                       weight_decay=weight_decay,
-                                      nesterov=nesterov))
+                                      nesterov=nesterov), bf16_mirror)
         self._momentum_buffers = [
             g.buffers_like() if momentum != 0 else None for g in self.groups]
 
@@ -149,7 +174,7 @@ class FusedSGD(FlatOptimizer):
         if group.device.type == "cuda":
             ops.fused_sgd(group.flat_p, group.flat_g, m, d["lr"],
                           d["momentum"], d["weight_decay"],
-                          nesterov=d["nesterov"])
+                          nesterov=d["nesterov"], p_bf16=group.flat_p16)
             return
         # CPU fallback, identical math
         with torch.no_grad():
@@ -160,6 +185,7 @@ class FusedSGD(FlatOptimizer):
                 m.mul_(d["momentum"]).add_(g)
                 g = g.add(m, alpha=d["momentum"]) if d["nesterov"] else m
             group.flat_p.add_(g, alpha=-d["lr"])
+            group.refresh_bf16()
 
     def _extra_state(self):
         return {"momentum_buffers": self._momentum_buffers}
@@ -178,10 +204,10 @@ class FusedAdam(FlatOptimizer):
     def __init__(self, params: ParamsT, lr: float = 1e-3,
                  betas: tp.Tuple[float, float] = (0.9, 0.999),
                  eps: float = 1e-8, weight_decay: float = 0.0,
-                 adamw: bool = False):
+                 adamw: bool = False, bf16_mirror: bool = False):
         super().__init__(params, dict(lr=lr, beta1=betas[0], beta2=betas[1],
                                       eps=eps, weight_decay=weight_decay,
-                                      adamw=adamw))
+                                      adamw=adamw), bf16_mirror)
         self._exp_avg = [g.buffers_like() for g in self.groups]
         self._exp_avg_sq = [g.buffers_like() for g in self.groups]
 
@@ -193,7 +219,7 @@ class FusedAdam(FlatOptimizer):
             ops.fused_adam(group.flat_p, group.flat_g, m, v, d["lr"],
                            d["beta1"], d["beta2"], d["eps"],
                            d["weight_decay"], self.step_count,
-                           adamw=d["adamw"])
+                           adamw=d["adamw"], p_bf16=group.flat_p16)
             return
         with torch.no_grad():
             g = group.flat_g
@@ -208,6 +234,7 @@ class FusedAdam(FlatOptimizer):
             bc2 = 1 - d["beta2"] ** self.step_count
             denom = (v / bc2).sqrt_().add_(d["eps"])
             p.addcdiv_(m / bc1, denom, value=-d["lr"])
+            group.refresh_bf16()
 
     def _extra_state(self):
         return {"exp_avg": self._exp_avg, "exp_avg_sq": self._exp_avg_sq}
