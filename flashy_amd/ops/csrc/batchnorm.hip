@@ -48,21 +48,41 @@ k_bn_stats(const uint16_t* __restrict__ x, float* __restrict__ partials,
     }
 }
 
-// work[0..C) = mean, [C..2C) = invstd, [2C..3C) = scale, [3C..4C) = shift
-__global__ void k_bn_finalize(const float* __restrict__ partials, int msplit,
-                              const float* __restrict__ gamma,
-                              const float* __restrict__ beta,
-                              float* __restrict__ running_mean,
-                              float* __restrict__ running_var,
-                              float* __restrict__ work, int64_t M, int C,
-                              float eps, float momentum, int update_running) {
-    const int c = blockIdx.x * blockDim.x + threadIdx.x;
-    if (c >= C) return;
+// Parallel combine of [msplit][2][C] partials: 256 threads = 64 channels x
+// 4 split-lanes, many loads in flight, LDS reduce.  Grid C/64.
+__device__ __forceinline__ void combine_partials(
+        const float* __restrict__ partials, int msplit, int C,
+        float* s_out, float* s2_out) {
+    const int c = blockIdx.x * 64 + (threadIdx.x & 63);
+    const int slane = threadIdx.x >> 6;  // 0..3
     float s = 0.f, s2 = 0.f;
-    for (int i = 0; i < msplit; ++i) {
+    for (int i = slane; i < msplit; i += 4) {
         s += partials[(int64_t)i * 2 * C + c];
         s2 += partials[(int64_t)i * 2 * C + C + c];
     }
+    __shared__ float red[2][4][64];
+    red[0][slane][threadIdx.x & 63] = s;
+    red[1][slane][threadIdx.x & 63] = s2;
+    __syncthreads();
+    *s_out = red[0][0][threadIdx.x & 63] + red[0][1][threadIdx.x & 63] +
+             red[0][2][threadIdx.x & 63] + red[0][3][threadIdx.x & 63];
+    *s2_out = red[1][0][threadIdx.x & 63] + red[1][1][threadIdx.x & 63] +
+              red[1][2][threadIdx.x & 63] + red[1][3][threadIdx.x & 63];
+}
+
+// work[0..C) = mean, [C..2C) = invstd, [2C..3C) = scale, [3C..4C) = shift
+__global__ void __launch_bounds__(256)
+k_bn_finalize(const float* __restrict__ partials, int msplit,
+              const float* __restrict__ gamma,
+              const float* __restrict__ beta,
+              float* __restrict__ running_mean,
+              float* __restrict__ running_var,
+              float* __restrict__ work, int64_t M, int C,
+              float eps, float momentum, int update_running) {
+    float s, s2;
+    combine_partials(partials, msplit, C, &s, &s2);
+    if (threadIdx.x >= 64) return;
+    const int c = blockIdx.x * 64 + threadIdx.x;
     const float mean = s / (float)M;
     float var = s2 / (float)M - mean * mean;
     var = fmaxf(var, 0.f);
@@ -144,17 +164,15 @@ k_bn_bwd_reduce(const uint16_t* __restrict__ dy, const uint16_t* __restrict__ y,
 }
 
 // combine partials -> bsums[2C]; dgamma/dbeta += (flat fp32 grads)
-__global__ void k_bn_bwd_grads(const float* __restrict__ partials, int msplit,
-                               float* __restrict__ bsums,
-                               float* __restrict__ dgamma,
-                               float* __restrict__ dbeta, int C) {
-    const int c = blockIdx.x * blockDim.x + threadIdx.x;
-    if (c >= C) return;
-    float s = 0.f, sx = 0.f;
-    for (int i = 0; i < msplit; ++i) {
-        s += partials[(int64_t)i * 2 * C + c];
-        sx += partials[(int64_t)i * 2 * C + C + c];
-    }
+__global__ void __launch_bounds__(256)
+k_bn_bwd_grads(const float* __restrict__ partials, int msplit,
+               float* __restrict__ bsums,
+               float* __restrict__ dgamma,
+               float* __restrict__ dbeta, int C) {
+    float s, sx;
+    combine_partials(partials, msplit, C, &s, &sx);
+    if (threadIdx.x >= 64) return;
+    const int c = blockIdx.x * 64 + threadIdx.x;
     bsums[c] = s;
     bsums[C + c] = sx;
     dbeta[c] += s;
@@ -207,8 +225,7 @@ extern "C" void launch_bn_finalize(const void* partials, int msplit,
                                    void* work, int64_t M, int C, float eps,
                                    float momentum, int update_running,
                                    hipStream_t stream) {
-    const int block = 256;
-    k_bn_finalize<<<(C + block - 1) / block, block, 0, stream>>>(
+    k_bn_finalize<<<C / 64, 256, 0, stream>>>(
         (const float*)partials, msplit, (const float*)gamma,
         (const float*)beta, (float*)running_mean, (float*)running_var,
         (float*)work, M, C, eps, momentum, update_running);
@@ -254,8 +271,7 @@ extern "C" void launch_bn_bwd_reduce(const void* dy, const void* y,
 extern "C" void launch_bn_bwd_grads(const void* partials, int msplit,
                                     void* bsums, void* dgamma, void* dbeta,
                                     int C, hipStream_t stream) {
-    const int block = 256;
-    k_bn_bwd_grads<<<(C + block - 1) / block, block, 0, stream>>>(
+    k_bn_bwd_grads<<<C / 64, 256, 0, stream>>>(
         (const float*)partials, msplit, (float*)bsums, (float*)dgamma,
         (float*)dbeta, C);
 }

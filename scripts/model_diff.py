@@ -18,9 +18,10 @@ print("max diff", (logits_n - logits_t).abs().max().item(),
       "std", logits_t.std().item())
 ln.backward()
 lt.backward()
-gn = model.layer1[0].conv1.weight.grad.permute(0, 3, 1, 2)
-gt = twin.layer1[0].conv1.weight.grad
-print("grad rel", ((gn - gt).abs().max() / (gt.abs().max() + 1e-8)).item())
+gn = model.layer1[0].conv1.weight.grad.permute(0, 3, 1, 2).flatten()
+gt = twin.layer1[0].conv1.weight.grad.flatten()
+print("grad relL2", ((gn - gt).norm() / (gt.norm() + 1e-8)).item(),
+      "cos", torch.nn.functional.cosine_similarity(gn, gt, dim=0).item())
 
 # per-stage comparison to localize drift
 with torch.no_grad():

@@ -194,11 +194,14 @@ def test_native_resnet18_matches_torch():
     assert diff / spread < 0.35, (diff, spread)
     assert abs(loss_n.item() - loss_t.item()) < 0.25, \
         (loss_n.item(), loss_t.item())
-    # spot-check a conv weight grad (bf16 path vs fp32 torch)
-    gn = model.layer1[0].conv1.weight.grad.permute(0, 3, 1, 2)
-    gt = twin.layer1[0].conv1.weight.grad
-    scale = gt.abs().max().item() + 1e-8
-    assert ((gn - gt).abs().max().item() / scale) < 0.15
+    # spot-check a conv weight grad (bf16 path vs fp32 torch): element-wise
+    # max is inflated by ReLU-mask flips at bf16 rounding boundaries, so use
+    # relative L2 + cosine similarity
+    gn = model.layer1[0].conv1.weight.grad.permute(0, 3, 1, 2).flatten()
+    gt = twin.layer1[0].conv1.weight.grad.flatten()
+    rel = (gn - gt).norm().item() / (gt.norm().item() + 1e-8)
+    cos = torch.nn.functional.cosine_similarity(gn, gt, dim=0).item()
+    assert rel < 0.2 and cos > 0.98, (rel, cos)
 
 
 @requires_gpu
