@@ -176,9 +176,10 @@ def conv_wgrad(x: torch.Tensor, dout: torch.Tensor, dw: torch.Tensor,
 # ---------------------------------------------------------------------------
 
 def bn_msplit(M: int, C: int) -> int:
-    """Blocks along M for the BN reductions (target ~1024 workgroups)."""
+    """Blocks along M for the BN reductions: ~256 blocks saturate the chip
+    while keeping the partial-combine kernels cheap."""
     cols = max(1, C // 64)
-    msplit = max(1, min(1024 // cols, 1024))
+    msplit = max(1, min(256 // cols, 256))
     return max(1, min(msplit, (M + 3) // 4))
 
 

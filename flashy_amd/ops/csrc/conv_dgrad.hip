@@ -21,7 +21,8 @@ k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_r
     constexpr int WAVES_N = 4 / WAVES_M;
     constexpr int MF = BM / WAVES_M / 16;
     constexpr int NF = CONV_BN / WAVES_N / 16;
-    constexpr int CHUNKS = BM * (CONV_BK / 8);
+    constexpr int BK2 = 2 * CONV_BK;
+    constexpr int CHUNKS = BM * (BK2 / 8);
     constexpr int CPT = (CHUNKS + CONV_THREADS - 1) / CONV_THREADS;
 
     const int rsk = d.R * d.S * d.K;
@@ -34,14 +35,14 @@ k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_r
     const int64_t m0 = (int64_t)blockIdx.x * BM;
     const int col0 = blockIdx.y * CONV_BN;
 
-    __shared__ uint16_t A_lds[BM * CONV_APITCH];
+    __shared__ uint16_t A_lds[2][BM * 2 * CONV_APITCH];
 
     int st_row[CPT], st_hi[CPT], st_wi[CPT];
     int64_t st_n[CPT];
 #pragma unroll
     for (int t = 0; t < CPT; ++t) {
         const int chunk = tid + t * CONV_THREADS;
-        const int row = chunk >> 2;
+        const int row = chunk >> 3;
         st_row[t] = row;
         const int64_t m = m0 + row;
         if (chunk < CHUNKS && m < M) {
@@ -57,9 +58,9 @@ k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_r
 
     auto load_chunk = [&](int t, int kc) -> short8 {
         short8 v = {};
-        if (st_n[t] >= 0) {
-            const int chunk = tid + t * CONV_THREADS;
-            const int kk = kc + (chunk & 3) * 8;
+        const int chunk = tid + t * CONV_THREADS;
+        const int kk = kc + (chunk & 7) * 8;
+        if (st_n[t] >= 0 && kk < rsk) {
             const int r = kk / (d.S * d.K);
             const int sk = kk - r * d.S * d.K;
             const int s = sk / d.K;
@@ -75,47 +76,70 @@ k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_r
         }
         return v;
     };
+    auto lds_write = [&](uint16_t* buf, int t, short8 v) {
+        const int chunk = tid + t * CONV_THREADS;
+        const int koff = (chunk & 7) * 8;
+        const int sub = koff >> 5;
+        *reinterpret_cast<short8*>(
+            &buf[(sub * BM + st_row[t]) * CONV_APITCH + (koff & 31)]) = v;
+    };
 
     floatx4 acc[MF][NF] = {};
     const int a_row = wave_m * (BM / WAVES_M) + (lane & 15);
     const int a_koff = (lane >> 4) * 8;
     const int b_col = col0 + wave_n * (CONV_BN / WAVES_N) + (lane & 15);
 
+    const int n_stages = (rsk + BK2 - 1) / BK2;
     short8 stage[CPT];
 #pragma unroll
     for (int t = 0; t < CPT; ++t) stage[t] = load_chunk(t, 0);
+#pragma unroll
+    for (int t = 0; t < CPT; ++t)
+        if (tid + t * CONV_THREADS < CHUNKS) lds_write(A_lds[0], t, stage[t]);
+    if (n_stages > 1) {
+#pragma unroll
+        for (int t = 0; t < CPT; ++t) stage[t] = load_chunk(t, BK2);
+    }
+    __syncthreads();
 
-    for (int kc = 0; kc < rsk; kc += CONV_BK) {
-        __syncthreads();
+    for (int i = 0; i < n_stages; ++i) {
+        const uint16_t* buf = A_lds[i & 1];
+        if (i + 1 < n_stages) {
+            uint16_t* nbuf = A_lds[(i + 1) & 1];
 #pragma unroll
-        for (int t = 0; t < CPT; ++t)
-            if (tid + t * CONV_THREADS < CHUNKS)
-                *reinterpret_cast<short8*>(
-                    &A_lds[st_row[t] * CONV_APITCH + ((tid + t * CONV_THREADS) & 3) * 8]) =
-                    stage[t];
-        __syncthreads();
-        if (kc + CONV_BK < rsk) {
+            for (int t = 0; t < CPT; ++t)
+                if (tid + t * CONV_THREADS < CHUNKS) lds_write(nbuf, t, stage[t]);
+            if (i + 2 < n_stages) {
 #pragma unroll
-            for (int t = 0; t < CPT; ++t) stage[t] = load_chunk(t, kc + CONV_BK);
+                for (int t = 0; t < CPT; ++t)
+                    stage[t] = load_chunk(t, (i + 2) * BK2);
+            }
         }
-        // B fragments: (r,s) constant across the 32-chunk since K % 32 == 0
-        const int r = (kc + a_koff) / (d.S * d.K);
-        const int sk = (kc + a_koff) - r * d.S * d.K;
-        const int s = sk / d.K;
-        const int k = sk - s * d.K;
-        short8 b[NF];
+        const int kc = i * BK2;
 #pragma unroll
-        for (int nf = 0; nf < NF; ++nf)
-            b[nf] = *reinterpret_cast<const short8*>(
-                w_rsck + ((int64_t)(r * d.S + s) * d.C + b_col + nf * 16) * d.K + k);
-#pragma unroll
-        for (int mf = 0; mf < MF; ++mf) {
-            const short8 a = *reinterpret_cast<const short8*>(
-                &A_lds[(a_row + mf * 16) * CONV_APITCH + a_koff]);
+        for (int sub = 0; sub < 2; ++sub) {
+            const int kbase = kc + sub * CONV_BK;
+            if (kbase >= rsk) break;
+            // (r,s) constant across the 32-chunk since K % 32 == 0
+            const int r = (kbase + a_koff) / (d.S * d.K);
+            const int sk = (kbase + a_koff) - r * d.S * d.K;
+            const int sidx = sk / d.K;
+            const int k = sk - sidx * d.K;
+            short8 b[NF];
 #pragma unroll
             for (int nf = 0; nf < NF; ++nf)
-                acc[mf][nf] = MFMA_BF16(a, b[nf], acc[mf][nf]);
+                b[nf] = *reinterpret_cast<const short8*>(
+                    w_rsck + ((int64_t)(r * d.S + sidx) * d.C + b_col + nf * 16) * d.K + k);
+#pragma unroll
+            for (int mf = 0; mf < MF; ++mf) {
+                const short8 a = *reinterpret_cast<const short8*>(
+                    &buf[(sub * BM + a_row + mf * 16) * CONV_APITCH + a_koff]);
+#pragma unroll
+                for (int nf = 0; nf < NF; ++nf)
+                    acc[mf][nf] = MFMA_BF16(a, b[nf], acc[mf][nf]);
+            }
         }
+        __syncthreads();
     }
 
     const int64_t out_row0 = m0 + wave_m * (BM / WAVES_M) + (lane >> 4) * 4;

@@ -27,7 +27,8 @@ k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
     constexpr int WAVES_N = 4 / WAVES_M;
     constexpr int MF = BM / WAVES_M / 16;      // m fragments per wave
     constexpr int NF = CONV_BN / WAVES_N / 16; // n fragments per wave
-    constexpr int CHUNKS = BM * (CONV_BK / 8); // 16B staging chunks
+    constexpr int BK2 = 2 * CONV_BK;           // 64-deep stage (2 MFMA-K)
+    constexpr int CHUNKS = BM * (BK2 / 8);     // 16B staging chunks per stage
     constexpr int CPT = (CHUNKS + CONV_THREADS - 1) / CONV_THREADS;
 
     const int rsc = d.R * d.S * d.C;
@@ -40,14 +41,16 @@ k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
     const int64_t m0 = (int64_t)blockIdx.x * BM;
     const int col0 = blockIdx.y * CONV_BN;
 
-    __shared__ uint16_t A_lds[BM * CONV_APITCH];
+    // two stage buffers, each [2 kk-subchunks][BM][APITCH] (sub-major keeps
+    // the row pitch at 48 elements = conflict-free ds_read_b128 groups)
+    __shared__ uint16_t A_lds[2][BM * 2 * CONV_APITCH];
 
     int st_row[CPT], st_hi[CPT], st_wi[CPT];
     int64_t st_n[CPT];
 #pragma unroll
     for (int t = 0; t < CPT; ++t) {
         const int chunk = tid + t * CONV_THREADS;
-        const int row = chunk >> 2;
+        const int row = chunk >> 3;            // 8 chunks per row (64 kk)
         st_row[t] = row;
         const int64_t m = m0 + row;
         if (chunk < CHUNKS && m < M) {
@@ -61,12 +64,12 @@ k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
         }
     }
 
-    // chunk load for reduction offset kc (16 B per staged chunk)
+    // chunk load for stage base offset kc (16 B per staged chunk)
     auto load_chunk = [&](int t, int kc) -> short8 {
         short8 v = {};
-        if (st_n[t] >= 0) {
-            const int chunk = tid + t * CONV_THREADS;
-            const int kk = kc + (chunk & 3) * 8;
+        const int chunk = tid + t * CONV_THREADS;
+        const int kk = kc + (chunk & 7) * 8;
+        if (st_n[t] >= 0 && kk < rsc) {
             const int r = kk / (d.S * d.C);
             const int sc = kk - r * d.S * d.C;
             const int s = sc / d.C;
@@ -79,42 +82,68 @@ k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
         }
         return v;
     };
+    // LDS layout: [sub][row][APITCH], sub = kk/32 within the 64-deep stage
+    auto lds_write = [&](uint16_t* buf, int t, short8 v) {
+        const int chunk = tid + t * CONV_THREADS;
+        const int koff = (chunk & 7) * 8;       // 0..56 within the 64 stage
+        const int sub = koff >> 5;
+        *reinterpret_cast<short8*>(
+            &buf[(sub * BM + st_row[t]) * CONV_APITCH + (koff & 31)]) = v;
+    };
 
     floatx4 acc[MF][NF] = {};
     const int a_row = wave_m * (BM / WAVES_M) + (lane & 15);
     const int a_koff = (lane >> 4) * 8;
     const int b_col = col0 + wave_n * (CONV_BN / WAVES_N) + (lane & 15);
 
+    const int n_stages = (rsc + BK2 - 1) / BK2;
     short8 stage[CPT];
+    // prologue: chunk 0 -> buf0; chunk 1 -> regs
 #pragma unroll
     for (int t = 0; t < CPT; ++t) stage[t] = load_chunk(t, 0);
+#pragma unroll
+    for (int t = 0; t < CPT; ++t)
+        if (tid + t * CONV_THREADS < CHUNKS) lds_write(A_lds[0], t, stage[t]);
+    if (n_stages > 1) {
+#pragma unroll
+        for (int t = 0; t < CPT; ++t) stage[t] = load_chunk(t, BK2);
+    }
+    __syncthreads();
 
-    for (int kc = 0; kc < rsc; kc += CONV_BK) {
-        __syncthreads();  // previous chunk's LDS reads complete
+    for (int i = 0; i < n_stages; ++i) {
+        const uint16_t* buf = A_lds[i & 1];
+        // regs hold stage i+1: write them to the other buffer, then start
+        // loading stage i+2 (latency hides under this stage's MFMAs)
+        if (i + 1 < n_stages) {
+            uint16_t* nbuf = A_lds[(i + 1) & 1];
 #pragma unroll
-        for (int t = 0; t < CPT; ++t)
-            if (tid + t * CONV_THREADS < CHUNKS)
-                *reinterpret_cast<short8*>(
-                    &A_lds[st_row[t] * CONV_APITCH + ((tid + t * CONV_THREADS) & 3) * 8]) =
-                    stage[t];
-        __syncthreads();
-        if (kc + CONV_BK < rsc) {
+            for (int t = 0; t < CPT; ++t)
+                if (tid + t * CONV_THREADS < CHUNKS) lds_write(nbuf, t, stage[t]);
+            if (i + 2 < n_stages) {
 #pragma unroll
-            for (int t = 0; t < CPT; ++t) stage[t] = load_chunk(t, kc + CONV_BK);
+                for (int t = 0; t < CPT; ++t)
+                    stage[t] = load_chunk(t, (i + 2) * BK2);
+            }
         }
-        short8 b[NF];
+        const int kc = i * BK2;
 #pragma unroll
-        for (int nf = 0; nf < NF; ++nf)
-            b[nf] = *reinterpret_cast<const short8*>(
-                w + (int64_t)(b_col + nf * 16) * rsc + kc + a_koff);
-#pragma unroll
-        for (int mf = 0; mf < MF; ++mf) {
-            const short8 a = *reinterpret_cast<const short8*>(
-                &A_lds[(a_row + mf * 16) * CONV_APITCH + a_koff]);
+        for (int sub = 0; sub < 2; ++sub) {
+            if (kc + sub * CONV_BK >= rsc) break;
+            short8 b[NF];
 #pragma unroll
             for (int nf = 0; nf < NF; ++nf)
-                acc[mf][nf] = MFMA_BF16(a, b[nf], acc[mf][nf]);
+                b[nf] = *reinterpret_cast<const short8*>(
+                    w + (int64_t)(b_col + nf * 16) * rsc + kc + sub * CONV_BK + a_koff);
+#pragma unroll
+            for (int mf = 0; mf < MF; ++mf) {
+                const short8 a = *reinterpret_cast<const short8*>(
+                    &buf[(sub * BM + a_row + mf * 16) * CONV_APITCH + a_koff]);
+#pragma unroll
+                for (int nf = 0; nf < NF; ++nf)
+                    acc[mf][nf] = MFMA_BF16(a, b[nf], acc[mf][nf]);
+            }
         }
+        __syncthreads();
     }
 
     const int64_t out_row0 = m0 + wave_m * (BM / WAVES_M) + (lane >> 4) * 4;
@@ -169,16 +198,30 @@ extern "C" void launch_conv_fwd(const void* x, const void* w, void* y,
 __global__ void __launch_bounds__(256)
 k_conv_stem_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
                 uint16_t* __restrict__ y, ConvDims d) {
-    const int64_t total = (int64_t)d.N * d.Ho * d.Wo * d.K;
+    // weights staged once per block into LDS as fp32 [rsc][K] (K = 64);
+    // each thread computes one output pixel x 16 consecutive channels.
+    __shared__ float w_lds[32 * 64];
+    const int rsc = d.R * d.S * d.C;       // <= 32 (3x3x3 = 27)
+    for (int i = threadIdx.x; i < rsc * 64; i += blockDim.x) {
+        const int j = i >> 6;              // tap
+        const int k = i & 63;              // channel
+        w_lds[j * 64 + k] = bf16_to_f32(w[(int64_t)k * rsc + j]);
+    }
+    __syncthreads();
+
+    const int64_t M = (int64_t)d.N * d.Ho * d.Wo;
+    const int64_t total = M * 4;           // 4 channel-quads of 16
     const int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          idx < total; idx += stride) {
-        const int k = (int)(idx % d.K);
-        const int64_t m = idx / d.K;
+        const int kq = (int)(idx & 3) * 16;
+        const int64_t m = idx >> 2;
         const int wo = (int)(m % d.Wo);
         const int ho = (int)((m / d.Wo) % d.Ho);
         const int64_t n = m / ((int64_t)d.Ho * d.Wo);
-        float acc = 0.f;
+        float acc[16];
+#pragma unroll
+        for (int t = 0; t < 16; ++t) acc[t] = 0.f;
         for (int r = 0; r < d.R; ++r) {
             const int hi = ho * d.stride + r - d.pad;
             if (hi < 0 || hi >= d.H) continue;
@@ -186,38 +229,49 @@ k_conv_stem_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
                 const int wi = wo * d.stride + s - d.pad;
                 if (wi < 0 || wi >= d.W) continue;
                 const uint16_t* xp = x + ((n * d.H + hi) * d.W + wi) * d.C;
-                const uint16_t* wp = w + ((int64_t)k * d.R * d.S + r * d.S + s) * d.C;
-                for (int c = 0; c < d.C; ++c)
-                    acc = fmaf(bf16_to_f32(xp[c]), bf16_to_f32(wp[c]), acc);
+                const float* wp = w_lds + (r * d.S + s) * d.C * 64 + kq;
+                for (int c = 0; c < d.C; ++c) {
+                    const float xv = bf16_to_f32(xp[c]);
+#pragma unroll
+                    for (int t = 0; t < 16; ++t)
+                        acc[t] = fmaf(xv, wp[c * 64 + t], acc[t]);
+                }
             }
         }
-        y[idx] = f32_to_bf16(acc);
+        short8 out[2];
+#pragma unroll
+        for (int t = 0; t < 16; ++t)
+            ((uint16_t*)out)[t] = f32_to_bf16(acc[t]);
+        *reinterpret_cast<short8*>(y + m * d.K + kq) = out[0];
+        *reinterpret_cast<short8*>(y + m * d.K + kq + 8) = out[1];
     }
 }
 
 extern "C" void launch_conv_stem_fwd(const void* x, const void* w, void* y,
                                      ConvDims d, hipStream_t stream) {
-    const int64_t total = (int64_t)d.N * d.Ho * d.Wo * d.K;
+    const int64_t total = (int64_t)d.N * d.Ho * d.Wo * 4;
     k_conv_stem_fwd<<<ew_grid(total, 256, 1), 256, 0, stream>>>(
         (const uint16_t*)x, (const uint16_t*)w, (uint16_t*)y, d);
 }
 
 // Stem wgrad: dw[k][rsc] += sum_m dout[m][k] * im2col(x)[m][rsc]
-// grid (K, msplit): each block reduces its m-slice for one k, wave-reduces,
-// atomics into fp32 dw.  RSC <= 32 (3x3x3 = 27).
+// grid (msplit): each wave owns one m at a time (lane = k), the 27 input
+// taps are same-address broadcast loads; per-thread fp32 partials combine
+// through LDS, then one atomicAdd per (k, tap) per block.
 __global__ void __launch_bounds__(256)
 k_conv_stem_wgrad(const uint16_t* __restrict__ x,
                   const uint16_t* __restrict__ dout,
                   float* __restrict__ dw, ConvDims d, int m_per_block) {
-    const int k = blockIdx.x;
+    const int k = threadIdx.x & 63;
+    const int mlane = threadIdx.x >> 6;  // 0..3
     const int rsc = d.R * d.S * d.C;
     const int64_t M = (int64_t)d.N * d.Ho * d.Wo;
-    const int64_t ms = (int64_t)blockIdx.y * m_per_block;
+    const int64_t ms = (int64_t)blockIdx.x * m_per_block;
     const int64_t me = min(ms + (int64_t)m_per_block, M);
     float part[32];
 #pragma unroll
     for (int j = 0; j < 32; ++j) part[j] = 0.f;
-    for (int64_t m = ms + threadIdx.x; m < me; m += blockDim.x) {
+    for (int64_t m = ms + mlane; m < me; m += 4) {
         const float go = bf16_to_f32(dout[m * d.K + k]);
         const int wo = (int)(m % d.Wo);
         const int ho = (int)((m / d.Wo) % d.Ho);
@@ -235,20 +289,26 @@ k_conv_stem_wgrad(const uint16_t* __restrict__ x,
             }
         }
     }
-    for (int j = 0; j < rsc; ++j) {
-        float v = wave_sum(part[j]);
-        if ((threadIdx.x & 63) == 0 && v != 0.f)
-            atomicAdd(&dw[(int64_t)k * rsc + j], v);
+    __shared__ float red[4][64 * 32];
+#pragma unroll
+    for (int j = 0; j < 32; ++j) red[mlane][k * 32 + j] = part[j];
+    __syncthreads();
+    if (mlane == 0) {
+        for (int j = 0; j < rsc; ++j) {
+            const float v = red[0][k * 32 + j] + red[1][k * 32 + j] +
+                            red[2][k * 32 + j] + red[3][k * 32 + j];
+            if (v != 0.f) atomicAdd(&dw[(int64_t)k * rsc + j], v);
+        }
     }
 }
 
 extern "C" void launch_conv_stem_wgrad(const void* x, const void* dout, void* dw,
                                        ConvDims d, hipStream_t stream) {
     const int64_t M = (int64_t)d.N * d.Ho * d.Wo;
-    const int64_t cap = (M + 255) / 256;
-    const int msplit = (int)(cap < 32 ? (cap < 1 ? 1 : cap) : 32);
+    int64_t msplit = (M + 255) / 256;
+    if (msplit > 256) msplit = 256;
+    if (msplit < 1) msplit = 1;
     const int m_per_block = (int)((M + msplit - 1) / msplit);
-    dim3 grid((unsigned)d.K, (unsigned)msplit);
-    k_conv_stem_wgrad<<<grid, 256, 0, stream>>>(
+    k_conv_stem_wgrad<<<(unsigned)msplit, 256, 0, stream>>>(
         (const uint16_t*)x, (const uint16_t*)dout, (float*)dw, d, m_per_block);
 }

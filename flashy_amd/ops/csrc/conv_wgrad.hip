@@ -41,45 +41,51 @@ k_conv_wgrad(const uint16_t* __restrict__ x, const uint16_t* __restrict__ dout,
     const int frag_col = wave_j * 32 + (lane & 15);   // + jf*16  (rsc dim)
     const int moff = (lane >> 4) * 8;
 
-    for (int64_t mc = ms; mc < me; mc += CONV_BK) {
-        // ---- stage dout^T: thread t loads 8 contiguous k of one m-row ----
-        {
-            const int m_r = tid >> 3;          // 0..31
-            const int k8 = (tid & 7) * 8;      // 0..56
-            const int64_t m = mc + m_r;
-            short8 v = {};
-            if (m < M)
-                v = *reinterpret_cast<const short8*>(dout + m * d.K + k0 + k8);
-#pragma unroll
-            for (int j = 0; j < 8; ++j)
-                doutT[(k8 + j) * WG_MP + m_r] = ((const uint16_t*)&v)[j];
+    // register-prefetch pipeline: the next m-chunk's global loads are in
+    // flight while this chunk's MFMAs run.
+    const int m_r = tid >> 3;
+    const int k8 = (tid & 7) * 8;
+    auto load_dout = [&](int64_t mc) -> short8 {
+        short8 v = {};
+        const int64_t m = mc + m_r;
+        if (m < M)
+            v = *reinterpret_cast<const short8*>(dout + m * d.K + k0 + k8);
+        return v;
+    };
+    auto load_x = [&](int64_t mc) -> short8 {
+        short8 v = {};
+        const int64_t m = mc + m_r;
+        if (m < M) {
+            const int jj = j0 + k8;
+            const int r = jj / (d.S * d.C);
+            const int sc = jj - r * d.S * d.C;
+            const int s = sc / d.C;
+            const int c = sc - s * d.C;
+            const int wo = (int)(m % d.Wo);
+            const int ho = (int)((m / d.Wo) % d.Ho);
+            const int64_t n = m / ((int64_t)d.Ho * d.Wo);
+            const int hi = ho * d.stride + r - d.pad;
+            const int wi = wo * d.stride + s - d.pad;
+            if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W)
+                v = *reinterpret_cast<const short8*>(
+                    x + (((n * d.H + hi) * d.W + wi) * (int64_t)d.C + c));
         }
-        // ---- stage im2col^T --------------------------------------------
-        {
-            const int m_r = tid >> 3;
-            const int j8 = (tid & 7) * 8;
-            const int64_t m = mc + m_r;
-            short8 v = {};
-            if (m < M) {
-                const int jj = j0 + j8;
-                const int r = jj / (d.S * d.C);
-                const int sc = jj - r * d.S * d.C;
-                const int s = sc / d.C;
-                const int c = sc - s * d.C;
-                const int wo = (int)(m % d.Wo);
-                const int ho = (int)((m / d.Wo) % d.Ho);
-                const int64_t n = m / ((int64_t)d.Ho * d.Wo);
-                const int hi = ho * d.stride + r - d.pad;
-                const int wi = wo * d.stride + s - d.pad;
-                if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W)
-                    v = *reinterpret_cast<const short8*>(
-                        x + (((n * d.H + hi) * d.W + wi) * (int64_t)d.C + c));
-            }
+        return v;
+    };
+
+    short8 dv = load_dout(ms), xv = load_x(ms);
+    for (int64_t mc = ms; mc < me; mc += CONV_BK) {
+        __syncthreads();  // previous chunk's fragment reads complete
 #pragma unroll
-            for (int j = 0; j < 8; ++j)
-                xT[(j8 + j) * WG_MP + m_r] = ((const uint16_t*)&v)[j];
+        for (int j = 0; j < 8; ++j) {
+            doutT[(k8 + j) * WG_MP + m_r] = ((const uint16_t*)&dv)[j];
+            xT[(k8 + j) * WG_MP + m_r] = ((const uint16_t*)&xv)[j];
         }
         __syncthreads();
+        if (mc + CONV_BK < me) {
+            dv = load_dout(mc + CONV_BK);
+            xv = load_x(mc + CONV_BK);
+        }
 
         short8 a[2], b[2];
 #pragma unroll
@@ -94,8 +100,8 @@ k_conv_wgrad(const uint16_t* __restrict__ x, const uint16_t* __restrict__ dout,
 #pragma unroll
             for (int jf = 0; jf < 2; ++jf)
                 acc[kf][jf] = MFMA_BF16(a[kf], b[jf], acc[kf][jf]);
-        __syncthreads();
     }
+    __syncthreads();
 
     // ---- accumulate into fp32 dw (flat KRSC layout) ---------------------
     const int out_k0 = k0 + wave_k * 32 + (lane >> 4) * 4;
