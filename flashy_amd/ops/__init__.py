@@ -179,8 +179,8 @@ def bn_msplit(M: int, C: int) -> int:
     """Blocks along M for the BN reductions: ~256 blocks saturate the chip
     while keeping the partial-combine kernels cheap."""
     cols = max(1, C // 64)
-    msplit = max(1, min(256 // cols, 256))
-    return max(1, min(msplit, (M + 3) // 4))
+    msplit = max(1, min(512 // cols, 512))
+    return max(1, min(msplit, (M + 31) // 32))
 
 
 def bn_stats(x: torch.Tensor, partials: torch.Tensor, M: int, C: int,

@@ -2,63 +2,82 @@
 // NHWC BatchNorm (training) for gfx950, operating on [M, C] with
 // M = N*H*W, bf16 activations, fp32 stats/params.
 //
-// forward:  stats (per-channel sum/sumsq partials, one slot per block —
-//                  no atomics, no pre-zeroed buffers)
-//        -> finalize (tiny: combine partials, mean/invstd/scale/shift,
-//                     running stats)
-//        -> apply (y = relu(scale*x + shift [+ res]))  [fused add+ReLU]
-// backward: reduce (dz = dy * relu-mask; per-channel partial sums;
-//                   dz written out — it IS the residual gradient)
-//        -> grads (tiny: combine partials -> bsums; dgamma/dbeta += into
-//                  flat fp32 grads)
+// forward:  stats (per-channel sum/sumsq partials; short8 loads, one
+//                  partial slot per block — no atomics, no pre-zeroing)
+//        -> finalize (combine partials, mean/invstd/scale/shift, running)
+//        -> apply (y = relu(scale*x + shift [+ res]), float4 param loads)
+// backward: reduce (dz = dy * relu-mask; partial sums; dz written out —
+//                   it IS the residual gradient)
+//        -> grads (combine -> bsums; dgamma/dbeta += into flat fp32 grads)
 //        -> apply (dx = scale*(dz - (sum_dz + xhat*sum_dzxhat)/M))
+//
+// partials layout [2][C][msplit] so the combine kernels read each channel's
+// msplit partials CONTIGUOUSLY.
 //
 // Replaces the BatchNorm + ReLU + residual-add chains of the reference's
 // ResNet workload with NHWC-native fused kernels.
 
 #include "common.h"
 
-// partials layout: [msplit][2][C]  (sum, then sumsq/dzxhat)
+// ---------------------------------------------------------------------------
+// fwd 1: stats.  Block = 256 threads = 8 channel-octets x 32 m-lanes over a
+// 64-channel group; short8 loads (16 B/lane).  Grid (C/64, msplit).
+// ---------------------------------------------------------------------------
 
 __global__ void __launch_bounds__(256)
 k_bn_stats(const uint16_t* __restrict__ x, float* __restrict__ partials,
-           int64_t M, int C, int m_per_block) {
-    const int c = blockIdx.x * 64 + (threadIdx.x & 63);
-    const int mlane = threadIdx.x >> 6;  // 0..3
+           int64_t M, int C, int msplit, int m_per_block) {
+    const int c8 = (threadIdx.x & 7) * 8;           // channel octet in group
+    const int mlane = threadIdx.x >> 3;             // 0..31
+    const int cbase = blockIdx.x * 64;
     const int64_t m0 = (int64_t)blockIdx.y * m_per_block;
     const int64_t m1 = min(m0 + (int64_t)m_per_block, M);
-    float s = 0.f, s2 = 0.f;
-    for (int64_t m = m0 + mlane; m < m1; m += 4) {
-        const float v = bf16_to_f32(x[m * C + c]);
-        s += v;
-        s2 = fmaf(v, v, s2);
+    float s[8] = {}, s2[8] = {};
+    for (int64_t m = m0 + mlane; m < m1; m += 32) {
+        const short8 v8 = *reinterpret_cast<const short8*>(
+            x + m * C + cbase + c8);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            const float v = bf16_to_f32(((const uint16_t*)&v8)[j]);
+            s[j] += v;
+            s2[j] = fmaf(v, v, s2[j]);
+        }
     }
-    __shared__ float red[2][4][64];
-    red[0][mlane][threadIdx.x & 63] = s;
-    red[1][mlane][threadIdx.x & 63] = s2;
+    // fold 32 m-lanes: LDS [2][32][64]
+    __shared__ float red[2][32][64];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+        red[0][mlane][c8 + j] = s[j];
+        red[1][mlane][c8 + j] = s2[j];
+    }
     __syncthreads();
-    if (mlane == 0) {
-        s = red[0][0][threadIdx.x] + red[0][1][threadIdx.x] +
-            red[0][2][threadIdx.x] + red[0][3][threadIdx.x];
-        s2 = red[1][0][threadIdx.x] + red[1][1][threadIdx.x] +
-             red[1][2][threadIdx.x] + red[1][3][threadIdx.x];
-        float* slot = partials + (int64_t)blockIdx.y * 2 * C;
-        slot[c] = s;
-        slot[C + c] = s2;
+    // 256 threads: tid<64 handles sum, tid in [64,128) handles sumsq
+    const int c = threadIdx.x & 63;
+    const int which = threadIdx.x >> 6;
+    if (which < 2) {
+        float acc = 0.f;
+#pragma unroll 8
+        for (int i = 0; i < 32; ++i) acc += red[which][i][c];
+        partials[((int64_t)which * C + cbase + c) * msplit + blockIdx.y] = acc;
     }
 }
 
-// Parallel combine of [msplit][2][C] partials: 256 threads = 64 channels x
-// 4 split-lanes, many loads in flight, LDS reduce.  Grid C/64.
+// ---------------------------------------------------------------------------
+// combine helper: [2][C][msplit] -> (s, s2) per channel.  Block = 256
+// threads = 64 channels x 4 split-lanes; contiguous per-channel reads.
+// ---------------------------------------------------------------------------
+
 __device__ __forceinline__ void combine_partials(
         const float* __restrict__ partials, int msplit, int C,
         float* s_out, float* s2_out) {
     const int c = blockIdx.x * 64 + (threadIdx.x & 63);
     const int slane = threadIdx.x >> 6;  // 0..3
+    const float* row0 = partials + (int64_t)c * msplit;
+    const float* row1 = partials + ((int64_t)C + c) * msplit;
     float s = 0.f, s2 = 0.f;
     for (int i = slane; i < msplit; i += 4) {
-        s += partials[(int64_t)i * 2 * C + c];
-        s2 += partials[(int64_t)i * 2 * C + C + c];
+        s += row0[i];
+        s2 += row1[i];
     }
     __shared__ float red[2][4][64];
     red[0][slane][threadIdx.x & 63] = s;
@@ -99,6 +118,10 @@ k_bn_finalize(const float* __restrict__ partials, int msplit,
     }
 }
 
+// ---------------------------------------------------------------------------
+// fwd 3: y = [relu](scale*x + shift [+ res]).  short8 data, float4 params.
+// ---------------------------------------------------------------------------
+
 template <bool RELU, bool RES>
 __global__ void __launch_bounds__(256)
 k_bn_apply(const uint16_t* __restrict__ x, const uint16_t* __restrict__ res,
@@ -106,11 +129,13 @@ k_bn_apply(const uint16_t* __restrict__ x, const uint16_t* __restrict__ res,
            int64_t M, int C) {
     const int64_t total8 = M * C / 8;
     const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    const float* scale = work + 2 * (int64_t)C;
-    const float* shift = work + 3 * (int64_t)C;
+    const float4* scale4 = reinterpret_cast<const float4*>(work + 2 * (int64_t)C);
+    const float4* shift4 = reinterpret_cast<const float4*>(work + 3 * (int64_t)C);
     for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total8;
          i += stride) {
-        const int c0 = (int)((i * 8) % C);
+        const int co = (int)((i * 8) % C) / 4;
+        const float4 sc[2] = {scale4[co], scale4[co + 1]};
+        const float4 sh[2] = {shift4[co], shift4[co + 1]};
         short8 xv = *reinterpret_cast<const short8*>(x + i * 8);
         short8 rv = {};
         if (RES) rv = *reinterpret_cast<const short8*>(res + i * 8);
@@ -118,7 +143,7 @@ k_bn_apply(const uint16_t* __restrict__ x, const uint16_t* __restrict__ res,
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
             float v = fmaf(bf16_to_f32(((const uint16_t*)&xv)[j]),
-                           scale[c0 + j], shift[c0 + j]);
+                           ((const float*)sc)[j], ((const float*)sh)[j]);
             if (RES) v += bf16_to_f32(((const uint16_t*)&rv)[j]);
             if (RELU) v = fmaxf(v, 0.f);
             ((uint16_t*)&out)[j] = f32_to_bf16(v);
@@ -127,39 +152,61 @@ k_bn_apply(const uint16_t* __restrict__ x, const uint16_t* __restrict__ res,
     }
 }
 
+// ---------------------------------------------------------------------------
+// bwd 1: dz = dy * (y > 0) [if relu]; per-channel partial sums of dz and
+// dz*xhat.  Same geometry as stats.
+// ---------------------------------------------------------------------------
+
 template <bool RELU>
 __global__ void __launch_bounds__(256)
 k_bn_bwd_reduce(const uint16_t* __restrict__ dy, const uint16_t* __restrict__ y,
                 const uint16_t* __restrict__ x, const float* __restrict__ work,
                 uint16_t* __restrict__ dz_out, float* __restrict__ partials,
-                int64_t M, int C, int m_per_block) {
-    const int c = blockIdx.x * 64 + (threadIdx.x & 63);
-    const int mlane = threadIdx.x >> 6;
+                int64_t M, int C, int msplit, int m_per_block) {
+    const int c8 = (threadIdx.x & 7) * 8;
+    const int mlane = threadIdx.x >> 3;
+    const int cbase = blockIdx.x * 64;
     const int64_t m0 = (int64_t)blockIdx.y * m_per_block;
     const int64_t m1 = min(m0 + (int64_t)m_per_block, M);
-    const float mean = work[c];
-    const float invstd = work[C + c];
-    float s = 0.f, sx = 0.f;
-    for (int64_t m = m0 + mlane; m < m1; m += 4) {
-        const int64_t i = m * C + c;
-        float g = bf16_to_f32(dy[i]);
-        if (RELU && bf16_to_f32(y[i]) <= 0.f) g = 0.f;
-        dz_out[i] = f32_to_bf16(g);
-        s += g;
-        sx = fmaf(g, (bf16_to_f32(x[i]) - mean) * invstd, sx);
+    float mean[8], invstd[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+        mean[j] = work[cbase + c8 + j];
+        invstd[j] = work[C + cbase + c8 + j];
     }
-    __shared__ float red[2][4][64];
-    red[0][mlane][threadIdx.x & 63] = s;
-    red[1][mlane][threadIdx.x & 63] = sx;
+    float s[8] = {}, sx[8] = {};
+    for (int64_t m = m0 + mlane; m < m1; m += 32) {
+        const int64_t off = m * C + cbase + c8;
+        const short8 g8 = *reinterpret_cast<const short8*>(dy + off);
+        const short8 x8 = *reinterpret_cast<const short8*>(x + off);
+        short8 y8 = {};
+        if (RELU) y8 = *reinterpret_cast<const short8*>(y + off);
+        short8 dz8;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            float g = bf16_to_f32(((const uint16_t*)&g8)[j]);
+            if (RELU && bf16_to_f32(((const uint16_t*)&y8)[j]) <= 0.f) g = 0.f;
+            ((uint16_t*)&dz8)[j] = f32_to_bf16(g);
+            s[j] += g;
+            const float xh = (bf16_to_f32(((const uint16_t*)&x8)[j]) - mean[j]) * invstd[j];
+            sx[j] = fmaf(g, xh, sx[j]);
+        }
+        *reinterpret_cast<short8*>(dz_out + off) = dz8;
+    }
+    __shared__ float red[2][32][64];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+        red[0][mlane][c8 + j] = s[j];
+        red[1][mlane][c8 + j] = sx[j];
+    }
     __syncthreads();
-    if (mlane == 0) {
-        s = red[0][0][threadIdx.x] + red[0][1][threadIdx.x] +
-            red[0][2][threadIdx.x] + red[0][3][threadIdx.x];
-        sx = red[1][0][threadIdx.x] + red[1][1][threadIdx.x] +
-             red[1][2][threadIdx.x] + red[1][3][threadIdx.x];
-        float* slot = partials + (int64_t)blockIdx.y * 2 * C;
-        slot[c] = s;
-        slot[C + c] = sx;
+    const int c = threadIdx.x & 63;
+    const int which = threadIdx.x >> 6;
+    if (which < 2) {
+        float acc = 0.f;
+#pragma unroll 8
+        for (int i = 0; i < 32; ++i) acc += red[which][i][c];
+        partials[((int64_t)which * C + cbase + c) * msplit + blockIdx.y] = acc;
     }
 }
 
@@ -179,6 +226,10 @@ k_bn_bwd_grads(const float* __restrict__ partials, int msplit,
     dgamma[c] += sx;
 }
 
+// ---------------------------------------------------------------------------
+// bwd 3: dx = scale * (dz - (sum_dz + xhat * sum_dz_xhat) / M)
+// ---------------------------------------------------------------------------
+
 __global__ void __launch_bounds__(256)
 k_bn_bwd_apply(const uint16_t* __restrict__ dz, const uint16_t* __restrict__ x,
                const float* __restrict__ work, const float* __restrict__ bsums,
@@ -186,21 +237,29 @@ k_bn_bwd_apply(const uint16_t* __restrict__ dz, const uint16_t* __restrict__ x,
     const int64_t total8 = M * C / 8;
     const int64_t stride = (int64_t)gridDim.x * blockDim.x;
     const float invM = 1.f / (float)M;
+    const float4* mean4 = reinterpret_cast<const float4*>(work);
+    const float4* invstd4 = reinterpret_cast<const float4*>(work + C);
+    const float4* scale4 = reinterpret_cast<const float4*>(work + 2 * (int64_t)C);
+    const float4* b04 = reinterpret_cast<const float4*>(bsums);
+    const float4* b14 = reinterpret_cast<const float4*>(bsums + C);
     for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total8;
          i += stride) {
-        const int c0 = (int)((i * 8) % C);
+        const int co = (int)((i * 8) % C) / 4;
+        const float4 mn[2] = {mean4[co], mean4[co + 1]};
+        const float4 is[2] = {invstd4[co], invstd4[co + 1]};
+        const float4 sc[2] = {scale4[co], scale4[co + 1]};
+        const float4 b0[2] = {b04[co], b04[co + 1]};
+        const float4 b1[2] = {b14[co], b14[co + 1]};
         short8 gz = *reinterpret_cast<const short8*>(dz + i * 8);
         short8 xv = *reinterpret_cast<const short8*>(x + i * 8);
         short8 out;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
-            const int c = c0 + j;
-            const float mean = work[c];
-            const float invstd = work[C + c];
-            const float scale = work[2 * C + c];
-            const float xhat = (bf16_to_f32(((const uint16_t*)&xv)[j]) - mean) * invstd;
+            const float xhat = (bf16_to_f32(((const uint16_t*)&xv)[j]) -
+                                ((const float*)mn)[j]) * ((const float*)is)[j];
             const float g = bf16_to_f32(((const uint16_t*)&gz)[j]);
-            const float v = scale * (g - (bsums[c] + xhat * bsums[C + c]) * invM);
+            const float v = ((const float*)sc)[j] *
+                (g - (((const float*)b0)[j] + xhat * ((const float*)b1)[j]) * invM);
             ((uint16_t*)&out)[j] = f32_to_bf16(v);
         }
         *reinterpret_cast<short8*>(dx + i * 8) = out;
@@ -216,7 +275,7 @@ extern "C" void launch_bn_stats(const void* x, void* partials, int64_t M, int C,
     const int mpb = (int)((M + msplit - 1) / msplit);
     dim3 grid((unsigned)(C / 64), (unsigned)msplit);
     k_bn_stats<<<grid, 256, 0, stream>>>((const uint16_t*)x, (float*)partials,
-                                         M, C, mpb);
+                                         M, C, msplit, mpb);
 }
 
 extern "C" void launch_bn_finalize(const void* partials, int msplit,
@@ -261,11 +320,13 @@ extern "C" void launch_bn_bwd_reduce(const void* dy, const void* y,
     if (relu)
         k_bn_bwd_reduce<true><<<grid, 256, 0, stream>>>(
             (const uint16_t*)dy, (const uint16_t*)y, (const uint16_t*)x,
-            (const float*)work, (uint16_t*)dz_out, (float*)partials, M, C, mpb);
+            (const float*)work, (uint16_t*)dz_out, (float*)partials, M, C,
+            msplit, mpb);
     else
         k_bn_bwd_reduce<false><<<grid, 256, 0, stream>>>(
             (const uint16_t*)dy, (const uint16_t*)y, (const uint16_t*)x,
-            (const float*)work, (uint16_t*)dz_out, (float*)partials, M, C, mpb);
+            (const float*)work, (uint16_t*)dz_out, (float*)partials, M, C,
+            msplit, mpb);
 }
 
 extern "C" void launch_bn_bwd_grads(const void* partials, int msplit,

@@ -255,23 +255,25 @@ extern "C" void launch_conv_stem_fwd(const void* x, const void* w, void* y,
 }
 
 // Stem wgrad: dw[k][rsc] += sum_m dout[m][k] * im2col(x)[m][rsc]
-// grid (msplit): each wave owns one m at a time (lane = k), the 27 input
-// taps are same-address broadcast loads; per-thread fp32 partials combine
-// through LDS, then one atomicAdd per (k, tap) per block.
+// grid (K/8 octets, msplit).  Thread = one k of its octet x one of 32
+// m-lanes: dout loads are 16B-coalesced across the octet, the 27 input taps
+// are same-address broadcasts within the octet; LDS fold over m-lanes, then
+// 8x27 atomicAdds per block.  RSC <= 32 (3x3x3 = 27).
 __global__ void __launch_bounds__(256)
 k_conv_stem_wgrad(const uint16_t* __restrict__ x,
                   const uint16_t* __restrict__ dout,
                   float* __restrict__ dw, ConvDims d, int m_per_block) {
-    const int k = threadIdx.x & 63;
-    const int mlane = threadIdx.x >> 6;  // 0..3
+    const int kl = threadIdx.x & 7;          // k within octet
+    const int k = blockIdx.x * 8 + kl;
+    const int mlane = threadIdx.x >> 3;      // 0..31
     const int rsc = d.R * d.S * d.C;
     const int64_t M = (int64_t)d.N * d.Ho * d.Wo;
-    const int64_t ms = (int64_t)blockIdx.x * m_per_block;
+    const int64_t ms = (int64_t)blockIdx.y * m_per_block;
     const int64_t me = min(ms + (int64_t)m_per_block, M);
     float part[32];
 #pragma unroll
     for (int j = 0; j < 32; ++j) part[j] = 0.f;
-    for (int64_t m = ms + mlane; m < me; m += 4) {
+    for (int64_t m = ms + mlane; m < me; m += 32) {
         const float go = bf16_to_f32(dout[m * d.K + k]);
         const int wo = (int)(m % d.Wo);
         const int ho = (int)((m / d.Wo) % d.Ho);
@@ -289,26 +291,31 @@ k_conv_stem_wgrad(const uint16_t* __restrict__ x,
             }
         }
     }
-    __shared__ float red[4][64 * 32];
+    // fold the 32 m-lanes per k: LDS [8 k][32 lanes spread over 27 taps]
+    __shared__ float red[8][32][32];
 #pragma unroll
-    for (int j = 0; j < 32; ++j) red[mlane][k * 32 + j] = part[j];
+    for (int j = 0; j < 32; ++j) red[kl][mlane][j] = part[j];
     __syncthreads();
-    if (mlane == 0) {
-        for (int j = 0; j < rsc; ++j) {
-            const float v = red[0][k * 32 + j] + red[1][k * 32 + j] +
-                            red[2][k * 32 + j] + red[3][k * 32 + j];
-            if (v != 0.f) atomicAdd(&dw[(int64_t)k * rsc + j], v);
-        }
+    // 256 threads: thread -> (k-octet slot, tap); fold 32 lanes
+    const int kk = threadIdx.x >> 5;         // 0..7
+    const int j0 = threadIdx.x & 31;         // tap (first 27 valid)
+    if (j0 < rsc) {
+        float acc = 0.f;
+#pragma unroll 8
+        for (int i = 0; i < 32; ++i) acc += red[kk][i][j0];
+        if (acc != 0.f)
+            atomicAdd(&dw[(int64_t)(blockIdx.x * 8 + kk) * rsc + j0], acc);
     }
 }
 
 extern "C" void launch_conv_stem_wgrad(const void* x, const void* dout, void* dw,
                                        ConvDims d, hipStream_t stream) {
     const int64_t M = (int64_t)d.N * d.Ho * d.Wo;
-    int64_t msplit = (M + 255) / 256;
-    if (msplit > 256) msplit = 256;
+    int64_t msplit = (M + 511) / 512;
+    if (msplit > 128) msplit = 128;
     if (msplit < 1) msplit = 1;
     const int m_per_block = (int)((M + msplit - 1) / msplit);
-    k_conv_stem_wgrad<<<(unsigned)msplit, 256, 0, stream>>>(
+    dim3 grid((unsigned)(d.K / 8), (unsigned)msplit);
+    k_conv_stem_wgrad<<<grid, 256, 0, stream>>>(
         (const uint16_t*)x, (const uint16_t*)dout, (float*)dw, d, m_per_block);
 }
