@@ -160,7 +160,7 @@ def conv_wgrad(x: torch.Tensor, dout: torch.Tensor, dw: torch.Tensor,
         assert d.K % 64 == 0 and rsc % 64 == 0, d
         if n_splits is None:
             tiles = (d.K // 64) * (rsc // 64)
-            n_splits = max(1, min(512 // tiles if tiles else 1, 64))
+            n_splits = max(1, min(1024 // tiles if tiles else 1, 128))
             M = d.N * d.Ho * d.Wo
             n_splits = max(1, min(n_splits, M // 32 or 1))
         ext.conv_wgrad(x.data_ptr(), dout.data_ptr(), dw.data_ptr(), *d,

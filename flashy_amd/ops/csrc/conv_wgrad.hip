@@ -30,9 +30,9 @@ k_conv_wgrad(const uint16_t* __restrict__ x, const uint16_t* __restrict__ dout,
     const int64_t ms = (int64_t)blockIdx.z * m_per_split;
     const int64_t me = min(ms + (int64_t)m_per_split, M);
 
-    // transposed chunks: [64 channels][32 m]
-    __shared__ uint16_t doutT[64 * WG_MP];
-    __shared__ uint16_t xT[64 * WG_MP];
+    // transposed chunks: [64 channels][32 m], double-buffered
+    __shared__ uint16_t doutT[2][64 * WG_MP];
+    __shared__ uint16_t xT[2][64 * WG_MP];
 
     // wave computes 32(K) x 32(rsc): 2x2 fragments of 16x16
     floatx4 acc[2][2] = {};
@@ -73,35 +73,50 @@ k_conv_wgrad(const uint16_t* __restrict__ x, const uint16_t* __restrict__ dout,
         return v;
     };
 
+    // schedule per chunk i: write regs(i+1)->buf^1, issue loads(i+2),
+    // MFMA over buf, ONE barrier.  Global latency hides under ~2 chunks.
+    const int64_t n_chunks = (me - ms + CONV_BK - 1) / CONV_BK;
     short8 dv = load_dout(ms), xv = load_x(ms);
-    for (int64_t mc = ms; mc < me; mc += CONV_BK) {
-        __syncthreads();  // previous chunk's fragment reads complete
+    {   // prologue: chunk 0 -> buf 0; chunk 1 -> regs
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
-            doutT[(k8 + j) * WG_MP + m_r] = ((const uint16_t*)&dv)[j];
-            xT[(k8 + j) * WG_MP + m_r] = ((const uint16_t*)&xv)[j];
+            doutT[0][(k8 + j) * WG_MP + m_r] = ((const uint16_t*)&dv)[j];
+            xT[0][(k8 + j) * WG_MP + m_r] = ((const uint16_t*)&xv)[j];
+        }
+        if (n_chunks > 1) {
+            dv = load_dout(ms + CONV_BK);
+            xv = load_x(ms + CONV_BK);
         }
         __syncthreads();
-        if (mc + CONV_BK < me) {
-            dv = load_dout(mc + CONV_BK);
-            xv = load_x(mc + CONV_BK);
+    }
+    for (int64_t i = 0; i < n_chunks; ++i) {
+        const int cur = (int)(i & 1);
+        if (i + 1 < n_chunks) {
+#pragma unroll
+            for (int j = 0; j < 8; ++j) {
+                doutT[cur ^ 1][(k8 + j) * WG_MP + m_r] = ((const uint16_t*)&dv)[j];
+                xT[cur ^ 1][(k8 + j) * WG_MP + m_r] = ((const uint16_t*)&xv)[j];
+            }
+            if (i + 2 < n_chunks) {
+                dv = load_dout(ms + (i + 2) * CONV_BK);
+                xv = load_x(ms + (i + 2) * CONV_BK);
+            }
         }
-
         short8 a[2], b[2];
 #pragma unroll
         for (int f = 0; f < 2; ++f) {
             a[f] = *reinterpret_cast<const short8*>(
-                &doutT[(frag_row + f * 16) * WG_MP + moff]);
+                &doutT[cur][(frag_row + f * 16) * WG_MP + moff]);
             b[f] = *reinterpret_cast<const short8*>(
-                &xT[(frag_col + f * 16) * WG_MP + moff]);
+                &xT[cur][(frag_col + f * 16) * WG_MP + moff]);
         }
 #pragma unroll
         for (int kf = 0; kf < 2; ++kf)
 #pragma unroll
             for (int jf = 0; jf < 2; ++jf)
                 acc[kf][jf] = MFMA_BF16(a[kf], b[jf], acc[kf][jf]);
+        __syncthreads();
     }
-    __syncthreads();
 
     // ---- accumulate into fp32 dw (flat KRSC layout) ---------------------
     const int out_k0 = k0 + wave_k * 32 + (lane >> 4) * 4;
