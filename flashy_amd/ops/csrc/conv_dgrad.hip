@@ -8,8 +8,9 @@
 //   B[(r,s,k)][c] = w[k][r][s][c] read from the RSCK transposed copy
 //   (k_weight_transpose below) so each MFMA B-fragment lane reads 8
 //   consecutive k — contiguous 16 B.
-// Same pipelined BMx64x32 structure as the forward kernel.
-// Requires: K % 32 == 0, C % 64 == 0.
+// Same pipelined 64-deep double-buffered single-barrier structure as the
+// forward kernel, with incremental (r,s,k) tap walk and B prefetch.
+// Requires: K % 64 == 0, C % 64 == 0.
 
 #include "conv_common.h"
 
@@ -35,15 +36,14 @@ k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_r
     const int64_t m0 = (int64_t)blockIdx.x * BM;
     const int col0 = blockIdx.y * CONV_BN;
 
-    __shared__ uint16_t A_lds[2][BM * 2 * CONV_APITCH];
+    __shared__ uint16_t A_lds[2][2 * BM * CONV_APITCH];
 
-    int st_row[CPT], st_hi[CPT], st_wi[CPT];
+    int st_hi[CPT], st_wi[CPT], st_r[CPT], st_s[CPT], st_k[CPT];
     int64_t st_n[CPT];
 #pragma unroll
     for (int t = 0; t < CPT; ++t) {
         const int chunk = tid + t * CONV_THREADS;
         const int row = chunk >> 3;
-        st_row[t] = row;
         const int64_t m = m0 + row;
         if (chunk < CHUNKS && m < M) {
             const int hw = d.H * d.W;
@@ -54,89 +54,112 @@ k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_r
         } else {
             st_n[t] = -1;
         }
+        st_r[t] = 0;
+        st_s[t] = 0;
+        st_k[t] = (chunk & 7) * 8;   // < 64 <= K
     }
 
-    auto load_chunk = [&](int t, int kc) -> short8 {
-        short8 v = {};
-        const int chunk = tid + t * CONV_THREADS;
-        const int kk = kc + (chunk & 7) * 8;
-        if (st_n[t] >= 0 && kk < rsk) {
-            const int r = kk / (d.S * d.K);
-            const int sk = kk - r * d.S * d.K;
-            const int s = sk / d.K;
-            const int k = sk - s * d.K;
-            const int hnum = st_hi[t] - r;
-            const int wnum = st_wi[t] - s;
-            const int ho = hnum / d.stride;
-            const int wo = wnum / d.stride;
-            if (hnum >= 0 && wnum >= 0 && ho * d.stride == hnum &&
-                wo * d.stride == wnum && ho < d.Ho && wo < d.Wo)
-                v = *reinterpret_cast<const short8*>(
-                    dout + (((st_n[t] * d.Ho + ho) * d.Wo + wo) * (int64_t)d.K + k));
+    auto load_stage = [&](short8* dst) {
+#pragma unroll
+        for (int t = 0; t < CPT; ++t) {
+            short8 v = {};
+            if (st_n[t] >= 0 && st_r[t] < d.R) {
+                const int hnum = st_hi[t] - st_r[t];  // = ho * stride
+                const int wnum = st_wi[t] - st_s[t];
+                const int ho = hnum / d.stride;
+                const int wo = wnum / d.stride;
+                if (hnum >= 0 && wnum >= 0 && ho * d.stride == hnum &&
+                    wo * d.stride == wnum && ho < d.Ho && wo < d.Wo)
+                    v = *reinterpret_cast<const short8*>(
+                        dout + (((st_n[t] * d.Ho + ho) * d.Wo + wo) * (int64_t)d.K +
+                                st_k[t]));
+            }
+            dst[t] = v;
+            int k = st_k[t] + BK2;
+            while (k >= d.K) {
+                k -= d.K;
+                if (++st_s[t] == d.S) { st_s[t] = 0; ++st_r[t]; }
+            }
+            st_k[t] = k;
         }
-        return v;
     };
-    auto lds_write = [&](uint16_t* buf, int t, short8 v) {
-        const int chunk = tid + t * CONV_THREADS;
-        const int koff = (chunk & 7) * 8;
-        const int sub = koff >> 5;
-        *reinterpret_cast<short8*>(
-            &buf[(sub * BM + st_row[t]) * CONV_APITCH + (koff & 31)]) = v;
+    auto lds_write = [&](uint16_t* buf, const short8* src) {
+#pragma unroll
+        for (int t = 0; t < CPT; ++t) {
+            const int chunk = tid + t * CONV_THREADS;
+            if (chunk < CHUNKS) {
+                const int row = chunk >> 3;
+                const int koff = (chunk & 7) * 8;
+                const int sub = koff >> 5;
+                *reinterpret_cast<short8*>(
+                    &buf[(sub * BM + row) * CONV_APITCH + (koff & 31)]) = src[t];
+            }
+        }
     };
 
-    floatx4 acc[MF][NF] = {};
     const int a_row = wave_m * (BM / WAVES_M) + (lane & 15);
     const int a_koff = (lane >> 4) * 8;
     const int b_col = col0 + wave_n * (CONV_BN / WAVES_N) + (lane & 15);
 
+    // B tap state: (r, s, k) of (stage base + a_koff); (r,s) constant across
+    // each 32-subchunk since K % 32 == 0, so track per sub ∈ {0,1}.
+    int b_r[2], b_s[2], b_k[2];
+#pragma unroll
+    for (int sub = 0; sub < 2; ++sub) {
+        b_r[sub] = 0;
+        b_s[sub] = 0;
+        b_k[sub] = sub * CONV_BK + a_koff;
+        // K >= 64 so initial k < K always
+    }
+    auto load_b = [&](short8 (*dst)[NF]) {
+#pragma unroll
+        for (int sub = 0; sub < 2; ++sub) {
+#pragma unroll
+            for (int nf = 0; nf < NF; ++nf)
+                dst[sub][nf] = (b_r[sub] < d.R)
+                    ? *reinterpret_cast<const short8*>(
+                          w_rsck + ((int64_t)(b_r[sub] * d.S + b_s[sub]) * d.C +
+                                    b_col + nf * 16) * d.K + b_k[sub])
+                    : short8{};
+            int k = b_k[sub] + BK2;
+            while (k >= d.K) {
+                k -= d.K;
+                if (++b_s[sub] == d.S) { b_s[sub] = 0; ++b_r[sub]; }
+            }
+            b_k[sub] = k;
+        }
+    };
+
+    floatx4 acc[MF][NF] = {};
     const int n_stages = (rsk + BK2 - 1) / BK2;
     short8 stage[CPT];
-#pragma unroll
-    for (int t = 0; t < CPT; ++t) stage[t] = load_chunk(t, 0);
-#pragma unroll
-    for (int t = 0; t < CPT; ++t)
-        if (tid + t * CONV_THREADS < CHUNKS) lds_write(A_lds[0], t, stage[t]);
-    if (n_stages > 1) {
-#pragma unroll
-        for (int t = 0; t < CPT; ++t) stage[t] = load_chunk(t, BK2);
-    }
+    short8 breg[2][2][NF];
+
+    load_stage(stage);
+    lds_write(A_lds[0], stage);
+    if (n_stages > 1) load_stage(stage);
+    load_b(breg[0]);
     __syncthreads();
 
     for (int i = 0; i < n_stages; ++i) {
         const uint16_t* buf = A_lds[i & 1];
         if (i + 1 < n_stages) {
-            uint16_t* nbuf = A_lds[(i + 1) & 1];
-#pragma unroll
-            for (int t = 0; t < CPT; ++t)
-                if (tid + t * CONV_THREADS < CHUNKS) lds_write(nbuf, t, stage[t]);
-            if (i + 2 < n_stages) {
-#pragma unroll
-                for (int t = 0; t < CPT; ++t)
-                    stage[t] = load_chunk(t, (i + 2) * BK2);
-            }
+            lds_write(A_lds[(i + 1) & 1], stage);
+            if (i + 2 < n_stages) load_stage(stage);
+            load_b(breg[(i + 1) & 1]);
         }
+        const short8(*b)[NF] = breg[i & 1];
         const int kc = i * BK2;
 #pragma unroll
         for (int sub = 0; sub < 2; ++sub) {
-            const int kbase = kc + sub * CONV_BK;
-            if (kbase >= rsk) break;
-            // (r,s) constant across the 32-chunk since K % 32 == 0
-            const int r = (kbase + a_koff) / (d.S * d.K);
-            const int sk = (kbase + a_koff) - r * d.S * d.K;
-            const int sidx = sk / d.K;
-            const int k = sk - sidx * d.K;
-            short8 b[NF];
-#pragma unroll
-            for (int nf = 0; nf < NF; ++nf)
-                b[nf] = *reinterpret_cast<const short8*>(
-                    w_rsck + ((int64_t)(r * d.S + sidx) * d.C + b_col + nf * 16) * d.K + k);
+            if (kc + sub * CONV_BK >= rsk) break;
 #pragma unroll
             for (int mf = 0; mf < MF; ++mf) {
                 const short8 a = *reinterpret_cast<const short8*>(
                     &buf[(sub * BM + a_row + mf * 16) * CONV_APITCH + a_koff]);
 #pragma unroll
                 for (int nf = 0; nf < NF; ++nf)
-                    acc[mf][nf] = MFMA_BF16(a, b[nf], acc[mf][nf]);
+                    acc[mf][nf] = MFMA_BF16(a, b[sub][nf], acc[mf][nf]);
             }
         }
         __syncthreads();

@@ -7,13 +7,14 @@
 //   B = weights in [K][R][S][C] layout (= B^T: each MFMA B-fragment lane
 //       reads 8 consecutive rsc for its column K — 16 B contiguous, L2).
 //
-// Pipelined staging (guide §6 G15, T14 shape): the next A chunk's global
-// loads are issued right after the LDS write barrier, so HBM/L2 latency
-// hides under the MFMA cluster of the current chunk.
+// Schedule per 64-deep stage (two MFMA-K subchunks), double-buffered LDS,
+// ONE barrier per stage: write regs(stage i+1) -> other buffer, issue
+// global loads for stage i+2 and the B fragments of stage i+1, then the
+// MFMA cluster over stage i.  Tap indices (r,s,c) advance INCREMENTALLY
+// (+64 with carry) — no integer divisions in the steady state.
 //
-// Tile template: BM in {128, 64, 32} x BN=64 x BK=32, 4 waves.  Smaller BM
-// keeps the deep ResNet layers (M = N*Ho*Wo as small as 1024) above ~256
-// workgroups so the 256-CU chip stays filled.
+// Tile template: BM in {128, 64, 32} x BN=64, 4 waves.  Smaller BM keeps
+// the deep ResNet layers (M as small as 1024) above ~208 workgroups.
 // Requires: C % 8 == 0, K % 64 == 0, rsc % 32 == 0 (ResNet bodies; the
 // C=3 stem has its own direct kernels below).
 
@@ -27,8 +28,8 @@ k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
     constexpr int WAVES_N = 4 / WAVES_M;
     constexpr int MF = BM / WAVES_M / 16;      // m fragments per wave
     constexpr int NF = CONV_BN / WAVES_N / 16; // n fragments per wave
-    constexpr int BK2 = 2 * CONV_BK;           // 64-deep stage (2 MFMA-K)
-    constexpr int CHUNKS = BM * (BK2 / 8);     // 16B staging chunks per stage
+    constexpr int BK2 = 2 * CONV_BK;           // 64-deep stage
+    constexpr int CHUNKS = BM * (BK2 / 8);
     constexpr int CPT = (CHUNKS + CONV_THREADS - 1) / CONV_THREADS;
 
     const int rsc = d.R * d.S * d.C;
@@ -41,17 +42,16 @@ k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
     const int64_t m0 = (int64_t)blockIdx.x * BM;
     const int col0 = blockIdx.y * CONV_BN;
 
-    // two stage buffers, each [2 kk-subchunks][BM][APITCH] (sub-major keeps
-    // the row pitch at 48 elements = conflict-free ds_read_b128 groups)
-    __shared__ uint16_t A_lds[2][BM * 2 * CONV_APITCH];
+    // [sub][row][APITCH] per buffer: row pitch 48 = conflict-free b128 groups
+    __shared__ uint16_t A_lds[2][2 * BM * CONV_APITCH];
 
-    int st_row[CPT], st_hi[CPT], st_wi[CPT];
+    // --- staging state: row geometry + incremental (r,s,c) tap walk ------
+    int st_hi[CPT], st_wi[CPT], st_r[CPT], st_s[CPT], st_c[CPT];
     int64_t st_n[CPT];
 #pragma unroll
     for (int t = 0; t < CPT; ++t) {
         const int chunk = tid + t * CONV_THREADS;
-        const int row = chunk >> 3;            // 8 chunks per row (64 kk)
-        st_row[t] = row;
+        const int row = chunk >> 3;
         const int64_t m = m0 + row;
         if (chunk < CHUNKS && m < M) {
             const int hw = d.Ho * d.Wo;
@@ -62,85 +62,94 @@ k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
         } else {
             st_n[t] = -1;
         }
+        const int kk = (chunk & 7) * 8;   // tap offset of stage 0
+        st_r[t] = kk / (d.S * d.C);
+        const int sc = kk - st_r[t] * d.S * d.C;
+        st_s[t] = sc / d.C;
+        st_c[t] = sc - st_s[t] * d.C;
     }
 
-    // chunk load for stage base offset kc (16 B per staged chunk)
-    auto load_chunk = [&](int t, int kc) -> short8 {
-        short8 v = {};
-        const int chunk = tid + t * CONV_THREADS;
-        const int kk = kc + (chunk & 7) * 8;
-        if (st_n[t] >= 0 && kk < rsc) {
-            const int r = kk / (d.S * d.C);
-            const int sc = kk - r * d.S * d.C;
-            const int s = sc / d.C;
-            const int c = sc - s * d.C;
-            const int hi = st_hi[t] + r;
-            const int wi = st_wi[t] + s;
-            if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W)
-                v = *reinterpret_cast<const short8*>(
-                    x + (((st_n[t] * d.H + hi) * d.W + wi) * (int64_t)d.C + c));
+    auto load_stage = [&](short8* dst) {
+#pragma unroll
+        for (int t = 0; t < CPT; ++t) {
+            short8 v = {};
+            if (st_n[t] >= 0 && st_r[t] < d.R) {
+                const int hi = st_hi[t] + st_r[t];
+                const int wi = st_wi[t] + st_s[t];
+                if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W)
+                    v = *reinterpret_cast<const short8*>(
+                        x + (((st_n[t] * d.H + hi) * d.W + wi) * (int64_t)d.C +
+                             st_c[t]));
+            }
+            dst[t] = v;
+            // advance tap by one stage (+BK2) with carries — no divides
+            int c = st_c[t] + BK2;
+            while (c >= d.C) {
+                c -= d.C;
+                if (++st_s[t] == d.S) { st_s[t] = 0; ++st_r[t]; }
+            }
+            st_c[t] = c;
         }
-        return v;
     };
-    // LDS layout: [sub][row][APITCH], sub = kk/32 within the 64-deep stage
-    auto lds_write = [&](uint16_t* buf, int t, short8 v) {
-        const int chunk = tid + t * CONV_THREADS;
-        const int koff = (chunk & 7) * 8;       // 0..56 within the 64 stage
-        const int sub = koff >> 5;
-        *reinterpret_cast<short8*>(
-            &buf[(sub * BM + st_row[t]) * CONV_APITCH + (koff & 31)]) = v;
+    auto lds_write = [&](uint16_t* buf, const short8* src) {
+#pragma unroll
+        for (int t = 0; t < CPT; ++t) {
+            const int chunk = tid + t * CONV_THREADS;
+            if (chunk < CHUNKS) {
+                const int row = chunk >> 3;
+                const int koff = (chunk & 7) * 8;
+                const int sub = koff >> 5;
+                *reinterpret_cast<short8*>(
+                    &buf[(sub * BM + row) * CONV_APITCH + (koff & 31)]) = src[t];
+            }
+        }
     };
 
-    floatx4 acc[MF][NF] = {};
     const int a_row = wave_m * (BM / WAVES_M) + (lane & 15);
     const int a_koff = (lane >> 4) * 8;
     const int b_col = col0 + wave_n * (CONV_BN / WAVES_N) + (lane & 15);
 
+    auto load_b = [&](short8 (*dst)[NF], int stage) {
+#pragma unroll
+        for (int sub = 0; sub < 2; ++sub)
+#pragma unroll
+            for (int nf = 0; nf < NF; ++nf)
+                dst[sub][nf] = *reinterpret_cast<const short8*>(
+                    w + (int64_t)(b_col + nf * 16) * rsc + stage * BK2 +
+                    sub * CONV_BK + a_koff);
+    };
+
+    floatx4 acc[MF][NF] = {};
     const int n_stages = (rsc + BK2 - 1) / BK2;
     short8 stage[CPT];
-    // prologue: chunk 0 -> buf0; chunk 1 -> regs
-#pragma unroll
-    for (int t = 0; t < CPT; ++t) stage[t] = load_chunk(t, 0);
-#pragma unroll
-    for (int t = 0; t < CPT; ++t)
-        if (tid + t * CONV_THREADS < CHUNKS) lds_write(A_lds[0], t, stage[t]);
-    if (n_stages > 1) {
-#pragma unroll
-        for (int t = 0; t < CPT; ++t) stage[t] = load_chunk(t, BK2);
-    }
+    short8 breg[2][2][NF];   // [parity][sub][nf]
+
+    // prologue: stage 0 -> buf0; stage 1 -> regs; B(0) -> breg[0]
+    load_stage(stage);
+    lds_write(A_lds[0], stage);
+    if (n_stages > 1) load_stage(stage);
+    load_b(breg[0], 0);
     __syncthreads();
 
     for (int i = 0; i < n_stages; ++i) {
         const uint16_t* buf = A_lds[i & 1];
-        // regs hold stage i+1: write them to the other buffer, then start
-        // loading stage i+2 (latency hides under this stage's MFMAs)
         if (i + 1 < n_stages) {
-            uint16_t* nbuf = A_lds[(i + 1) & 1];
-#pragma unroll
-            for (int t = 0; t < CPT; ++t)
-                if (tid + t * CONV_THREADS < CHUNKS) lds_write(nbuf, t, stage[t]);
-            if (i + 2 < n_stages) {
-#pragma unroll
-                for (int t = 0; t < CPT; ++t)
-                    stage[t] = load_chunk(t, (i + 2) * BK2);
-            }
+            lds_write(A_lds[(i + 1) & 1], stage);
+            if (i + 2 < n_stages) load_stage(stage);
+            load_b(breg[(i + 1) & 1], i + 1);
         }
+        const short8(*b)[NF] = breg[i & 1];
         const int kc = i * BK2;
 #pragma unroll
         for (int sub = 0; sub < 2; ++sub) {
             if (kc + sub * CONV_BK >= rsc) break;
-            short8 b[NF];
-#pragma unroll
-            for (int nf = 0; nf < NF; ++nf)
-                b[nf] = *reinterpret_cast<const short8*>(
-                    w + (int64_t)(b_col + nf * 16) * rsc + kc + sub * CONV_BK + a_koff);
 #pragma unroll
             for (int mf = 0; mf < MF; ++mf) {
                 const short8 a = *reinterpret_cast<const short8*>(
                     &buf[(sub * BM + a_row + mf * 16) * CONV_APITCH + a_koff]);
 #pragma unroll
                 for (int nf = 0; nf < NF; ++nf)
-                    acc[mf][nf] = MFMA_BF16(a, b[nf], acc[mf][nf]);
+                    acc[mf][nf] = MFMA_BF16(a, b[sub][nf], acc[mf][nf]);
             }
         }
         __syncthreads();
@@ -193,6 +202,7 @@ extern "C" void launch_conv_fwd(const void* x, const void* w, void* y,
 
 // ---------------------------------------------------------------------------
 // Direct kernels for the C=3 stem conv (implicit-GEMM needs C%8==0).
+// K must be 64 (the ResNet stem).
 // ---------------------------------------------------------------------------
 
 __global__ void __launch_bounds__(256)
@@ -291,12 +301,11 @@ k_conv_stem_wgrad(const uint16_t* __restrict__ x,
             }
         }
     }
-    // fold the 32 m-lanes per k: LDS [8 k][32 lanes spread over 27 taps]
+    // fold the 32 m-lanes per k: LDS [8 k][32 lanes][32 taps]
     __shared__ float red[8][32][32];
 #pragma unroll
     for (int j = 0; j < 32; ++j) red[kl][mlane][j] = part[j];
     __syncthreads();
-    // 256 threads: thread -> (k-octet slot, tap); fold 32 lanes
     const int kk = threadIdx.x >> 5;         // 0..7
     const int j0 = threadIdx.x & 31;         // tap (first 27 valid)
     if (j0 < rsc) {
