@@ -133,22 +133,21 @@ k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_r
     floatx4 acc[MF][NF] = {};
     const int n_stages = (rsk + BK2 - 1) / BK2;
     short8 stage[CPT];
-    short8 breg[2][2][NF];
+    short8 breg_a[2][NF], breg_b[2][NF];
 
     load_stage(stage);
     lds_write(A_lds[0], stage);
     if (n_stages > 1) load_stage(stage);
-    load_b(breg[0]);
+    load_b(breg_a);
     __syncthreads();
 
-    for (int i = 0; i < n_stages; ++i) {
-        const uint16_t* buf = A_lds[i & 1];
+    auto step = [&](int i, const uint16_t* buf, uint16_t* nbuf,
+                    short8 (&bcur)[2][NF], short8 (&bnext)[2][NF]) {
         if (i + 1 < n_stages) {
-            lds_write(A_lds[(i + 1) & 1], stage);
+            lds_write(nbuf, stage);
             if (i + 2 < n_stages) load_stage(stage);
-            load_b(breg[(i + 1) & 1]);
+            load_b(bnext);
         }
-        const short8(*b)[NF] = breg[i & 1];
         const int kc = i * BK2;
 #pragma unroll
         for (int sub = 0; sub < 2; ++sub) {
@@ -159,10 +158,16 @@ k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_r
                     &buf[(sub * BM + a_row + mf * 16) * CONV_APITCH + a_koff]);
 #pragma unroll
                 for (int nf = 0; nf < NF; ++nf)
-                    acc[mf][nf] = MFMA_BF16(a, b[sub][nf], acc[mf][nf]);
+                    acc[mf][nf] = MFMA_BF16(a, bcur[sub][nf], acc[mf][nf]);
             }
         }
         __syncthreads();
+    };
+    for (int i = 0; i < n_stages;) {
+        step(i, A_lds[0], A_lds[1], breg_a, breg_b);
+        if (++i >= n_stages) break;
+        step(i, A_lds[1], A_lds[0], breg_b, breg_a);
+        ++i;
     }
 
     const int64_t out_row0 = m0 + wave_m * (BM / WAVES_M) + (lane >> 4) * 4;
