@@ -180,7 +180,10 @@ def bn_msplit(M: int, C: int) -> int:
     while keeping the partial-combine kernels cheap."""
     cols = max(1, C // 64)
     msplit = max(1, min(512 // cols, 512))
-    return max(1, min(msplit, (M + 31) // 32))
+    msplit = max(1, min(msplit, (M + 31) // 32))
+    if msplit >= 4:
+        msplit &= ~3  # multiple of 4: per-channel partial rows stay 16B-aligned
+    return msplit
 
 
 def bn_stats(x: torch.Tensor, partials: torch.Tensor, M: int, C: int,

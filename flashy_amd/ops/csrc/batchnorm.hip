@@ -75,10 +75,18 @@ __device__ __forceinline__ void combine_partials(
     const float* row0 = partials + (int64_t)c * msplit;
     const float* row1 = partials + ((int64_t)C + c) * msplit;
     float s = 0.f, s2 = 0.f;
-    for (int i = slane; i < msplit; i += 4) {
-        s += row0[i];
-        s2 += row1[i];
+    int i = slane * 4;
+    for (; i + 4 <= msplit; i += 16) {   // float4 rows: 4x fewer, wider loads
+        const float4 a = *reinterpret_cast<const float4*>(row0 + i);
+        const float4 b = *reinterpret_cast<const float4*>(row1 + i);
+        s += a.x + a.y + a.z + a.w;
+        s2 += b.x + b.y + b.z + b.w;
     }
+    if (slane == 0)  // scalar tail when msplit % 4 != 0
+        for (int j = msplit & ~3; j < msplit; ++j) {
+            s += row0[j];
+            s2 += row1[j];
+        }
     __shared__ float red[2][4][64];
     red[0][slane][threadIdx.x & 63] = s;
     red[1][slane][threadIdx.x & 63] = s2;

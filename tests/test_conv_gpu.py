@@ -194,14 +194,30 @@ def test_native_resnet18_matches_torch():
     assert diff / spread < 0.35, (diff, spread)
     assert abs(loss_n.item() - loss_t.item()) < 0.25, \
         (loss_n.item(), loss_t.item())
-    # spot-check a conv weight grad (bf16 path vs fp32 torch): element-wise
-    # max is inflated by ReLU-mask flips at bf16 rounding boundaries, so use
-    # relative L2 + cosine similarity
-    gn = model.layer1[0].conv1.weight.grad.permute(0, 3, 1, 2).flatten()
-    gt = twin.layer1[0].conv1.weight.grad.flatten()
-    rel = (gn - gt).norm().item() / (gt.norm().item() + 1e-8)
-    cos = torch.nn.functional.cosine_similarity(gn, gt, dim=0).item()
-    assert rel < 0.2 and cos > 0.98, (rel, cos)
+    # Early-layer weight grads of a bf16 stack legitimately drift from the
+    # fp32 twin (long backward chains + ReLU mask flips).  Calibrate the
+    # tolerance against the same twin run under bf16 autocast: our kernels
+    # must not be much noisier than torch's own bf16 path.
+    twin2 = resnet18(num_classes=10, small_input=True).cuda().train()
+    twin2.load_state_dict(twin.state_dict())
+    twin2.zero_grad()
+    with torch.autocast("cuda", torch.bfloat16):
+        logits_a = twin2(x)
+    F.cross_entropy(logits_a.float(), y).backward()
+
+    def _rel_cos(a, b):
+        a, b = a.flatten(), b.flatten()
+        rel = (a - b).norm().item() / (b.norm().item() + 1e-8)
+        cos = torch.nn.functional.cosine_similarity(a, b, dim=0).item()
+        return rel, cos
+
+    gn = model.layer1[0].conv1.weight.grad.permute(0, 3, 1, 2)
+    gt = twin.layer1[0].conv1.weight.grad
+    ga = twin2.layer1[0].conv1.weight.grad
+    ours_rel, ours_cos = _rel_cos(gn, gt)
+    floor_rel, floor_cos = _rel_cos(ga, gt)
+    assert ours_rel < max(2.5 * floor_rel, 0.05), (ours_rel, floor_rel)
+    assert ours_cos > 1 - 2.5 * (1 - floor_cos) - 1e-3, (ours_cos, floor_cos)
 
 
 @requires_gpu
