@@ -171,7 +171,9 @@ class EntryPoint:
             shutil.rmtree(xp.folder, ignore_errors=True)
         xp.enter()
         # Persist the resolved config so get_xp_from_sig can rebuild the XP.
-        with write_and_rename(xp.folder / "config.yaml", "w") as fh:
+        # pid-suffixed temp: concurrent DDP workers of one run write the same
+        # content and must not race each other's rename.
+        with write_and_rename(xp.folder / "config.yaml", "w", pid=True) as fh:
             import yaml
             yaml.safe_dump(xp.cfg.to_plain(), fh)
         return self._fn(xp.cfg)
