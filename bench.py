@@ -38,10 +38,10 @@ MODELS = {"resnet18": resnet18, "resnet50": resnet50}
 NATIVE_MODELS = {"resnet18": native_resnet18, "resnet50": native_resnet50}
 
 
-def build_step(model, optim, static_x, static_y, autocast: bool, distributed: bool):
+def build_fwd_bwd(model, optim, static_x, static_y, autocast: bool):
     fused = isinstance(optim, FusedSGD)
 
-    def step():
+    def fwd_bwd():
         optim.zero_grad(set_to_none=False)
         with torch.autocast("cuda", torch.bfloat16, enabled=autocast):
             logits = model(static_x)
@@ -50,6 +50,17 @@ def build_step(model, optim, static_x, static_y, autocast: bool, distributed: bo
         else:
             loss = torch.nn.functional.cross_entropy(logits, static_y)
         loss.backward()
+        return loss
+
+    return fwd_bwd
+
+
+def build_step(model, optim, static_x, static_y, autocast: bool, distributed: bool):
+    fused = isinstance(optim, FusedSGD)
+    fwd_bwd = build_fwd_bwd(model, optim, static_x, static_y, autocast)
+
+    def step():
+        loss = fwd_bwd()
         if distributed:
             if fused:
                 distrib.sync_flat_gradients(optim)
@@ -143,10 +154,28 @@ def main():
     static_y = torch.zeros_like(ys[0], device=device)
 
     autocast = use_cuda and not native  # the native model is bf16 internally
-    step = build_step(model, optim, static_x, static_y, autocast, ws > 1)
+    use_graph = use_cuda and not args.no_graph and not args.ref
 
-    use_graph = use_cuda and ws == 1 and not args.no_graph and not args.ref
-    runner = CapturedStep(step, warmup=3).capture() if use_graph else step
+    if use_graph and ws > 1 and isinstance(optim, FusedSGD):
+        # multi-GPU: capture fwd+bwd as one graph (grads land in the static
+        # flat buffers), then RCCL all-reduce + fused optimizer step eagerly
+        # — collectives stay outside the graph.
+        fwd_bwd = CapturedStep(
+            build_fwd_bwd(model, optim, static_x, static_y, autocast),
+            warmup=3).capture()
+
+        def runner():
+            loss = fwd_bwd()
+            distrib.sync_flat_gradients(optim)
+            optim.step()
+            return loss
+    elif use_graph and ws == 1:
+        runner = CapturedStep(
+            build_step(model, optim, static_x, static_y, autocast, False),
+            warmup=3).capture()
+    else:
+        use_graph = False
+        runner = build_step(model, optim, static_x, static_y, autocast, ws > 1)
 
     def one_step(i: int):
         static_x.copy_(xs[i % pool_n], non_blocking=True)

@@ -125,12 +125,33 @@ class ConvDims(tp.NamedTuple):
         return ConvDims(N, H, W, C, K, R, S, Ho, Wo, stride, pad)
 
 
+def _splitk_plan(M: int, ktiles: int, red_stages: int):
+    """If the best single-pass grid underfills the 256-CU chip, split the
+    reduction over grid.z into fp32 workspace slices (returns zn or 0)."""
+    blocks64 = ((M + 63) // 64) * ktiles
+    if blocks64 >= 208 or red_stages < 4:
+        return 0  # a single-pass BM>=64 grid already fills the chip
+    zn = max(2, min((256 + blocks64 - 1) // max(1, blocks64), red_stages // 2))
+    return min(zn, 8)
+
+
 def conv_fwd(x: torch.Tensor, w: torch.Tensor, y: torch.Tensor, d: ConvDims,
              relu: bool = False) -> None:
     ext = require()
     if d.C % 8 == 0:
-        assert d.K % 64 == 0 and (d.R * d.S * d.C) % 32 == 0, d
-        ext.conv_fwd(x.data_ptr(), w.data_ptr(), y.data_ptr(), *d, relu, _stream())
+        rsc = d.R * d.S * d.C
+        assert d.K % 64 == 0 and rsc % 32 == 0, d
+        M = d.N * d.Ho * d.Wo
+        zn = _splitk_plan(M, d.K // 64, (rsc + 63) // 64)
+        if zn:
+            ws = torch.empty(zn * M * d.K, dtype=torch.float32, device=x.device)
+            ext.conv_fwd_splitk(x.data_ptr(), w.data_ptr(), ws.data_ptr(), *d,
+                                zn, _stream())
+            ext.splitk_combine(ws.data_ptr(), y.data_ptr(), M * d.K, zn, relu,
+                               _stream())
+        else:
+            ext.conv_fwd(x.data_ptr(), w.data_ptr(), y.data_ptr(), *d, relu,
+                         _stream())
     else:
         assert not relu
         ext.conv_stem_fwd(x.data_ptr(), w.data_ptr(), y.data_ptr(), *d, _stream())
@@ -140,7 +161,18 @@ def conv_dgrad(dout: torch.Tensor, w_rsck: torch.Tensor, dx: torch.Tensor,
                d: ConvDims) -> None:
     ext = require()
     assert d.C % 64 == 0 and d.K % 32 == 0, d
-    ext.conv_dgrad(dout.data_ptr(), w_rsck.data_ptr(), dx.data_ptr(), *d, _stream())
+    M = d.N * d.H * d.W
+    rsk = d.R * d.S * d.K
+    zn = _splitk_plan(M, d.C // 64, (rsk + 63) // 64)
+    if zn:
+        ws = torch.empty(zn * M * d.C, dtype=torch.float32, device=dout.device)
+        ext.conv_dgrad_splitk(dout.data_ptr(), w_rsck.data_ptr(), ws.data_ptr(),
+                              *d, zn, _stream())
+        ext.splitk_combine(ws.data_ptr(), dx.data_ptr(), M * d.C, zn, False,
+                           _stream())
+    else:
+        ext.conv_dgrad(dout.data_ptr(), w_rsck.data_ptr(), dx.data_ptr(), *d,
+                       _stream())
 
 
 def weight_transpose(w: torch.Tensor, wt: torch.Tensor) -> None:

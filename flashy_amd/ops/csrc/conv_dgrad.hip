@@ -14,10 +14,11 @@
 
 #include "conv_common.h"
 
-template <int BM>
+template <int BM, bool SPLITK>
 __global__ void __launch_bounds__(CONV_THREADS)
 k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_rsck,
-             uint16_t* __restrict__ dx, ConvDims d) {
+             uint16_t* __restrict__ dx, float* __restrict__ ws_out,
+             ConvDims d, int stages_per_split) {
     constexpr int WAVES_M = BM >= 64 ? 2 : 1;
     constexpr int WAVES_N = 4 / WAVES_M;
     constexpr int MF = BM / WAVES_M / 16;
@@ -54,9 +55,12 @@ k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_r
         } else {
             st_n[t] = -1;
         }
-        st_r[t] = 0;
-        st_s[t] = 0;
-        st_k[t] = (chunk & 7) * 8;   // < 64 <= K
+        const int kk0 = (SPLITK ? blockIdx.z * stages_per_split * BK2 : 0) +
+                        (chunk & 7) * 8;
+        st_r[t] = kk0 / (d.S * d.K);
+        const int sk0 = kk0 - st_r[t] * d.S * d.K;
+        st_s[t] = sk0 / d.K;
+        st_k[t] = sk0 - st_s[t] * d.K;
     }
 
     auto load_stage = [&](short8* dst) {
@@ -106,10 +110,12 @@ k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_r
     int b_r[2], b_s[2], b_k[2];
 #pragma unroll
     for (int sub = 0; sub < 2; ++sub) {
-        b_r[sub] = 0;
-        b_s[sub] = 0;
-        b_k[sub] = sub * CONV_BK + a_koff;
-        // K >= 64 so initial k < K always
+        const int kk0 = (SPLITK ? blockIdx.z * stages_per_split * BK2 : 0) +
+                        sub * CONV_BK + a_koff;
+        b_r[sub] = kk0 / (d.S * d.K);
+        const int sk0 = kk0 - b_r[sub] * d.S * d.K;
+        b_s[sub] = sk0 / d.K;
+        b_k[sub] = sk0 - b_s[sub] * d.K;
     }
     auto load_b = [&](short8 (*dst)[NF]) {
 #pragma unroll
@@ -131,7 +137,11 @@ k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_r
     };
 
     floatx4 acc[MF][NF] = {};
-    const int n_stages = (rsk + BK2 - 1) / BK2;
+    const int all_stages = (rsk + BK2 - 1) / BK2;
+    const int s0 = SPLITK ? blockIdx.z * stages_per_split : 0;
+    const int n_stages = SPLITK
+        ? (all_stages - s0 < stages_per_split ? all_stages - s0 : stages_per_split)
+        : all_stages;
     short8 stage[CPT];
     short8 breg_a[2][NF], breg_b[2][NF];
 
@@ -148,7 +158,7 @@ k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_r
             if (i + 2 < n_stages) load_stage(stage);
             load_b(bnext);
         }
-        const int kc = i * BK2;
+        const int kc = (s0 + i) * BK2;
 #pragma unroll
         for (int sub = 0; sub < 2; ++sub) {
             if (kc + sub * CONV_BK >= rsk) break;
@@ -179,9 +189,14 @@ k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_r
 #pragma unroll
             for (int rr = 0; rr < 4; ++rr) {
                 const int64_t row = out_row0 + mf * 16 + rr;
-                if (row < M)
-                    dx[row * d.C + out_col0 + nf * 16] =
-                        f32_to_bf16(acc[mf][nf][rr]);
+                if (row < M) {
+                    if (SPLITK)
+                        ws_out[((int64_t)blockIdx.z * M + row) * d.C +
+                               out_col0 + nf * 16] = acc[mf][nf][rr];
+                    else
+                        dx[row * d.C + out_col0 + nf * 16] =
+                            f32_to_bf16(acc[mf][nf][rr]);
+                }
             }
 }
 
@@ -197,9 +212,23 @@ extern "C" void launch_conv_dgrad(const void* dout, const void* w_rsck,
     auto dd = (const uint16_t*)dout;
     auto ww = (const uint16_t*)w_rsck;
     auto xx = (uint16_t*)dx;
-    if (bm == 128) k_conv_dgrad<128><<<grid, CONV_THREADS, 0, stream>>>(dd, ww, xx, d);
-    else if (bm == 64) k_conv_dgrad<64><<<grid, CONV_THREADS, 0, stream>>>(dd, ww, xx, d);
-    else k_conv_dgrad<32><<<grid, CONV_THREADS, 0, stream>>>(dd, ww, xx, d);
+    if (bm == 128) k_conv_dgrad<128, false><<<grid, CONV_THREADS, 0, stream>>>(dd, ww, xx, nullptr, d, 0);
+    else if (bm == 64) k_conv_dgrad<64, false><<<grid, CONV_THREADS, 0, stream>>>(dd, ww, xx, nullptr, d, 0);
+    else k_conv_dgrad<32, false><<<grid, CONV_THREADS, 0, stream>>>(dd, ww, xx, nullptr, d, 0);
+}
+
+extern "C" void launch_conv_dgrad_splitk(const void* dout, const void* w_rsck,
+                                         void* ws, ConvDims d, int zn,
+                                         hipStream_t stream) {
+    const int64_t M = (int64_t)d.N * d.H * d.W;
+    const int rsk = d.R * d.S * d.K;
+    const int all_stages = (rsk + 63) / 64;
+    const int spz = (all_stages + zn - 1) / zn;
+    const int zeff = (all_stages + spz - 1) / spz;
+    dim3 grid((unsigned)((M + 63) / 64), (unsigned)(d.C / CONV_BN), (unsigned)zeff);
+    k_conv_dgrad<64, true><<<grid, CONV_THREADS, 0, stream>>>(
+        (const uint16_t*)dout, (const uint16_t*)w_rsck, nullptr, (float*)ws,
+        d, spz);
 }
 
 // ---------------------------------------------------------------------------

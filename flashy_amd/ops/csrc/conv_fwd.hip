@@ -20,10 +20,11 @@
 
 #include "conv_common.h"
 
-template <int BM, bool RELU>
+template <int BM, bool RELU, bool SPLITK>
 __global__ void __launch_bounds__(CONV_THREADS)
 k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
-           uint16_t* __restrict__ y, ConvDims d) {
+           uint16_t* __restrict__ y, float* __restrict__ ws_out,
+           ConvDims d, int stages_per_split) {
     constexpr int WAVES_M = BM >= 64 ? 2 : 1;
     constexpr int WAVES_N = 4 / WAVES_M;
     constexpr int MF = BM / WAVES_M / 16;      // m fragments per wave
@@ -62,7 +63,8 @@ k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
         } else {
             st_n[t] = -1;
         }
-        const int kk = (chunk & 7) * 8;   // tap offset of stage 0
+        const int kk = (SPLITK ? blockIdx.z * stages_per_split * BK2 : 0) +
+                       (chunk & 7) * 8;   // tap offset of the first stage
         st_r[t] = kk / (d.S * d.C);
         const int sc = kk - st_r[t] * d.S * d.C;
         st_s[t] = sc / d.C;
@@ -120,15 +122,19 @@ k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
     };
 
     floatx4 acc[MF][NF] = {};
-    const int n_stages = (rsc + BK2 - 1) / BK2;
+    const int all_stages = (rsc + BK2 - 1) / BK2;
+    const int s0 = SPLITK ? blockIdx.z * stages_per_split : 0;
+    const int n_stages = SPLITK
+        ? (all_stages - s0 < stages_per_split ? all_stages - s0 : stages_per_split)
+        : all_stages;
     short8 stage[CPT];
     short8 breg_a[2][NF], breg_b[2][NF];
 
-    // prologue: stage 0 -> buf0; stage 1 -> regs; B(0) -> breg_a
+    // prologue: first stage -> buf0; next -> regs; its B -> breg_a
     load_stage(stage);
     lds_write(A_lds[0], stage);
     if (n_stages > 1) load_stage(stage);
-    load_b(breg_a, 0);
+    load_b(breg_a, s0);
     __syncthreads();
 
     // even/odd bodies keep buffer/B-register parity COMPILE-TIME (a runtime
@@ -138,9 +144,9 @@ k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
         if (i + 1 < n_stages) {
             lds_write(nbuf, stage);
             if (i + 2 < n_stages) load_stage(stage);
-            load_b(bnext, i + 1);
+            load_b(bnext, s0 + i + 1);
         }
-        const int kc = i * BK2;
+        const int kc = (s0 + i) * BK2;
 #pragma unroll
         for (int sub = 0; sub < 2; ++sub) {
             if (kc + sub * CONV_BK >= rsc) break;
@@ -173,10 +179,45 @@ k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
                 const int64_t row = out_row0 + mf * 16 + rr;
                 if (row < M) {
                     float v = acc[mf][nf][rr];
-                    if (RELU) v = fmaxf(v, 0.f);
-                    y[row * d.K + out_col0 + nf * 16] = f32_to_bf16(v);
+                    if (SPLITK) {
+                        ws_out[((int64_t)blockIdx.z * M + row) * d.K +
+                               out_col0 + nf * 16] = v;
+                    } else {
+                        if (RELU) v = fmaxf(v, 0.f);
+                        y[row * d.K + out_col0 + nf * 16] = f32_to_bf16(v);
+                    }
                 }
             }
+}
+
+// combine split-K fp32 partials -> bf16 (+optional relu)
+__global__ void __launch_bounds__(256)
+k_splitk_combine(const float* __restrict__ ws, uint16_t* __restrict__ out,
+                 int64_t total, int zn, int relu) {
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         i * 4 < total; i += stride) {
+        float4 acc = *reinterpret_cast<const float4*>(ws + i * 4);
+        for (int z = 1; z < zn; ++z) {
+            const float4 v = *reinterpret_cast<const float4*>(
+                ws + (int64_t)z * total + i * 4);
+            acc.x += v.x; acc.y += v.y; acc.z += v.z; acc.w += v.w;
+        }
+        if (relu) {
+            acc.x = fmaxf(acc.x, 0.f); acc.y = fmaxf(acc.y, 0.f);
+            acc.z = fmaxf(acc.z, 0.f); acc.w = fmaxf(acc.w, 0.f);
+        }
+        ushort4 o;
+        o.x = f32_to_bf16(acc.x); o.y = f32_to_bf16(acc.y);
+        o.z = f32_to_bf16(acc.z); o.w = f32_to_bf16(acc.w);
+        *reinterpret_cast<ushort4*>(out + i * 4) = o;
+    }
+}
+
+extern "C" void launch_splitk_combine(const void* ws, void* out, int64_t total,
+                                      int zn, int relu, hipStream_t stream) {
+    k_splitk_combine<<<ew_grid(total / 4, 256, 2), 256, 0, stream>>>(
+        (const float*)ws, (uint16_t*)out, total, zn, relu);
 }
 
 // pick BM so the grid keeps >= ~208 workgroups where possible (256 CUs)
@@ -197,14 +238,28 @@ extern "C" void launch_conv_fwd(const void* x, const void* w, void* y,
     auto ww = (const uint16_t*)w;
     auto yy = (uint16_t*)y;
     if (relu) {
-        if (bm == 128) k_conv_fwd<128, true><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, d);
-        else if (bm == 64) k_conv_fwd<64, true><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, d);
-        else k_conv_fwd<32, true><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, d);
+        if (bm == 128) k_conv_fwd<128, true, false><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, d, 0);
+        else if (bm == 64) k_conv_fwd<64, true, false><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, d, 0);
+        else k_conv_fwd<32, true, false><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, d, 0);
     } else {
-        if (bm == 128) k_conv_fwd<128, false><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, d);
-        else if (bm == 64) k_conv_fwd<64, false><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, d);
-        else k_conv_fwd<32, false><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, d);
+        if (bm == 128) k_conv_fwd<128, false, false><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, d, 0);
+        else if (bm == 64) k_conv_fwd<64, false, false><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, d, 0);
+        else k_conv_fwd<32, false, false><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, d, 0);
     }
+}
+
+// split-K path: BM=64 tiles, grid.z over stage ranges, fp32 workspace
+// [zn][M][K], then k_splitk_combine.
+extern "C" void launch_conv_fwd_splitk(const void* x, const void* w, void* ws,
+                                       ConvDims d, int zn, hipStream_t stream) {
+    const int64_t M = (int64_t)d.N * d.Ho * d.Wo;
+    const int rsc = d.R * d.S * d.C;
+    const int all_stages = (rsc + 63) / 64;
+    const int spz = (all_stages + zn - 1) / zn;
+    const int zeff = (all_stages + spz - 1) / spz;
+    dim3 grid((unsigned)((M + 63) / 64), (unsigned)(d.K / CONV_BN), (unsigned)zeff);
+    k_conv_fwd<64, false, true><<<grid, CONV_THREADS, 0, stream>>>(
+        (const uint16_t*)x, (const uint16_t*)w, nullptr, (float*)ws, d, spz);
 }
 
 // ---------------------------------------------------------------------------

@@ -29,6 +29,12 @@ void launch_conv_stem_wgrad(const void* x, const void* dout, void* dw,
                             ConvDims d, hipStream_t stream);
 void launch_conv_dgrad(const void* dout, const void* w_rsck, void* dx,
                        ConvDims d, hipStream_t stream);
+void launch_conv_fwd_splitk(const void* x, const void* w, void* ws, ConvDims d,
+                            int zn, hipStream_t stream);
+void launch_conv_dgrad_splitk(const void* dout, const void* w_rsck, void* ws,
+                              ConvDims d, int zn, hipStream_t stream);
+void launch_splitk_combine(const void* ws, void* out, int64_t total, int zn,
+                           int relu, hipStream_t stream);
 void launch_weight_transpose(const void* w, void* wt, int K, int rsc,
                              hipStream_t stream);
 void launch_conv_wgrad(const void* x, const void* dout, void* dw, ConvDims d,
@@ -121,6 +127,32 @@ PYBIND11_MODULE(_hip_ops, m) {
                                 (void*)dx,
                                 make_dims(N, H, W, C, K, R, S, Ho, Wo, stride, pad),
                                 as_stream(stream));
+              check_last();
+          });
+    m.def("conv_fwd_splitk",
+          [](uintptr_t x, uintptr_t w, uintptr_t ws, int N, int H, int W,
+             int C, int K, int R, int S, int Ho, int Wo, int stride, int pad,
+             int zn, uintptr_t stream) {
+              launch_conv_fwd_splitk((const void*)x, (const void*)w, (void*)ws,
+                                     make_dims(N, H, W, C, K, R, S, Ho, Wo, stride, pad),
+                                     zn, as_stream(stream));
+              check_last();
+          });
+    m.def("conv_dgrad_splitk",
+          [](uintptr_t dout, uintptr_t w_rsck, uintptr_t ws, int N, int H,
+             int W, int C, int K, int R, int S, int Ho, int Wo, int stride,
+             int pad, int zn, uintptr_t stream) {
+              launch_conv_dgrad_splitk((const void*)dout, (const void*)w_rsck,
+                                       (void*)ws,
+                                       make_dims(N, H, W, C, K, R, S, Ho, Wo, stride, pad),
+                                       zn, as_stream(stream));
+              check_last();
+          });
+    m.def("splitk_combine",
+          [](uintptr_t ws, uintptr_t out, int64_t total, int zn, bool relu,
+             uintptr_t stream) {
+              launch_splitk_combine((const void*)ws, (void*)out, total, zn,
+                                    relu ? 1 : 0, as_stream(stream));
               check_last();
           });
     m.def("weight_transpose",
