@@ -142,12 +142,15 @@ def conv_fwd(x: torch.Tensor, w: torch.Tensor, y: torch.Tensor, d: ConvDims,
         rsc = d.R * d.S * d.C
         assert d.K % 64 == 0 and rsc % 32 == 0, d
         M = d.N * d.Ho * d.Wo
-        zn = _splitk_plan(M, d.K // 64, (rsc + 63) // 64)
+        stages = (rsc + 63) // 64
+        zn = _splitk_plan(M, d.K // 64, stages)
         if zn:
-            ws = torch.empty(zn * M * d.K, dtype=torch.float32, device=x.device)
+            spz = (stages + zn - 1) // zn
+            zeff = (stages + spz - 1) // spz
+            ws = torch.empty(zeff * M * d.K, dtype=torch.float32, device=x.device)
             ext.conv_fwd_splitk(x.data_ptr(), w.data_ptr(), ws.data_ptr(), *d,
-                                zn, _stream())
-            ext.splitk_combine(ws.data_ptr(), y.data_ptr(), M * d.K, zn, relu,
+                                spz, _stream())
+            ext.splitk_combine(ws.data_ptr(), y.data_ptr(), M * d.K, zeff, relu,
                                _stream())
         else:
             ext.conv_fwd(x.data_ptr(), w.data_ptr(), y.data_ptr(), *d, relu,
@@ -163,12 +166,15 @@ def conv_dgrad(dout: torch.Tensor, w_rsck: torch.Tensor, dx: torch.Tensor,
     assert d.C % 64 == 0 and d.K % 32 == 0, d
     M = d.N * d.H * d.W
     rsk = d.R * d.S * d.K
-    zn = _splitk_plan(M, d.C // 64, (rsk + 63) // 64)
+    stages = (rsk + 63) // 64
+    zn = _splitk_plan(M, d.C // 64, stages)
     if zn:
-        ws = torch.empty(zn * M * d.C, dtype=torch.float32, device=dout.device)
+        spz = (stages + zn - 1) // zn
+        zeff = (stages + spz - 1) // spz
+        ws = torch.empty(zeff * M * d.C, dtype=torch.float32, device=dout.device)
         ext.conv_dgrad_splitk(dout.data_ptr(), w_rsck.data_ptr(), ws.data_ptr(),
-                              *d, zn, _stream())
-        ext.splitk_combine(ws.data_ptr(), dx.data_ptr(), M * d.C, zn, False,
+                              *d, spz, _stream())
+        ext.splitk_combine(ws.data_ptr(), dx.data_ptr(), M * d.C, zeff, False,
                            _stream())
     else:
         ext.conv_dgrad(dout.data_ptr(), w_rsck.data_ptr(), dx.data_ptr(), *d,

@@ -218,12 +218,11 @@ extern "C" void launch_conv_dgrad(const void* dout, const void* w_rsck,
 }
 
 extern "C" void launch_conv_dgrad_splitk(const void* dout, const void* w_rsck,
-                                         void* ws, ConvDims d, int zn,
+                                         void* ws, ConvDims d, int spz,
                                          hipStream_t stream) {
     const int64_t M = (int64_t)d.N * d.H * d.W;
     const int rsk = d.R * d.S * d.K;
     const int all_stages = (rsk + 63) / 64;
-    const int spz = (all_stages + zn - 1) / zn;
     const int zeff = (all_stages + spz - 1) / spz;
     dim3 grid((unsigned)((M + 63) / 64), (unsigned)(d.C / CONV_BN), (unsigned)zeff);
     k_conv_dgrad<64, true><<<grid, CONV_THREADS, 0, stream>>>(

@@ -251,11 +251,10 @@ extern "C" void launch_conv_fwd(const void* x, const void* w, void* y,
 // split-K path: BM=64 tiles, grid.z over stage ranges, fp32 workspace
 // [zn][M][K], then k_splitk_combine.
 extern "C" void launch_conv_fwd_splitk(const void* x, const void* w, void* ws,
-                                       ConvDims d, int zn, hipStream_t stream) {
+                                       ConvDims d, int spz, hipStream_t stream) {
     const int64_t M = (int64_t)d.N * d.Ho * d.Wo;
     const int rsc = d.R * d.S * d.C;
     const int all_stages = (rsc + 63) / 64;
-    const int spz = (all_stages + zn - 1) / zn;
     const int zeff = (all_stages + spz - 1) / spz;
     dim3 grid((unsigned)((M + 63) / 64), (unsigned)(d.K / CONV_BN), (unsigned)zeff);
     k_conv_fwd<64, false, true><<<grid, CONV_THREADS, 0, stream>>>(
