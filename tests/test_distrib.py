@@ -137,6 +137,31 @@ def _check_eager_small_buckets(rank: int, ws: int):
         assert torch.allclose(got, ref, atol=1e-6)
 
 
+def _check_flat_optimizer_equivalence(rank: int, ws: int):
+    """FusedSGD + sync_flat_gradients matches torch SGD on the virtual batch
+    (the flat-buffer DP path the bench and solvers use)."""
+    from flashy_amd.optim import FusedSGD
+    model = _make_model()
+    opt = FusedSGD(model.parameters(), lr=0.05, momentum=0.9)
+    x, y = _virtual_batch(ws)
+    xs, ys = x[rank * 4:(rank + 1) * 4], y[rank * 4:(rank + 1) * 4]
+    for _ in range(3):
+        loss = torch.nn.functional.mse_loss(model(xs), ys)
+        opt.zero_grad()
+        loss.backward()
+        distrib.sync_flat_gradients(opt)
+        opt.step()
+    ref = _make_model()
+    ref_opt = torch.optim.SGD(ref.parameters(), lr=0.05, momentum=0.9)
+    for _ in range(3):
+        loss = torch.nn.functional.mse_loss(ref(x), y)
+        ref_opt.zero_grad()
+        loss.backward()
+        ref_opt.step()
+    for p, q in zip(model.parameters(), ref.parameters()):
+        assert torch.allclose(p, q, atol=1e-5), (p - q).abs().max()
+
+
 def _check_broadcast_object(rank: int, ws: int):
     import collections
     if rank == 0:
@@ -194,6 +219,7 @@ ALL_CHECKS = [
     "_check_sync_model_equivalence",
     "_check_eager_sync_equivalence",
     "_check_eager_small_buckets",
+    "_check_flat_optimizer_equivalence",
     "_check_broadcast_object",
     "_check_average_metrics",
     "_check_broadcast_model_and_barrier",
