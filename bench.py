@@ -127,7 +127,8 @@ def main():
 
     native = use_cuda and not args.ref and not args.torch_model
     if native:
-        model = NATIVE_MODELS[args.model](num_classes=args.classes).to(device)
+        model = NATIVE_MODELS[args.model](
+            num_classes=args.classes, imagenet_stem=args.img > 64).to(device)
     else:
         model = MODELS[args.model](num_classes=args.classes,
                                    small_input=args.img <= 64).to(device)

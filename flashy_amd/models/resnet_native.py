@@ -75,12 +75,20 @@ class NativeBottleneck(nn.Module):
 
 
 class NativeResNet(nn.Module):
-    """CIFAR-stem ResNet; input [N, 3, H, W] (NCHW — permuted to NHWC once)."""
+    """ResNet on the native NHWC kernels; input [N, 3, H, W] (NCHW — permuted
+    to NHWC once).  ``imagenet_stem`` uses the 7x7/s2 conv + 3x3/s2 maxpool
+    stem (224-class inputs, BASELINE config 5); default is the CIFAR stem."""
 
     def __init__(self, block: type, layers: tp.Sequence[int],
-                 num_classes: int = 10):
+                 num_classes: int = 10, imagenet_stem: bool = False):
         super().__init__()
-        self.stem_conv = fnn.Conv2d(3, 64, 3, 1, 1, input_grad=False)
+        self.imagenet_stem = imagenet_stem
+        if imagenet_stem:
+            self.stem_conv = fnn.Conv2d(3, 64, 7, 2, 3, input_grad=False)
+            self.stem_pool = fnn.MaxPool2d(3, 2, 1)
+        else:
+            self.stem_conv = fnn.Conv2d(3, 64, 3, 1, 1, input_grad=False)
+            self.stem_pool = None
         self.stem_bn = fnn.BatchNorm2d(64)
         self.inplanes = 64
         self.layer1 = self._make_layer(block, 64, layers[0])
@@ -102,6 +110,8 @@ class NativeResNet(nn.Module):
         if x.dtype != torch.bfloat16:
             x = x.to(torch.bfloat16)
         x = self.stem_bn(self.stem_conv(x), relu=True)
+        if self.stem_pool is not None:
+            x = self.stem_pool(x)
         x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
         x = x.float().mean(dim=(1, 2))      # global average pool (NHWC)
         return self.fc(x)
@@ -141,9 +151,13 @@ class NativeResNet(nn.Module):
         return m
 
 
-def native_resnet18(num_classes: int = 10) -> NativeResNet:
-    return NativeResNet(NativeBasicBlock, [2, 2, 2, 2], num_classes)
+def native_resnet18(num_classes: int = 10,
+                    imagenet_stem: bool = False) -> NativeResNet:
+    return NativeResNet(NativeBasicBlock, [2, 2, 2, 2], num_classes,
+                        imagenet_stem)
 
 
-def native_resnet50(num_classes: int = 10) -> NativeResNet:
-    return NativeResNet(NativeBottleneck, [3, 4, 6, 3], num_classes)
+def native_resnet50(num_classes: int = 10,
+                    imagenet_stem: bool = False) -> NativeResNet:
+    return NativeResNet(NativeBottleneck, [3, 4, 6, 3], num_classes,
+                        imagenet_stem)

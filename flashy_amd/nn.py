@@ -147,6 +147,42 @@ class _BnFn(torch.autograd.Function):
                 None, None)
 
 
+class _MaxPoolFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, kernel: int, stride: int, pad: int):
+        N, H, W, C = x.shape
+        Ho = (H + 2 * pad - kernel) // stride + 1
+        Wo = (W + 2 * pad - kernel) // stride + 1
+        d = ops.ConvDims(N, H, W, C, C, kernel, kernel, Ho, Wo, stride, pad)
+        y = x.new_empty((N, Ho, Wo, C))
+        argmax = torch.empty(N * Ho * Wo * C, dtype=torch.uint8, device=x.device)
+        ops.maxpool_fwd(x, y, argmax, d)
+        ctx.save_for_backward(argmax)
+        ctx.dims = d
+        return y
+
+    @staticmethod
+    def backward(ctx, dy: torch.Tensor):
+        (argmax,) = ctx.saved_tensors
+        d: ops.ConvDims = ctx.dims
+        dx = dy.new_empty((d.N, d.H, d.W, d.C))
+        ops.maxpool_bwd(dy.contiguous(), argmax, dx, d)
+        return dx, None, None, None
+
+
+class MaxPool2d(nn.Module):
+    """NHWC bf16 max pool with deterministic gather-based backward."""
+
+    def __init__(self, kernel_size: int, stride: int, padding: int = 0):
+        super().__init__()
+        self.kernel_size = kernel_size
+        self.stride = stride
+        self.padding = padding
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return _MaxPoolFn.apply(x, self.kernel_size, self.stride, self.padding)
+
+
 class BatchNorm2d(nn.Module):
     """NHWC training BatchNorm with optional fused residual-add + ReLU."""
 

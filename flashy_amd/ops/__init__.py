@@ -95,13 +95,34 @@ def fused_adam(p: torch.Tensor, g: torch.Tensor, m: torch.Tensor,
                v: torch.Tensor, lr: float, beta1: float, beta2: float,
                eps: float, wd: float, step: int, grad_scale: float = 1.0,
                adamw: bool = False,
-               p_bf16: tp.Optional[torch.Tensor] = None) -> None:
+               p_bf16: tp.Optional[torch.Tensor] = None,
+               step_dev: tp.Optional[torch.Tensor] = None) -> None:
+    """step_dev: optional int64 device scalar holding the step count — makes
+    the bias correction graph-replay-safe (advance it with adam_step_inc)."""
     ext = require()
     assert p.is_contiguous() and g.is_contiguous()
     ext.fused_adam(p.data_ptr(), g.data_ptr(), m.data_ptr(), v.data_ptr(),
                    p_bf16.data_ptr() if p_bf16 is not None else 0,
+                   step_dev.data_ptr() if step_dev is not None else 0,
                    p.numel(), lr, beta1, beta2, eps, wd, step, grad_scale,
                    adamw, _stream())
+
+
+def adam_step_inc(step_dev: torch.Tensor) -> None:
+    require().adam_step_inc(step_dev.data_ptr(), _stream())
+
+
+def maxpool_fwd(x: torch.Tensor, y: torch.Tensor, argmax: torch.Tensor,
+                d: "ConvDims") -> None:
+    assert d.C % 8 == 0
+    require().maxpool_fwd(x.data_ptr(), y.data_ptr(), argmax.data_ptr(), *d,
+                          _stream())
+
+
+def maxpool_bwd(dy: torch.Tensor, argmax: torch.Tensor, dx: torch.Tensor,
+                d: "ConvDims") -> None:
+    require().maxpool_bwd(dy.data_ptr(), argmax.data_ptr(), dx.data_ptr(), *d,
+                          _stream())
 
 
 # ---------------------------------------------------------------------------

@@ -61,9 +61,14 @@ void launch_fused_sgd(void* p, const void* g, void* m, void* p_bf16, int64_t n,
                       float lr, float momentum, float wd, float grad_scale,
                       int nesterov, hipStream_t stream);
 void launch_fused_adam(void* p, const void* g, void* m, void* v, void* p_bf16,
-                       int64_t n, float lr, float beta1, float beta2, float eps,
-                       float wd, int64_t step, float grad_scale, int adamw,
-                       hipStream_t stream);
+                       void* step_dev, int64_t n, float lr, float beta1,
+                       float beta2, float eps, float wd, int64_t step,
+                       float grad_scale, int adamw, hipStream_t stream);
+void launch_adam_step_inc(void* p, hipStream_t stream);
+void launch_maxpool_fwd(const void* x, void* y, void* argmax, ConvDims d,
+                        hipStream_t stream);
+void launch_maxpool_bwd(const void* dy, const void* argmax, void* dx,
+                        ConvDims d, hipStream_t stream);
 void launch_cross_entropy(const void* logits, const void* target, void* dlogits,
                           void* loss_sum, int64_t B, int64_t C,
                           float loss_scale, float grad_scale, int is_bf16,
@@ -235,13 +240,36 @@ PYBIND11_MODULE(_hip_ops, m) {
 
     m.def("fused_adam",
           [](uintptr_t p, uintptr_t g, uintptr_t mom, uintptr_t var,
-             uintptr_t p_bf16, int64_t n, float lr, float beta1, float beta2,
-             float eps, float wd, int64_t step, float grad_scale, bool adamw,
-             uintptr_t stream) {
+             uintptr_t p_bf16, uintptr_t step_dev, int64_t n, float lr,
+             float beta1, float beta2, float eps, float wd, int64_t step,
+             float grad_scale, bool adamw, uintptr_t stream) {
               launch_fused_adam((void*)p, (const void*)g, (void*)mom,
-                                (void*)var, (void*)p_bf16, n, lr, beta1, beta2,
-                                eps, wd, step, grad_scale, adamw ? 1 : 0,
-                                as_stream(stream));
+                                (void*)var, (void*)p_bf16, (void*)step_dev, n,
+                                lr, beta1, beta2, eps, wd, step, grad_scale,
+                                adamw ? 1 : 0, as_stream(stream));
+              check_last();
+          });
+    m.def("adam_step_inc", [](uintptr_t p, uintptr_t stream) {
+        launch_adam_step_inc((void*)p, as_stream(stream));
+        check_last();
+    });
+    m.def("maxpool_fwd",
+          [](uintptr_t x, uintptr_t y, uintptr_t argmax, int N, int H, int W,
+             int C, int K, int R, int S, int Ho, int Wo, int stride, int pad,
+             uintptr_t stream) {
+              launch_maxpool_fwd((const void*)x, (void*)y, (void*)argmax,
+                                 make_dims(N, H, W, C, K, R, S, Ho, Wo, stride, pad),
+                                 as_stream(stream));
+              check_last();
+          });
+    m.def("maxpool_bwd",
+          [](uintptr_t dy, uintptr_t argmax, uintptr_t dx, int N, int H, int W,
+             int C, int K, int R, int S, int Ho, int Wo, int stride, int pad,
+             uintptr_t stream) {
+              launch_maxpool_bwd((const void*)dy, (const void*)argmax,
+                                 (void*)dx,
+                                 make_dims(N, H, W, C, K, R, S, Ho, Wo, stride, pad),
+                                 as_stream(stream));
               check_last();
           });
 

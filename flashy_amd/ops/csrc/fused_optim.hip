@@ -105,6 +105,7 @@ struct AdamArgs {
     float* __restrict__ m;
     float* __restrict__ v;
     uint16_t* __restrict__ p_bf16;
+    const long long* step_ptr;  // device step (graph-safe); null -> use bc1/bc2
     int64_t n;
     float lr, beta1, beta2, eps, wd, bc1, bc2, grad_scale;
     int adamw;  // 1: decoupled weight decay, 0: L2 into grad
@@ -123,6 +124,11 @@ __device__ __forceinline__ float adam_one(const AdamArgs& a, float p, float g,
 template <bool HAS_BF16>
 __global__ void __launch_bounds__(256)
 k_fused_adam(AdamArgs a) {
+    if (a.step_ptr != nullptr) {
+        const float t = (float)*a.step_ptr;
+        a.bc1 = 1.f - powf(a.beta1, t);
+        a.bc2 = 1.f - powf(a.beta2, t);
+    }
     const int64_t n4 = a.n / 4;
     const int64_t stride = (int64_t)gridDim.x * blockDim.x;
     float4* p4 = reinterpret_cast<float4*>(a.p);
@@ -159,13 +165,22 @@ k_fused_adam(AdamArgs a) {
     }
 }
 
+__global__ void k_step_inc(long long* p) {
+    if (threadIdx.x == 0 && blockIdx.x == 0) ++(*p);
+}
+
+extern "C" void launch_adam_step_inc(void* p, hipStream_t stream) {
+    k_step_inc<<<1, 1, 0, stream>>>((long long*)p);
+}
+
 extern "C" void launch_fused_adam(void* p, const void* g, void* m, void* v,
-                                  void* p_bf16, int64_t n, float lr,
-                                  float beta1, float beta2, float eps, float wd,
-                                  int64_t step, float grad_scale, int adamw,
-                                  hipStream_t stream) {
+                                  void* p_bf16, void* step_dev, int64_t n,
+                                  float lr, float beta1, float beta2, float eps,
+                                  float wd, int64_t step, float grad_scale,
+                                  int adamw, hipStream_t stream) {
     AdamArgs a{(float*)p, (const float*)g, (float*)m, (float*)v,
-               (uint16_t*)p_bf16, n, lr, beta1, beta2, eps, wd,
+               (uint16_t*)p_bf16, (const long long*)step_dev, n, lr, beta1,
+               beta2, eps, wd,
                1.f - powf(beta1, (float)step), 1.f - powf(beta2, (float)step),
                grad_scale, adamw};
     int grid = ew_grid(n / 4 + 1, 256, 1);

@@ -210,6 +210,14 @@ class FusedAdam(FlatOptimizer):
                                       adamw=adamw), bf16_mirror)
         self._exp_avg = [g.buffers_like() for g in self.groups]
         self._exp_avg_sq = [g.buffers_like() for g in self.groups]
+        # device-side step counter: the kernel reads it for bias correction,
+        # so the whole step is HIP-graph-replay-safe (a captured host step
+        # count would freeze).  Starts at 1 = the value used by step #1.
+        self._step_dev = {}
+        for g in self.groups:
+            if g.device.type == "cuda" and g.device not in self._step_dev:
+                self._step_dev[g.device] = torch.ones(
+                    1, dtype=torch.int64, device=g.device)
 
     def _step_group(self, group: _Group) -> None:
         d = self.defaults
@@ -219,7 +227,8 @@ class FusedAdam(FlatOptimizer):
             ops.fused_adam(group.flat_p, group.flat_g, m, v, d["lr"],
                            d["beta1"], d["beta2"], d["eps"],
                            d["weight_decay"], self.step_count,
-                           adamw=d["adamw"], p_bf16=group.flat_p16)
+                           adamw=d["adamw"], p_bf16=group.flat_p16,
+                           step_dev=self._step_dev[group.device])
             return
         with torch.no_grad():
             g = group.flat_g
@@ -236,6 +245,12 @@ class FusedAdam(FlatOptimizer):
             p.addcdiv_(m / bc1, denom, value=-d["lr"])
             group.refresh_bf16()
 
+    def step(self, closure=None):
+        out = super().step(closure)
+        for dev_step in self._step_dev.values():
+            ops.adam_step_inc(dev_step)
+        return out
+
     def _extra_state(self):
         return {"exp_avg": self._exp_avg, "exp_avg_sq": self._exp_avg_sq}
 
@@ -244,3 +259,5 @@ class FusedAdam(FlatOptimizer):
             mine.copy_(got.to(mine.device))
         for mine, got in zip(self._exp_avg_sq, state.get("exp_avg_sq", [])):
             mine.copy_(got.to(mine.device))
+        for dev_step in self._step_dev.values():
+            dev_step.fill_(self.step_count + 1)
