@@ -178,6 +178,7 @@ def conv_fwd(x: torch.Tensor, w: torch.Tensor, y: torch.Tensor, d: ConvDims,
                          _stream())
     else:
         assert not relu
+        assert d.R * d.S * d.C <= 160 and d.K == 64, d
         ext.conv_stem_fwd(x.data_ptr(), w.data_ptr(), y.data_ptr(), *d, _stream())
 
 
@@ -225,7 +226,7 @@ def conv_wgrad(x: torch.Tensor, dout: torch.Tensor, dw: torch.Tensor,
         ext.conv_wgrad(x.data_ptr(), dout.data_ptr(), dw.data_ptr(), *d,
                        n_splits, _stream())
     else:
-        assert rsc <= 32, d
+        assert rsc <= 160, d  # small-C stem kernels stage taps/weights in LDS
         ext.conv_stem_wgrad(x.data_ptr(), dout.data_ptr(), dw.data_ptr(), *d,
                             _stream())
 

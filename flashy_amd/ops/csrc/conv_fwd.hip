@@ -271,8 +271,9 @@ k_conv_stem_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
                 uint16_t* __restrict__ y, ConvDims d) {
     // weights staged once per block into LDS as fp32 [rsc][K] (K = 64);
     // each thread computes one output pixel x 16 consecutive channels.
-    __shared__ float w_lds[32 * 64];
-    const int rsc = d.R * d.S * d.C;       // <= 32 (3x3x3 = 27)
+    // rsc <= 160 (covers 3x3x3 = 27 and the ImageNet 7x7x3 = 147 stem).
+    __shared__ float w_lds[160 * 64];
+    const int rsc = d.R * d.S * d.C;
     for (int i = threadIdx.x; i < rsc * 64; i += blockDim.x) {
         const int j = i >> 6;              // tap
         const int k = i & 63;              // channel
@@ -326,10 +327,11 @@ extern "C" void launch_conv_stem_fwd(const void* x, const void* w, void* y,
 }
 
 // Stem wgrad: dw[k][rsc] += sum_m dout[m][k] * im2col(x)[m][rsc]
-// grid (K/8 octets, msplit).  Thread = one k of its octet x one of 32
-// m-lanes: dout loads are 16B-coalesced across the octet, the 27 input taps
-// are same-address broadcasts within the octet; LDS fold over m-lanes, then
-// 8x27 atomicAdds per block.  RSC <= 32 (3x3x3 = 27).
+// grid (K/8 octets, tap-chunks of 32, msplit).  Thread = one k of its octet
+// x one of 32 m-lanes; the chunk's (r,s,c) tap decode is staged in LDS once;
+// dout loads are coalesced across the octet, input taps are same-address
+// broadcasts; LDS fold over m-lanes, then 8x32 atomicAdds per block.
+// Any rsc (3x3x3 = 27, 7x7x3 = 147, ...).
 __global__ void __launch_bounds__(256)
 k_conv_stem_wgrad(const uint16_t* __restrict__ x,
                   const uint16_t* __restrict__ dout,
@@ -338,8 +340,22 @@ k_conv_stem_wgrad(const uint16_t* __restrict__ x,
     const int k = blockIdx.x * 8 + kl;
     const int mlane = threadIdx.x >> 3;      // 0..31
     const int rsc = d.R * d.S * d.C;
+    const int j0 = blockIdx.y * 32;          // tap chunk base
+    const int jn = min(32, rsc - j0);
+    __shared__ int tap_r[32], tap_s[32], tap_c[32];
+    if (threadIdx.x < 32) {
+        const int j = j0 + threadIdx.x;
+        if (j < rsc) {
+            tap_r[threadIdx.x] = j / (d.S * d.C);
+            const int sc = j - tap_r[threadIdx.x] * d.S * d.C;
+            tap_s[threadIdx.x] = sc / d.C;
+            tap_c[threadIdx.x] = sc - tap_s[threadIdx.x] * d.C;
+        }
+    }
+    __syncthreads();
+
     const int64_t M = (int64_t)d.N * d.Ho * d.Wo;
-    const int64_t ms = (int64_t)blockIdx.y * m_per_block;
+    const int64_t ms = (int64_t)blockIdx.z * m_per_block;
     const int64_t me = min(ms + (int64_t)m_per_block, M);
     float part[32];
 #pragma unroll
@@ -349,43 +365,44 @@ k_conv_stem_wgrad(const uint16_t* __restrict__ x,
         const int wo = (int)(m % d.Wo);
         const int ho = (int)((m / d.Wo) % d.Ho);
         const int64_t n = m / ((int64_t)d.Ho * d.Wo);
-        for (int r = 0; r < d.R; ++r) {
-            const int hi = ho * d.stride + r - d.pad;
-            if (hi < 0 || hi >= d.H) continue;
-            for (int s = 0; s < d.S; ++s) {
-                const int wi = wo * d.stride + s - d.pad;
-                if (wi < 0 || wi >= d.W) continue;
-                const uint16_t* xp = x + ((n * d.H + hi) * d.W + wi) * d.C;
-                const int base = (r * d.S + s) * d.C;
-                for (int c = 0; c < d.C; ++c)
-                    part[base + c] = fmaf(bf16_to_f32(xp[c]), go, part[base + c]);
-            }
+        const int hb = ho * d.stride - d.pad;
+        const int wb = wo * d.stride - d.pad;
+        for (int j = 0; j < jn; ++j) {
+            const int hi = hb + tap_r[j];
+            const int wi = wb + tap_s[j];
+            if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W)
+                part[j] = fmaf(
+                    bf16_to_f32(x[((n * d.H + hi) * d.W + wi) * (int64_t)d.C +
+                                  tap_c[j]]),
+                    go, part[j]);
         }
     }
-    // fold the 32 m-lanes per k: LDS [8 k][32 lanes][32 taps]
     __shared__ float red[8][32][32];
 #pragma unroll
     for (int j = 0; j < 32; ++j) red[kl][mlane][j] = part[j];
     __syncthreads();
     const int kk = threadIdx.x >> 5;         // 0..7
-    const int j0 = threadIdx.x & 31;         // tap (first 27 valid)
-    if (j0 < rsc) {
+    const int jj = threadIdx.x & 31;
+    if (jj < jn) {
         float acc = 0.f;
 #pragma unroll 8
-        for (int i = 0; i < 32; ++i) acc += red[kk][i][j0];
+        for (int i = 0; i < 32; ++i) acc += red[kk][i][jj];
         if (acc != 0.f)
-            atomicAdd(&dw[(int64_t)(blockIdx.x * 8 + kk) * rsc + j0], acc);
+            atomicAdd(&dw[(int64_t)(blockIdx.x * 8 + kk) * rsc + j0 + jj], acc);
     }
 }
 
 extern "C" void launch_conv_stem_wgrad(const void* x, const void* dout, void* dw,
                                        ConvDims d, hipStream_t stream) {
     const int64_t M = (int64_t)d.N * d.Ho * d.Wo;
+    const int rsc = d.R * d.S * d.C;
+    const int jchunks = (rsc + 31) / 32;
     int64_t msplit = (M + 511) / 512;
-    if (msplit > 128) msplit = 128;
+    const int64_t cap = 256 / jchunks > 0 ? 256 / jchunks : 1;
+    if (msplit > cap) msplit = cap;
     if (msplit < 1) msplit = 1;
     const int m_per_block = (int)((M + msplit - 1) / msplit);
-    dim3 grid((unsigned)(d.K / 8), (unsigned)msplit);
+    dim3 grid((unsigned)(d.K / 8), (unsigned)jchunks, (unsigned)msplit);
     k_conv_stem_wgrad<<<grid, 256, 0, stream>>>(
         (const uint16_t*)x, (const uint16_t*)dout, (float*)dw, d, m_per_block);
 }
