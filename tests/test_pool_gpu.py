@@ -30,9 +30,11 @@ def test_maxpool_fwd_bwd(shape):
     ref_y = ref.permute(0, 2, 3, 1)
     ref_dx = xr.grad.permute(0, 2, 3, 1)
     assert torch.allclose(y.float(), ref_y, atol=1e-2, rtol=1e-2)
-    # ties may pick different argmax between impls; compare sums and most elems
+    # our dx is bf16 (ref is fp32): pixels summing >1 overlapping window's
+    # grad round; single-contribution pixels are bit-exact
     diff = (x.grad.float() - ref_dx).abs()
-    assert (diff > 1e-3).float().mean().item() < 0.01
+    assert (diff > 0.01 * ref_dx.abs() + 0.02).float().mean().item() < 0.005, \
+        diff.max().item()
     assert torch.allclose(x.grad.float().sum(), ref_dx.sum(), rtol=1e-2, atol=1.0)
 
 
