@@ -128,23 +128,23 @@ k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
         ? (all_stages - s0 < stages_per_split ? all_stages - s0 : stages_per_split)
         : all_stages;
     short8 stage[CPT];
-    short8 breg_a[2][NF], breg_b[2][NF];
 
-    // prologue: first stage -> buf0; next -> regs; its B -> breg_a
+    // prologue: first stage -> buf0; next -> regs
     load_stage(stage);
     lds_write(A_lds[0], stage);
     if (n_stages > 1) load_stage(stage);
-    load_b(breg_a, s0);
     __syncthreads();
 
-    // even/odd bodies keep buffer/B-register parity COMPILE-TIME (a runtime
-    // breg[i&1] index lowers to thousands of cndmask selects — rule 20)
-    auto step = [&](int i, const uint16_t* buf, uint16_t* nbuf,
-                    short8 (&bcur)[2][NF], short8 (&bnext)[2][NF]) {
+    // even/odd bodies keep the LDS buffer parity COMPILE-TIME (a runtime
+    // [i&1] register index lowers to thousands of cndmask selects — rule 20).
+    // B fragments load at use: the weight panel is L2-resident (re-read by
+    // every M-block), and the freed 32 VGPRs buy a third wave per SIMD.
+    auto step = [&](int i, const uint16_t* buf, uint16_t* nbuf) {
+        short8 b[2][NF];
+        load_b(b, s0 + i);
         if (i + 1 < n_stages) {
             lds_write(nbuf, stage);
             if (i + 2 < n_stages) load_stage(stage);
-            load_b(bnext, s0 + i + 1);
         }
         const int kc = (s0 + i) * BK2;
 #pragma unroll
@@ -156,15 +156,15 @@ k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
                     &buf[(sub * BM + a_row + mf * 16) * CONV_APITCH + a_koff]);
 #pragma unroll
                 for (int nf = 0; nf < NF; ++nf)
-                    acc[mf][nf] = MFMA_BF16(a, bcur[sub][nf], acc[mf][nf]);
+                    acc[mf][nf] = MFMA_BF16(a, b[sub][nf], acc[mf][nf]);
             }
         }
         __syncthreads();
     };
     for (int i = 0; i < n_stages;) {
-        step(i, A_lds[0], A_lds[1], breg_a, breg_b);
+        step(i, A_lds[0], A_lds[1]);
         if (++i >= n_stages) break;
-        step(i, A_lds[1], A_lds[0], breg_b, breg_a);
+        step(i, A_lds[1], A_lds[0]);
         ++i;
     }
 

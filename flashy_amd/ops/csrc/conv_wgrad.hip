@@ -73,6 +73,12 @@ k_conv_wgrad(const uint16_t* __restrict__ x, const uint16_t* __restrict__ dout,
             xv[sc] = u;
         }
     };
+    // XOR the 8-m column group by a per-row pattern so the 8 lanes sharing
+    // an m_r (channel octets k8 = 0..56) land on distinct banks (was 8-way).
+    auto swz_col = [](int row, int col) {
+        const int g = ((row >> 2) ^ (row >> 3)) & 3;
+        return (col & 7) | ((((col >> 3) ^ g) & 3) << 3);
+    };
     auto stage_write = [&](uint16_t (&dT)[2 * 64 * WG_MP],
                            uint16_t (&xTb)[2 * 64 * WG_MP],
                            const short8* dv, const short8* xv) {
@@ -80,8 +86,10 @@ k_conv_wgrad(const uint16_t* __restrict__ x, const uint16_t* __restrict__ dout,
         for (int sc = 0; sc < 2; ++sc)
 #pragma unroll
             for (int j = 0; j < 8; ++j) {
-                dT[(sc * 64 + k8 + j) * WG_MP + m_r] = ((const uint16_t*)&dv[sc])[j];
-                xTb[(sc * 64 + k8 + j) * WG_MP + m_r] = ((const uint16_t*)&xv[sc])[j];
+                const int row = sc * 64 + k8 + j;
+                const int col = swz_col(k8 + j, m_r);
+                dT[row * WG_MP + col] = ((const uint16_t*)&dv[sc])[j];
+                xTb[row * WG_MP + col] = ((const uint16_t*)&xv[sc])[j];
             }
     };
 
@@ -105,10 +113,14 @@ k_conv_wgrad(const uint16_t* __restrict__ x, const uint16_t* __restrict__ dout,
             short8 a[2], b[2];
 #pragma unroll
             for (int f = 0; f < 2; ++f) {
+                // reads stay contiguous: the swizzle only permutes which
+                // 8-m group sits at moff for this row
                 a[f] = *reinterpret_cast<const short8*>(
-                    &dT[(sc * 64 + frag_row + f * 16) * WG_MP + moff]);
+                    &dT[(sc * 64 + frag_row + f * 16) * WG_MP +
+                        swz_col(frag_row + f * 16, moff)]);
                 b[f] = *reinterpret_cast<const short8*>(
-                    &xTb[(sc * 64 + frag_col + f * 16) * WG_MP + moff]);
+                    &xTb[(sc * 64 + frag_col + f * 16) * WG_MP +
+                         swz_col(frag_col + f * 16, moff)]);
             }
 #pragma unroll
             for (int kf = 0; kf < 2; ++kf)
