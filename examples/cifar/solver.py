@@ -12,6 +12,8 @@ import torch
 from torch.nn import functional as F
 
 from flashy_amd import BaseSolver, Formatter, distrib
+from flashy_amd.functional import cross_entropy
+from flashy_amd.models.resnet_native import NativeResNet
 from flashy_amd.utils import averager
 
 
@@ -42,7 +44,9 @@ class Solver(BaseSolver):
         self.loaders = loaders
         self.optim = optim
         self.device = next(model.parameters()).device
-        self.autocast = self.device.type == "cuda" and cfg.dtype == "bf16"
+        self.native = isinstance(model, NativeResNet)
+        self.autocast = (self.device.type == "cuda" and cfg.dtype == "bf16"
+                         and not self.native)
         self.register_stateful("model", "optim")
 
     def get_formatter(self, stage_name):
@@ -53,12 +57,18 @@ class Solver(BaseSolver):
         label = label.to(self.device, non_blocking=True)
         with torch.autocast("cuda", torch.bfloat16, enabled=self.autocast):
             est = self.model(img)
+        if self.native and train:
+            loss = cross_entropy(est, label)  # fused fwd+grad kernel
+        else:
             loss = F.cross_entropy(est, label)
         acc = (est.argmax(1) == label).float().mean()
         if train:
             self.optim.zero_grad()
             loss.backward()
-            distrib.sync_model(self.model)
+            if hasattr(self.optim, "grad_buffers"):
+                distrib.sync_flat_gradients(self.optim)
+            else:
+                distrib.sync_model(self.model)
             self.optim.step()
         return loss, acc
 

@@ -8,7 +8,8 @@ import torch
 
 import flashy_amd
 from flashy_amd import distrib
-from flashy_amd.models import resnet18
+from flashy_amd.models import native_resnet18, resnet18
+from flashy_amd.optim import FusedSGD
 from flashy_amd import xp as fxp
 
 from .solver import Solver, SyntheticCIFAR
@@ -19,18 +20,28 @@ main = fxp.entry_point("examples.cifar", Path(__file__).parent / "conf")
 def get_solver(cfg):
     device = torch.device("cuda" if torch.cuda.is_available() else "cpu") \
         if cfg.device == "auto" else torch.device(cfg.device)
+    native = device.type == "cuda" and cfg.get("native", True)
+    if native:  # the gfx950 NHWC kernel path + flat fused optimizer
+        model = native_resnet18(cfg.num_classes).to(device)
+        distrib.broadcast_model(model)
+        optim = FusedSGD(model.parameters(), lr=cfg.lr, momentum=cfg.momentum,
+                         weight_decay=cfg.weight_decay, bf16_mirror=True)
+        return Solver(cfg, model, _loaders(cfg), optim)
     model = resnet18(num_classes=cfg.num_classes, small_input=True).to(device)
     distrib.broadcast_model(model)
     optim = torch.optim.SGD(model.parameters(), lr=cfg.lr,
                             momentum=cfg.momentum, weight_decay=cfg.weight_decay)
-    loaders = {
+    return Solver(cfg, model, _loaders(cfg), optim)
+
+
+def _loaders(cfg):
+    return {
         "train": distrib.loader(SyntheticCIFAR(cfg.dataset_size, cfg.num_classes),
                                 batch_size=cfg.batch_size, shuffle=True),
         "valid": distrib.loader(SyntheticCIFAR(cfg.valid_size, cfg.num_classes,
                                                train=False),
                                 batch_size=cfg.batch_size, shuffle=False),
     }
-    return Solver(cfg, model, loaders, optim)
 
 
 def get_solver_from_sig(sig: str):
