@@ -138,6 +138,8 @@ def main():
     if use_cuda and not args.ref:
         optim = FusedSGD(model.parameters(), lr=0.1, momentum=0.9,
                          weight_decay=5e-4, bf16_mirror=native)
+        if native:
+            model.enable_wt_cache()
     else:
         optim = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9,
                                 weight_decay=5e-4)

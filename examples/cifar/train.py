@@ -26,6 +26,7 @@ def get_solver(cfg):
         distrib.broadcast_model(model)
         optim = FusedSGD(model.parameters(), lr=cfg.lr, momentum=cfg.momentum,
                          weight_decay=cfg.weight_decay, bf16_mirror=True)
+        model.enable_wt_cache()
         return Solver(cfg, model, _loaders(cfg), optim)
     model = resnet18(num_classes=cfg.num_classes, small_input=True).to(device)
     distrib.broadcast_model(model)

@@ -96,6 +96,7 @@ class NativeResNet(nn.Module):
         self.layer3 = self._make_layer(block, 256, layers[2], stride=2)
         self.layer4 = self._make_layer(block, 512, layers[3], stride=2)
         self.fc = nn.Linear(512 * block.expansion, num_classes)
+        self._wt_cache = None
 
     def _make_layer(self, block: type, planes: int, n: int, stride: int = 1):
         downsample = stride != 1 or self.inplanes != planes * block.expansion
@@ -104,7 +105,17 @@ class NativeResNet(nn.Module):
         blocks += [block(self.inplanes, planes) for _ in range(1, n)]
         return nn.Sequential(*blocks)
 
+    def enable_wt_cache(self) -> "NativeResNet":
+        """With a bf16-mirror flat optimizer attached: refresh every conv's
+        RSCK weight copy with ONE kernel per step (instead of one transpose
+        launch per conv per backward).  Call after the optimizer exists."""
+        cache = fnn.WtCache(self)
+        self._wt_cache = cache if cache.active else None
+        return self
+
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if self._wt_cache is not None and torch.is_grad_enabled():
+            self._wt_cache.refresh()
         if x.shape[1] == 3:  # NCHW input -> logical NHWC
             x = x.permute(0, 2, 3, 1).contiguous()
         if x.dtype != torch.bfloat16:

@@ -47,7 +47,7 @@ def _grad_target(p: torch.Tensor) -> tp.Tuple[torch.Tensor, bool]:
 class _ConvFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x: torch.Tensor, w: torch.Tensor, stride: int, pad: int,
-                input_grad: bool):
+                input_grad: bool, wt_cached: tp.Optional[torch.Tensor]):
         w16 = _weight_bf16(w)
         d = ops.ConvDims.infer(x, w16, stride, pad)
         y = x.new_empty((d.N, d.Ho, d.Wo, d.K))
@@ -56,6 +56,7 @@ class _ConvFn(torch.autograd.Function):
         ctx.dims = d
         ctx.input_grad = input_grad
         ctx.w_ref = w
+        ctx.wt_cached = wt_cached
         return y
 
     @staticmethod
@@ -67,11 +68,13 @@ class _ConvFn(torch.autograd.Function):
         ops.conv_wgrad(x, dy, dw_buf, d)
         dx = None
         if ctx.input_grad:
-            wt = w16.new_empty((d.R, d.S, d.C, d.K))
-            ops.weight_transpose(w16, wt)
+            wt = ctx.wt_cached  # RSCK copy refreshed once per step (WtCache)
+            if wt is None:
+                wt = w16.new_empty((d.R, d.S, d.C, d.K))
+                ops.weight_transpose(w16, wt)
             dx = x.new_empty(x.shape)
             ops.conv_dgrad(dy, wt, dx, d)
-        return dx, None if direct else dw_buf, None, None, None
+        return dx, None if direct else dw_buf, None, None, None, None
 
 
 class Conv2d(nn.Module):
@@ -83,6 +86,7 @@ class Conv2d(nn.Module):
         self.stride = stride
         self.padding = padding
         self.input_grad = input_grad
+        self._wt_view: tp.Optional[torch.Tensor] = None
         k = kernel_size
         self.weight = nn.Parameter(
             torch.empty(out_channels, k, k, in_channels))
@@ -91,7 +95,58 @@ class Conv2d(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         return _ConvFn.apply(x, self.weight, self.stride, self.padding,
-                             self.input_grad and x.requires_grad)
+                             self.input_grad and x.requires_grad,
+                             self._wt_view)
+
+
+class WtCache:
+    """One arena + ONE batched kernel refreshing every conv's RSCK weight
+    copy per step (replaces a transpose launch per conv per backward).
+
+    Requires the flat optimizer's bf16 mirror (``bf16_mirror=True``): the
+    kernel reads each weight's mirror view straight out of the flat buffer.
+    Call :meth:`refresh` once per step before backward — NativeResNet does it
+    at the top of ``forward``.
+    """
+
+    def __init__(self, model: nn.Module):
+        convs = [m for m in model.modules()
+                 if isinstance(m, Conv2d) and m.input_grad
+                 and m.weight.shape[-1] % 64 == 0]
+        metas, total = [], 0
+        mirrors = [getattr(c.weight, "_bf16_mirror", None) for c in convs]
+        if not convs or any(m is None for m in mirrors):
+            self.meta = None
+            return
+        base = mirrors[0]
+        for c, mir in zip(convs, mirrors):
+            assert mir.untyped_storage().data_ptr() == \
+                base.untyped_storage().data_ptr(), \
+                "all conv weights must share one flat bf16 mirror"
+            src_off = (mir.data_ptr() - base.data_ptr()) // 2
+            assert src_off >= 0
+            K, R, S, C = c.weight.shape
+            metas.append((src_off, total, K, R * S * C))
+            total += K * R * S * C
+        dev = convs[0].weight.device
+        self.src = mirrors[0]
+        self.arena = torch.empty(total, dtype=torch.bfloat16, device=dev)
+        self.meta = torch.tensor([list(m) for m in metas],
+                                 dtype=torch.int32, device=dev).flatten()
+        self.max_elems = max(m[2] * m[3] for m in metas)
+        self.n = len(convs)
+        for c, m in zip(convs, metas):
+            K, R, S, C = c.weight.shape
+            c._wt_view = self.arena[m[1]:m[1] + K * R * S * C].view(R, S, C, K)
+
+    @property
+    def active(self) -> bool:
+        return self.meta is not None
+
+    def refresh(self) -> None:
+        if self.meta is not None:
+            ops.weight_transpose_batched(self.src, self.arena, self.meta,
+                                         self.n, self.max_elems)
 
 
 class _BnFn(torch.autograd.Function):

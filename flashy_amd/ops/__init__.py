@@ -203,6 +203,14 @@ def conv_dgrad(dout: torch.Tensor, w_rsck: torch.Tensor, dx: torch.Tensor,
                        _stream())
 
 
+def weight_transpose_batched(src: torch.Tensor, dst: torch.Tensor,
+                             meta: torch.Tensor, n_convs: int,
+                             max_elems: int) -> None:
+    require().weight_transpose_batched(src.data_ptr(), dst.data_ptr(),
+                                       meta.data_ptr(), n_convs, max_elems,
+                                       _stream())
+
+
 def weight_transpose(w: torch.Tensor, wt: torch.Tensor) -> None:
     ext = require()
     K = w.shape[0]
@@ -239,7 +247,7 @@ def bn_msplit(M: int, C: int) -> int:
     """Blocks along M for the BN reductions: ~256 blocks saturate the chip
     while keeping the partial-combine kernels cheap."""
     cols = max(1, C // 64)
-    msplit = max(1, min(512 // cols, 512))
+    msplit = max(1, min(128 // cols, 128))
     msplit = max(1, min(msplit, (M + 31) // 32))
     if msplit >= 4:
         msplit &= ~3  # multiple of 4: per-channel partial rows stay 16B-aligned

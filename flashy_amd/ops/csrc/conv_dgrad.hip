@@ -250,3 +250,34 @@ extern "C" void launch_weight_transpose(const void* w, void* wt, int K, int rsc,
     k_weight_transpose<<<ew_grid((int64_t)K * rsc, 256, 1), 256, 0, stream>>>(
         (const uint16_t*)w, (uint16_t*)wt, K, rsc);
 }
+
+// Batched transpose of EVERY conv weight in one launch: meta[i] =
+// {src_off, dst_off, K, rsc} (elements), src = the optimizer's flat bf16
+// mirror, dst = the shared RSCK arena.  grid.y = conv index.
+__global__ void __launch_bounds__(256)
+k_weight_transpose_batched(const uint16_t* __restrict__ src_base,
+                           uint16_t* __restrict__ dst_base,
+                           const int* __restrict__ meta) {
+    const int* mi = meta + blockIdx.y * 4;
+    const uint16_t* w = src_base + mi[0];
+    uint16_t* wt = dst_base + mi[1];
+    const int K = mi[2];
+    const int rsc = mi[3];
+    const int64_t total = (int64_t)K * rsc;
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+         i += stride) {
+        const int k = (int)(i / rsc);
+        const int j = (int)(i - (int64_t)k * rsc);
+        wt[(int64_t)j * K + k] = w[i];
+    }
+}
+
+extern "C" void launch_weight_transpose_batched(const void* src, void* dst,
+                                                const void* meta, int n_convs,
+                                                int64_t max_elems,
+                                                hipStream_t stream) {
+    dim3 grid((unsigned)ew_grid(max_elems, 256, 4), (unsigned)n_convs);
+    k_weight_transpose_batched<<<grid, 256, 0, stream>>>(
+        (const uint16_t*)src, (uint16_t*)dst, (const int*)meta);
+}

@@ -37,6 +37,9 @@ void launch_splitk_combine(const void* ws, void* out, int64_t total, int zn,
                            int relu, hipStream_t stream);
 void launch_weight_transpose(const void* w, void* wt, int K, int rsc,
                              hipStream_t stream);
+void launch_weight_transpose_batched(const void* src, void* dst,
+                                     const void* meta, int n_convs,
+                                     int64_t max_elems, hipStream_t stream);
 void launch_conv_wgrad(const void* x, const void* dout, void* dw, ConvDims d,
                        int n_splits, hipStream_t stream);
 void launch_bn_stats(const void* x, void* partials, int64_t M, int C,
@@ -164,6 +167,14 @@ PYBIND11_MODULE(_hip_ops, m) {
           [](uintptr_t w, uintptr_t wt, int K, int rsc, uintptr_t stream) {
               launch_weight_transpose((const void*)w, (void*)wt, K, rsc,
                                       as_stream(stream));
+              check_last();
+          });
+    m.def("weight_transpose_batched",
+          [](uintptr_t src, uintptr_t dst, uintptr_t meta, int n_convs,
+             int64_t max_elems, uintptr_t stream) {
+              launch_weight_transpose_batched((const void*)src, (void*)dst,
+                                              (const void*)meta, n_convs,
+                                              max_elems, as_stream(stream));
               check_last();
           });
     m.def("conv_wgrad",
