@@ -14,7 +14,7 @@
 
 #include "conv_common.h"
 
-template <int BM, bool SPLITK>
+template <int BM, bool SPLITK, bool S1>
 __global__ void __launch_bounds__(CONV_THREADS)
 k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_rsck,
              uint16_t* __restrict__ dx, float* __restrict__ ws_out,
@@ -70,10 +70,11 @@ k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_r
             if (st_n[t] >= 0 && st_r[t] < d.R) {
                 const int hnum = st_hi[t] - st_r[t];  // = ho * stride
                 const int wnum = st_wi[t] - st_s[t];
-                const int ho = hnum / d.stride;
-                const int wo = wnum / d.stride;
-                if (hnum >= 0 && wnum >= 0 && ho * d.stride == hnum &&
-                    wo * d.stride == wnum && ho < d.Ho && wo < d.Wo)
+                const int ho = S1 ? hnum : hnum / d.stride;
+                const int wo = S1 ? wnum : wnum / d.stride;
+                if (hnum >= 0 && wnum >= 0 &&
+                    (S1 || (ho * d.stride == hnum && wo * d.stride == wnum)) &&
+                    ho < d.Ho && wo < d.Wo)
                     v = *reinterpret_cast<const short8*>(
                         dout + (((st_n[t] * d.Ho + ho) * d.Wo + wo) * (int64_t)d.K +
                                 st_k[t]));
@@ -210,9 +211,15 @@ extern "C" void launch_conv_dgrad(const void* dout, const void* w_rsck,
     auto dd = (const uint16_t*)dout;
     auto ww = (const uint16_t*)w_rsck;
     auto xx = (uint16_t*)dx;
-    if (bm == 128) k_conv_dgrad<128, false><<<grid, CONV_THREADS, 0, stream>>>(dd, ww, xx, nullptr, d, 0);
-    else if (bm == 64) k_conv_dgrad<64, false><<<grid, CONV_THREADS, 0, stream>>>(dd, ww, xx, nullptr, d, 0);
-    else k_conv_dgrad<32, false><<<grid, CONV_THREADS, 0, stream>>>(dd, ww, xx, nullptr, d, 0);
+    if (d.stride == 1) {
+        if (bm == 128) k_conv_dgrad<128, false, true><<<grid, CONV_THREADS, 0, stream>>>(dd, ww, xx, nullptr, d, 0);
+        else if (bm == 64) k_conv_dgrad<64, false, true><<<grid, CONV_THREADS, 0, stream>>>(dd, ww, xx, nullptr, d, 0);
+        else k_conv_dgrad<32, false, true><<<grid, CONV_THREADS, 0, stream>>>(dd, ww, xx, nullptr, d, 0);
+    } else {
+        if (bm == 128) k_conv_dgrad<128, false, false><<<grid, CONV_THREADS, 0, stream>>>(dd, ww, xx, nullptr, d, 0);
+        else if (bm == 64) k_conv_dgrad<64, false, false><<<grid, CONV_THREADS, 0, stream>>>(dd, ww, xx, nullptr, d, 0);
+        else k_conv_dgrad<32, false, false><<<grid, CONV_THREADS, 0, stream>>>(dd, ww, xx, nullptr, d, 0);
+    }
 }
 
 extern "C" void launch_conv_dgrad_splitk(const void* dout, const void* w_rsck,
@@ -223,9 +230,14 @@ extern "C" void launch_conv_dgrad_splitk(const void* dout, const void* w_rsck,
     const int all_stages = (rsk + 63) / 64;
     const int zeff = (all_stages + spz - 1) / spz;
     dim3 grid((unsigned)((M + 63) / 64), (unsigned)(d.C / CONV_BN), (unsigned)zeff);
-    k_conv_dgrad<64, true><<<grid, CONV_THREADS, 0, stream>>>(
-        (const uint16_t*)dout, (const uint16_t*)w_rsck, nullptr, (float*)ws,
-        d, spz);
+    if (d.stride == 1)
+        k_conv_dgrad<64, true, true><<<grid, CONV_THREADS, 0, stream>>>(
+            (const uint16_t*)dout, (const uint16_t*)w_rsck, nullptr, (float*)ws,
+            d, spz);
+    else
+        k_conv_dgrad<64, true, false><<<grid, CONV_THREADS, 0, stream>>>(
+            (const uint16_t*)dout, (const uint16_t*)w_rsck, nullptr, (float*)ws,
+            d, spz);
 }
 
 // ---------------------------------------------------------------------------
