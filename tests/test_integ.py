@@ -47,3 +47,25 @@ def test_local_ddp_two_workers(tmp_path):
     _run(tmp_path, "--clear", "-d", "--workers", "2", "epochs=2")
     hist = _history(tmp_path)
     assert len(hist) == 2
+
+
+def test_bench_distributed_contract(tmp_path):
+    """bench.py under torch.distributed.run exactly as the driver launches
+    it (one JSON line from rank 0, value aggregated over the world)."""
+    env = dict(os.environ)
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29517", "bench.py", "--gpus", "2", "--steps", "2",
+         "--warmup", "1", "--no-ckpt", "--batch", "8"],
+        cwd=REPO, env=env, capture_output=True, text=True, timeout=600)
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [ln for ln in out.stdout.splitlines() if ln.startswith("{")]
+    assert len(lines) == 1, out.stdout
+    result = json.loads(lines[0])
+    assert result["n_gpus"] == 2
+    assert result["config"]["parallelism"] == "dp2"
+    assert result["config"]["global_batch"] == 16
+    assert result["value"] > 0
