@@ -13,6 +13,7 @@ from torch.nn import functional as F
 
 from flashy_amd import BaseSolver, Formatter, distrib
 from flashy_amd.functional import cross_entropy
+from flashy_amd.graph import CapturedStep
 from flashy_amd.models.resnet_native import NativeResNet
 from flashy_amd.utils import averager
 
@@ -47,12 +48,44 @@ class Solver(BaseSolver):
         self.native = isinstance(model, NativeResNet)
         self.autocast = (self.device.type == "cuda" and cfg.dtype == "bf16"
                          and not self.native)
+        # HIP-graph whole-step capture: native path, single process only
+        self.use_graph = (cfg.get("use_graph", True) and self.native
+                          and not distrib.is_distributed())
+        self._graph = None
         self.register_stateful("model", "optim")
 
     def get_formatter(self, stage_name):
         return Formatter({"acc": ".1%", "loss": ".5f"})
 
+    def _capture(self, shape):
+        self._static_img = torch.zeros(shape, device=self.device)
+        self._static_label = torch.zeros(shape[0], dtype=torch.long,
+                                         device=self.device)
+
+        def step():
+            self.optim.zero_grad(set_to_none=False)
+            est = self.model(self._static_img)
+            loss = cross_entropy(est, self._static_label)
+            loss.backward()
+            self.optim.step()
+            return loss, est
+
+        self._graph = CapturedStep(step, warmup=3).capture()
+
+    def _graphed_step(self, img, label):
+        """One training step as a single hipGraph replay (static buffers)."""
+        if self._graph is None:
+            self._capture(img.shape)
+        self._static_img.copy_(img, non_blocking=True)
+        self._static_label.copy_(label, non_blocking=True)
+        loss, est = self._graph()
+        acc = (est.argmax(1) == self._static_label).float().mean()
+        return loss, acc
+
     def _step(self, img, label, train: bool):
+        if (train and self.use_graph and self.device.type == "cuda"
+                and img.shape[0] == self.cfg.batch_size):
+            return self._graphed_step(img, label)
         img = img.to(self.device, non_blocking=True)
         label = label.to(self.device, non_blocking=True)
         with torch.autocast("cuda", torch.bfloat16, enabled=self.autocast):
