@@ -11,3 +11,5 @@ from .resnet import ResNet, resnet18, resnet50  # noqa: F401
 from .resnet_native import (NativeResNet, native_resnet18,  # noqa: F401
                             native_resnet50)
 from .dcgan import DCGANGenerator, DCGANDiscriminator  # noqa: F401
+from .dcgan_native import (NativeDCGANGenerator,  # noqa: F401
+                           NativeDCGANDiscriminator)

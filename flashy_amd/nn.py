@@ -68,12 +68,15 @@ class _ConvFn(torch.autograd.Function):
         ops.conv_wgrad(x, dy, dw_buf, d)
         dx = None
         if ctx.input_grad:
-            wt = ctx.wt_cached  # RSCK copy refreshed once per step (WtCache)
-            if wt is None:
-                wt = w16.new_empty((d.R, d.S, d.C, d.K))
-                ops.weight_transpose(w16, wt)
             dx = x.new_empty(x.shape)
-            ops.conv_dgrad(dy, wt, dx, d)
+            if d.C % 64 == 0:
+                wt = ctx.wt_cached  # RSCK copy refreshed per step (WtCache)
+                if wt is None:
+                    wt = w16.new_empty((d.R, d.S, d.C, d.K))
+                    ops.weight_transpose(w16, wt)
+                ops.conv_dgrad(dy, wt, dx, d)
+            else:  # small-C edge conv (e.g. a discriminator RGB stem)
+                ops.conv_stem_dgrad(dy, w16, dx, d)
         return dx, None if direct else dw_buf, None, None, None, None
 
 
@@ -152,7 +155,8 @@ class WtCache:
 class _BnFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
-                res: tp.Optional[torch.Tensor], relu: bool, module):
+                res: tp.Optional[torch.Tensor], relu: bool, slope: float,
+                module):
         N, H, W, C = x.shape
         M = N * H * W
         y = torch.empty_like(x)
@@ -170,9 +174,10 @@ class _BnFn(torch.autograd.Function):
             scale = gamma.detach() * invstd
             shift = beta.detach() - module.running_mean * scale
             work = torch.cat([module.running_mean, invstd, scale, shift])
-        ops.bn_apply(x, res, y, work, M, C, relu)
+        ops.bn_apply(x, res, y, work, M, C, relu, slope)
         ctx.save_for_backward(x, y, work)
         ctx.relu = relu
+        ctx.slope = slope
         ctx.has_res = res is not None
         ctx.refs = (gamma, beta)
         return y
@@ -189,7 +194,8 @@ class _BnFn(torch.autograd.Function):
                                device=x.device)
         bsums = torch.empty(2 * C, dtype=torch.float32, device=x.device)
         dz = torch.empty_like(dy)
-        ops.bn_bwd_reduce(dy, y, x, work, dz, partials, M, C, msplit, ctx.relu)
+        ops.bn_bwd_reduce(dy, y, x, work, dz, partials, M, C, msplit,
+                          ctx.relu, ctx.slope)
         dgamma, g_direct = _grad_target(gamma)
         dbeta, b_direct = _grad_target(beta)
         ops.bn_bwd_grads(partials, msplit, bsums, dgamma, dbeta, C)
@@ -199,7 +205,75 @@ class _BnFn(torch.autograd.Function):
                 None if g_direct else dgamma,
                 None if b_direct else dbeta,
                 dz if ctx.has_res else None,
-                None, None)
+                None, None, None)
+
+
+class _ConvTransposeFn(torch.autograd.Function):
+    """Transposed conv via the equivalent regular conv C (weight [K=C_in]
+    [R][S][C=C_out]):  fwd = C's dgrad,  dx = C's fwd,  dw = C's wgrad with
+    the operand roles swapped (dL/dw = wgrad(x=dy, dout=x))."""
+
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, w: torch.Tensor, stride: int, pad: int,
+                input_grad: bool):
+        w16 = _weight_bf16(w)
+        N, Hi, Wi, Cin = x.shape
+        K, R, S, Cout = w16.shape
+        assert K == Cin, (x.shape, w16.shape)
+        Ho = (Hi - 1) * stride - 2 * pad + R
+        Wo = (Wi - 1) * stride - 2 * pad + S
+        # equivalent-conv dims: "input" = our OUTPUT, "output" = our input
+        d = ops.ConvDims(N, Ho, Wo, Cout, Cin, R, S, Hi, Wi, stride, pad)
+        y = x.new_empty((N, Ho, Wo, Cout))
+        if Cout % 64 == 0:
+            wt = w16.new_empty((R, S, Cout, Cin))
+            ops.weight_transpose(w16, wt)
+            ops.conv_dgrad(x, wt, y, d)
+        else:  # RGB head: small-C direct kernel, KRSC weights as stored
+            ops.conv_stem_dgrad(x, w16, y, d)
+        ctx.save_for_backward(x, w16)
+        ctx.dims = d
+        ctx.input_grad = input_grad
+        ctx.w_ref = w
+        return y
+
+    @staticmethod
+    def backward(ctx, dy: torch.Tensor):
+        x, w16 = ctx.saved_tensors
+        d: ops.ConvDims = ctx.dims
+        dy = dy.contiguous()
+        dw_buf, direct = _grad_target(ctx.w_ref)
+        ops.conv_wgrad(dy, x, dw_buf, d)   # roles swapped vs regular conv
+        dx = None
+        if ctx.input_grad:
+            dx = x.new_empty(x.shape)
+            ops.conv_fwd(dy, w16, dx, d)
+        return dx, None if direct else dw_buf, None, None, None
+
+
+class ConvTranspose2d(nn.Module):
+    """NHWC bf16 transposed conv.  Weight [C_in, R, S, C_out] fp32
+    (= the equivalent regular conv's [K][R][S][C]).
+    Trunk layers need C_in % 64 == 0 and C_out % 64 == 0; a small-C output
+    head (e.g. RGB, C_out <= 8 with C_in == 64) uses the direct edge
+    kernels."""
+
+    def __init__(self, in_channels: int, out_channels: int, kernel_size: int,
+                 stride: int = 1, padding: int = 0, input_grad: bool = True):
+        super().__init__()
+        self.stride = stride
+        self.padding = padding
+        self.input_grad = input_grad
+        k = kernel_size
+        self.weight = nn.Parameter(
+            torch.empty(in_channels, k, k, out_channels))
+        fan_out = out_channels * k * k
+        nn.init.normal_(self.weight, std=math.sqrt(2.0 / fan_out))
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return _ConvTransposeFn.apply(x, self.weight, self.stride,
+                                      self.padding,
+                                      self.input_grad and x.requires_grad)
 
 
 class _MaxPoolFn(torch.autograd.Function):
@@ -253,5 +327,6 @@ class BatchNorm2d(nn.Module):
 
     def forward(self, x: torch.Tensor,
                 res: tp.Optional[torch.Tensor] = None,
-                relu: bool = False) -> torch.Tensor:
-        return _BnFn.apply(x, self.weight, self.bias, res, relu, self)
+                relu: bool = False, slope: float = 0.0) -> torch.Tensor:
+        """``relu=True, slope=s`` applies LeakyReLU(s) (s=0: plain ReLU)."""
+        return _BnFn.apply(x, self.weight, self.bias, res, relu, slope, self)

@@ -182,6 +182,14 @@ def conv_fwd(x: torch.Tensor, w: torch.Tensor, y: torch.Tensor, d: ConvDims,
         ext.conv_stem_fwd(x.data_ptr(), w.data_ptr(), y.data_ptr(), *d, _stream())
 
 
+def conv_stem_dgrad(dout: torch.Tensor, w_krsc: torch.Tensor,
+                    dx: torch.Tensor, d: ConvDims) -> None:
+    """dX of a small-C edge conv (C <= 8, K == 64); weights in KRSC layout."""
+    assert d.C <= 8 and d.K == 64 and d.R * d.S * d.C <= 160, d
+    require().conv_stem_dgrad(dout.data_ptr(), w_krsc.data_ptr(),
+                              dx.data_ptr(), *d, _stream())
+
+
 def conv_dgrad(dout: torch.Tensor, w_rsck: torch.Tensor, dx: torch.Tensor,
                d: ConvDims) -> None:
     ext = require()
@@ -268,16 +276,19 @@ def bn_finalize(partials, msplit: int, gamma, beta, rmean, rvar, work,
                           update_running, _stream())
 
 
-def bn_apply(x, res, y, work, M: int, C: int, relu: bool) -> None:
+def bn_apply(x, res, y, work, M: int, C: int, relu: bool,
+             slope: float = 0.0) -> None:
     require().bn_apply(x.data_ptr(), res.data_ptr() if res is not None else 0,
-                       y.data_ptr(), work.data_ptr(), M, C, relu, _stream())
+                       y.data_ptr(), work.data_ptr(), M, C, relu, slope,
+                       _stream())
 
 
 def bn_bwd_reduce(dy, y, x, work, dz_out, partials, M: int, C: int,
-                  msplit: int, relu: bool) -> None:
+                  msplit: int, relu: bool, slope: float = 0.0) -> None:
     require().bn_bwd_reduce(dy.data_ptr(), y.data_ptr(), x.data_ptr(),
                             work.data_ptr(), dz_out.data_ptr(),
-                            partials.data_ptr(), M, C, msplit, relu, _stream())
+                            partials.data_ptr(), M, C, msplit, relu, slope,
+                            _stream())
 
 
 def bn_bwd_grads(partials, msplit: int, bsums, dgamma, dbeta, C: int) -> None:

@@ -134,7 +134,7 @@ template <bool RELU, bool RES>
 __global__ void __launch_bounds__(256)
 k_bn_apply(const uint16_t* __restrict__ x, const uint16_t* __restrict__ res,
            uint16_t* __restrict__ y, const float* __restrict__ work,
-           int64_t M, int C) {
+           int64_t M, int C, float slope) {
     const int64_t total8 = M * C / 8;
     const int64_t stride = (int64_t)gridDim.x * blockDim.x;
     const float4* scale4 = reinterpret_cast<const float4*>(work + 2 * (int64_t)C);
@@ -153,7 +153,7 @@ k_bn_apply(const uint16_t* __restrict__ x, const uint16_t* __restrict__ res,
             float v = fmaf(bf16_to_f32(((const uint16_t*)&xv)[j]),
                            ((const float*)sc)[j], ((const float*)sh)[j]);
             if (RES) v += bf16_to_f32(((const uint16_t*)&rv)[j]);
-            if (RELU) v = fmaxf(v, 0.f);
+            if (RELU) v = v > 0.f ? v : slope * v;  // slope 0 = plain ReLU
             ((uint16_t*)&out)[j] = f32_to_bf16(v);
         }
         *reinterpret_cast<short8*>(y + i * 8) = out;
@@ -170,7 +170,7 @@ __global__ void __launch_bounds__(256)
 k_bn_bwd_reduce(const uint16_t* __restrict__ dy, const uint16_t* __restrict__ y,
                 const uint16_t* __restrict__ x, const float* __restrict__ work,
                 uint16_t* __restrict__ dz_out, float* __restrict__ partials,
-                int64_t M, int C, int msplit, int m_per_block) {
+                int64_t M, int C, int msplit, int m_per_block, float slope) {
     const int c8 = (threadIdx.x & 7) * 8;
     const int mlane = threadIdx.x >> 3;
     const int cbase = blockIdx.x * 64;
@@ -193,7 +193,8 @@ k_bn_bwd_reduce(const uint16_t* __restrict__ dy, const uint16_t* __restrict__ y,
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
             float g = bf16_to_f32(((const uint16_t*)&g8)[j]);
-            if (RELU && bf16_to_f32(((const uint16_t*)&y8)[j]) <= 0.f) g = 0.f;
+            if (RELU && bf16_to_f32(((const uint16_t*)&y8)[j]) <= 0.f)
+                g *= slope;
             ((uint16_t*)&dz8)[j] = f32_to_bf16(g);
             s[j] += g;
             const float xh = (bf16_to_f32(((const uint16_t*)&x8)[j]) - mean[j]) * invstd[j];
@@ -300,28 +301,28 @@ extern "C" void launch_bn_finalize(const void* partials, int msplit,
 
 extern "C" void launch_bn_apply(const void* x, const void* res, void* y,
                                 const void* work, int64_t M, int C, int relu,
-                                hipStream_t stream) {
+                                float slope, hipStream_t stream) {
     const int grid = ew_grid(M * C / 8, 256, 4);
     if (relu && res)
         k_bn_apply<true, true><<<grid, 256, 0, stream>>>(
             (const uint16_t*)x, (const uint16_t*)res, (uint16_t*)y,
-            (const float*)work, M, C);
+            (const float*)work, M, C, slope);
     else if (relu)
         k_bn_apply<true, false><<<grid, 256, 0, stream>>>(
-            (const uint16_t*)x, nullptr, (uint16_t*)y, (const float*)work, M, C);
+            (const uint16_t*)x, nullptr, (uint16_t*)y, (const float*)work, M, C, slope);
     else if (res)
         k_bn_apply<false, true><<<grid, 256, 0, stream>>>(
             (const uint16_t*)x, (const uint16_t*)res, (uint16_t*)y,
-            (const float*)work, M, C);
+            (const float*)work, M, C, slope);
     else
         k_bn_apply<false, false><<<grid, 256, 0, stream>>>(
-            (const uint16_t*)x, nullptr, (uint16_t*)y, (const float*)work, M, C);
+            (const uint16_t*)x, nullptr, (uint16_t*)y, (const float*)work, M, C, slope);
 }
 
 extern "C" void launch_bn_bwd_reduce(const void* dy, const void* y,
                                      const void* x, const void* work,
                                      void* dz_out, void* partials, int64_t M,
-                                     int C, int msplit, int relu,
+                                     int C, int msplit, int relu, float slope,
                                      hipStream_t stream) {
     const int mpb = (int)((M + msplit - 1) / msplit);
     dim3 grid((unsigned)(C / 64), (unsigned)msplit);
@@ -329,12 +330,12 @@ extern "C" void launch_bn_bwd_reduce(const void* dy, const void* y,
         k_bn_bwd_reduce<true><<<grid, 256, 0, stream>>>(
             (const uint16_t*)dy, (const uint16_t*)y, (const uint16_t*)x,
             (const float*)work, (uint16_t*)dz_out, (float*)partials, M, C,
-            msplit, mpb);
+            msplit, mpb, slope);
     else
         k_bn_bwd_reduce<false><<<grid, 256, 0, stream>>>(
             (const uint16_t*)dy, (const uint16_t*)y, (const uint16_t*)x,
             (const float*)work, (uint16_t*)dz_out, (float*)partials, M, C,
-            msplit, mpb);
+            msplit, mpb, slope);
 }
 
 extern "C" void launch_bn_bwd_grads(const void* partials, int msplit,

@@ -406,3 +406,58 @@ extern "C" void launch_conv_stem_wgrad(const void* x, const void* dout, void* dw
     k_conv_stem_wgrad<<<grid, 256, 0, stream>>>(
         (const uint16_t*)x, (const uint16_t*)dout, (float*)dw, d, m_per_block);
 }
+
+// Stem dgrad (small-C edge layers: the RGB ends of the DCGAN generator /
+// discriminator).  Computes the equivalent conv's dX for C <= 8, K == 64:
+//   dx[n,hi,wi,c] = sum over valid (r,s):  dout[n,ho,wo,k] * w[k,r,s,c]
+// with ho = (hi+pad-r)/stride when exact.  Weights staged to LDS as fp32.
+__global__ void __launch_bounds__(256)
+k_conv_stem_dgrad(const uint16_t* __restrict__ dout,
+                  const uint16_t* __restrict__ w,
+                  uint16_t* __restrict__ dx, ConvDims d) {
+    __shared__ float w_lds[64 * 160];
+    const int rsc = d.R * d.S * d.C;
+    for (int i = threadIdx.x; i < 64 * rsc; i += blockDim.x)
+        w_lds[i] = bf16_to_f32(w[i]);   // [k][rsc] as stored
+    __syncthreads();
+
+    const int64_t total = (int64_t)d.N * d.H * d.W;
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t m = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         m < total; m += stride) {
+        const int wi = (int)(m % d.W);
+        const int hi = (int)((m / d.W) % d.H);
+        const int64_t n = m / ((int64_t)d.H * d.W);
+        float acc[8] = {};
+        for (int r = 0; r < d.R; ++r) {
+            const int hnum = hi + d.pad - r;
+            if (hnum < 0 || hnum % d.stride) continue;
+            const int ho = hnum / d.stride;
+            if (ho >= d.Ho) continue;
+            for (int s = 0; s < d.S; ++s) {
+                const int wnum = wi + d.pad - s;
+                if (wnum < 0 || wnum % d.stride) continue;
+                const int wo = wnum / d.stride;
+                if (wo >= d.Wo) continue;
+                const uint16_t* gp =
+                    dout + ((n * d.Ho + ho) * d.Wo + wo) * (int64_t)d.K;
+                const int base = (r * d.S + s) * d.C;
+                for (int k = 0; k < 64; ++k) {
+                    const float g = bf16_to_f32(gp[k]);
+                    for (int c = 0; c < d.C; ++c)
+                        acc[c] = fmaf(g, w_lds[k * rsc + base + c], acc[c]);
+                }
+            }
+        }
+        for (int c = 0; c < d.C; ++c)
+            dx[m * d.C + c] = f32_to_bf16(acc[c]);
+    }
+}
+
+extern "C" void launch_conv_stem_dgrad(const void* dout, const void* w,
+                                       void* dx, ConvDims d,
+                                       hipStream_t stream) {
+    const int64_t total = (int64_t)d.N * d.H * d.W;
+    k_conv_stem_dgrad<<<ew_grid(total, 256, 1), 256, 0, stream>>>(
+        (const uint16_t*)dout, (const uint16_t*)w, (uint16_t*)dx, d);
+}

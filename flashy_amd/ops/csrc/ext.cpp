@@ -27,6 +27,8 @@ void launch_conv_stem_fwd(const void* x, const void* w, void* y, ConvDims d,
                           hipStream_t stream);
 void launch_conv_stem_wgrad(const void* x, const void* dout, void* dw,
                             ConvDims d, hipStream_t stream);
+void launch_conv_stem_dgrad(const void* dout, const void* w, void* dx,
+                            ConvDims d, hipStream_t stream);
 void launch_conv_dgrad(const void* dout, const void* w_rsck, void* dx,
                        ConvDims d, hipStream_t stream);
 void launch_conv_fwd_splitk(const void* x, const void* w, void* ws, ConvDims d,
@@ -50,10 +52,11 @@ void launch_bn_finalize(const void* partials, int msplit, const void* gamma,
                         float eps, float momentum, int update_running,
                         hipStream_t stream);
 void launch_bn_apply(const void* x, const void* res, void* y, const void* work,
-                     int64_t M, int C, int relu, hipStream_t stream);
+                     int64_t M, int C, int relu, float slope,
+                     hipStream_t stream);
 void launch_bn_bwd_reduce(const void* dy, const void* y, const void* x,
                           const void* work, void* dz_out, void* partials,
-                          int64_t M, int C, int msplit, int relu,
+                          int64_t M, int C, int msplit, int relu, float slope,
                           hipStream_t stream);
 void launch_bn_bwd_grads(const void* partials, int msplit, void* bsums,
                          void* dgamma, void* dbeta, int C, hipStream_t stream);
@@ -123,6 +126,16 @@ PYBIND11_MODULE(_hip_ops, m) {
              uintptr_t stream) {
               launch_conv_stem_wgrad((const void*)x, (const void*)dout,
                                      (void*)dw,
+                                     make_dims(N, H, W, C, K, R, S, Ho, Wo, stride, pad),
+                                     as_stream(stream));
+              check_last();
+          });
+    m.def("conv_stem_dgrad",
+          [](uintptr_t dout, uintptr_t w, uintptr_t dx, int N, int H, int W,
+             int C, int K, int R, int S, int Ho, int Wo, int stride, int pad,
+             uintptr_t stream) {
+              launch_conv_stem_dgrad((const void*)dout, (const void*)w,
+                                     (void*)dx,
                                      make_dims(N, H, W, C, K, R, S, Ho, Wo, stride, pad),
                                      as_stream(stream));
               check_last();
@@ -206,20 +219,21 @@ PYBIND11_MODULE(_hip_ops, m) {
           });
     m.def("bn_apply",
           [](uintptr_t x, uintptr_t res, uintptr_t y, uintptr_t work, int64_t M,
-             int C, bool relu, uintptr_t stream) {
+             int C, bool relu, float slope, uintptr_t stream) {
               launch_bn_apply((const void*)x, (const void*)res, (void*)y,
-                              (const void*)work, M, C, relu ? 1 : 0,
+                              (const void*)work, M, C, relu ? 1 : 0, slope,
                               as_stream(stream));
               check_last();
           });
     m.def("bn_bwd_reduce",
           [](uintptr_t dy, uintptr_t y, uintptr_t x, uintptr_t work,
              uintptr_t dz_out, uintptr_t partials, int64_t M, int C,
-             int msplit, bool relu, uintptr_t stream) {
+             int msplit, bool relu, float slope, uintptr_t stream) {
               launch_bn_bwd_reduce((const void*)dy, (const void*)y,
                                    (const void*)x, (const void*)work,
                                    (void*)dz_out, (void*)partials, M, C,
-                                   msplit, relu ? 1 : 0, as_stream(stream));
+                                   msplit, relu ? 1 : 0, slope,
+                                   as_stream(stream));
               check_last();
           });
     m.def("bn_bwd_grads",

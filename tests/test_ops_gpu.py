@@ -149,7 +149,7 @@ def test_solver_end_to_end_gpu(tmp_path, monkeypatch):
         "epochs": 1, "lr": 0.1, "momentum": 0.9, "weight_decay": 5e-4,
         "batch_size": 32, "dataset_size": 256, "valid_size": 64,
         "num_classes": 10, "device": "auto", "dtype": "bf16",
-        "use_graph": False, "seed": 0, "run": {"exclude": []}})
+        "use_graph": True, "seed": 0, "run": {"exclude": []}})
     fxp.create_xp(cfg).enter()
     solver = get_solver(cfg)
     solver.run()
