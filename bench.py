@@ -114,7 +114,12 @@ def main():
     parser.add_argument("--no-ckpt", action="store_true",
                         help="skip the checkpoint save/restore measurement")
     parser.add_argument("--channels-last", action="store_true")
+    parser.add_argument("--workload", default="cifar", choices=["cifar", "gan"],
+                        help="gan = DCGAN-style G/D on 64x64 synthetic images"
+                             " (BASELINE config 4)")
     args = parser.parse_args()
+    if args.workload == "gan":
+        return main_gan(args)
 
     distrib.init()
     ws = distrib.world_size()
@@ -240,6 +245,85 @@ def main():
             },
         }
         print(json.dumps(result))
+
+
+def main_gan(args):
+    """Adversarial G/D step benchmark (BASELINE config 4): native DCGAN
+    kernels + flat fused Adam on GPU; one G update + one D update per step."""
+    from flashy_amd.adversarial import AdversarialLoss
+    from flashy_amd.models import (DCGANDiscriminator, DCGANGenerator,
+                                   NativeDCGANDiscriminator,
+                                   NativeDCGANGenerator)
+    from flashy_amd.optim import FusedAdam
+
+    distrib.init()
+    ws = distrib.world_size()
+    rank = distrib.rank()
+    use_cuda = torch.cuda.is_available()
+    device = torch.device("cuda", torch.cuda.current_device()) if use_cuda \
+        else torch.device("cpu")
+    torch.manual_seed(1234 + rank)
+    nz = 128
+    native = use_cuda and not args.ref
+    if native:
+        gen = NativeDCGANGenerator(nz).to(device)
+        disc = NativeDCGANDiscriminator().to(device)
+        g_optim = FusedAdam(gen.parameters(), lr=2e-4, betas=(0.5, 0.999))
+        d_optim = FusedAdam(disc.parameters(), lr=2e-4, betas=(0.5, 0.999))
+    else:
+        gen = DCGANGenerator(nz).to(device)
+        disc = DCGANDiscriminator().to(device)
+        g_optim = torch.optim.Adam(gen.parameters(), lr=2e-4, betas=(0.5, 0.999))
+        d_optim = torch.optim.Adam(disc.parameters(), lr=2e-4, betas=(0.5, 0.999))
+    distrib.broadcast_model(gen)
+    adv = AdversarialLoss(disc, d_optim)
+    batch = args.batch
+    real = torch.tanh(torch.randn(batch, 3, 64, 64, device=device))
+
+    def step():
+        z = torch.randn(batch, nz, 1, 1, device=device)
+        fake = gen(z)
+        adv.train_adv(fake, real)
+        g_loss = adv(fake)
+        g_optim.zero_grad()
+        if hasattr(g_optim, "grad_buffers"):
+            g_loss.backward()
+            distrib.sync_flat_gradients(g_optim)
+        else:
+            with distrib.eager_sync_model(gen):
+                g_loss.backward()
+        g_optim.step()
+        return g_loss
+
+    for _ in range(args.warmup):
+        step()
+    distrib.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    distrib.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    t = torch.tensor([elapsed], device=distrib.device(), dtype=torch.float64)
+    if ws > 1:
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+    elapsed = float(t.item())
+    if rank == 0:
+        print(json.dumps({
+            "metric": "img/sec", "value": ws * batch * args.steps / elapsed,
+            "unit": "img/s", "n_gpus": ws, "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000,
+            "higher_is_better": True, "scaling": "weak", "vs_baseline": None,
+            "dtype": "bf16" if native else "fp32", "data": "synthetic",
+            "config": {"model": "dcgan64", "dataset": "synthetic-64x64",
+                       "global_batch": ws * batch, "img_size": 64, "nz": nz,
+                       "parallelism": f"dp{ws}",
+                       "mode": "native-kernels" if native else "reference-torch-ops"},
+        }))
 
 
 if __name__ == "__main__":

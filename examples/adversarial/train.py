@@ -11,7 +11,9 @@ import torch
 import flashy_amd
 from flashy_amd import BaseSolver, Formatter, distrib
 from flashy_amd.adversarial import AdversarialLoss
-from flashy_amd.models import DCGANDiscriminator, DCGANGenerator
+from flashy_amd.models import (DCGANDiscriminator, DCGANGenerator,
+                               NativeDCGANDiscriminator, NativeDCGANGenerator)
+from flashy_amd.optim import FusedAdam
 from flashy_amd.utils import averager
 from flashy_amd import xp as fxp
 
@@ -24,13 +26,25 @@ class Solver(BaseSolver):
         self.cfg = cfg
         self.device = torch.device("cuda" if torch.cuda.is_available() else "cpu") \
             if cfg.device == "auto" else torch.device(cfg.device)
-        self.generator = DCGANGenerator(cfg.nz, cfg.ngf).to(self.device)
-        distrib.broadcast_model(self.generator)
-        self.g_optim = torch.optim.Adam(self.generator.parameters(), lr=cfg.lr,
-                                        betas=(cfg.beta1, 0.999))
-        disc = DCGANDiscriminator(cfg.ndf).to(self.device)
-        d_optim = torch.optim.Adam(disc.parameters(), lr=cfg.lr,
-                                   betas=(cfg.beta1, 0.999))
+        self.native = self.device.type == "cuda" and cfg.get("native", True)
+        if self.native:  # gfx950 kernel G/D (trunk needs nz % 64 == 0)
+            self.nz = max(64, (cfg.nz + 63) // 64 * 64)
+            self.generator = NativeDCGANGenerator(self.nz, cfg.ngf).to(self.device)
+            distrib.broadcast_model(self.generator)
+            self.g_optim = FusedAdam(self.generator.parameters(), lr=cfg.lr,
+                                     betas=(cfg.beta1, 0.999))
+            disc = NativeDCGANDiscriminator(cfg.ndf).to(self.device)
+            d_optim = FusedAdam(disc.parameters(), lr=cfg.lr,
+                                betas=(cfg.beta1, 0.999))
+        else:
+            self.nz = cfg.nz
+            self.generator = DCGANGenerator(self.nz, cfg.ngf).to(self.device)
+            distrib.broadcast_model(self.generator)
+            self.g_optim = torch.optim.Adam(self.generator.parameters(), lr=cfg.lr,
+                                            betas=(cfg.beta1, 0.999))
+            disc = DCGANDiscriminator(cfg.ndf).to(self.device)
+            d_optim = torch.optim.Adam(disc.parameters(), lr=cfg.lr,
+                                       betas=(cfg.beta1, 0.999))
         self.adv = AdversarialLoss(disc, d_optim)
         self.register_stateful("generator", "g_optim", "adv")
 
@@ -47,13 +61,17 @@ class Solver(BaseSolver):
         lp = self.log_progress("train", range(self.cfg.steps_per_epoch), updates=5)
         for _ in lp:
             real = self._real_batch()
-            z = torch.randn(self.cfg.batch_size, self.cfg.nz, 1, 1, device=self.device)
+            z = torch.randn(self.cfg.batch_size, self.nz, 1, 1, device=self.device)
             fake = self.generator(z)
             d_loss = self.adv.train_adv(fake, real)
             g_loss = self.adv(fake)
             self.g_optim.zero_grad()
-            with distrib.eager_sync_model(self.generator):
+            if hasattr(self.g_optim, "grad_buffers"):
                 g_loss.backward()
+                distrib.sync_flat_gradients(self.g_optim)
+            else:
+                with distrib.eager_sync_model(self.generator):
+                    g_loss.backward()
             self.g_optim.step()
             metrics = avg({"g_loss": g_loss.item(), "d_loss": d_loss.item()})
             lp.update(**metrics)
@@ -66,7 +84,7 @@ class Solver(BaseSolver):
             self.run_stage("train", self.train_stage)
             if distrib.is_rank_zero():
                 with torch.no_grad():
-                    z = torch.randn(8, self.cfg.nz, 1, 1, device=self.device)
+                    z = torch.randn(8, self.nz, 1, 1, device=self.device)
                     sample = self.generator(z).add(1).div(2).clamp(0, 1)
                 self.log_image("train", "samples", sample.cpu())
             self.commit()

@@ -69,8 +69,17 @@ class AdversarialLoss(nn.Module):
         loss = self.loss(self.adversary(fake.detach()), 1.0) + \
             self.loss(self.adversary(real.detach()), 0.0)
         self.optimizer.zero_grad()
-        with distrib.eager_sync_model(self.adversary):
+        if hasattr(self.optimizer, "grad_buffers"):
+            # flat optimizer (native kernels write grads straight into the
+            # flat buffer, bypassing autograd hooks): one all-reduce per group
             loss.backward()
+            distrib.sync_flat_gradients(self.optimizer)
+            if distrib.is_distributed():
+                distrib.average_tensors(
+                    [b for b in self.adversary.buffers() if b.is_floating_point()])
+        else:
+            with distrib.eager_sync_model(self.adversary):
+                loss.backward()
         self.optimizer.step()
         return loss.detach()
 
