@@ -255,7 +255,7 @@ def bn_msplit(M: int, C: int) -> int:
     """Blocks along M for the BN reductions: ~256 blocks saturate the chip
     while keeping the partial-combine kernels cheap."""
     cols = max(1, C // 64)
-    msplit = max(1, min(128 // cols, 128))
+    msplit = max(1, min(512 // cols, 512))
     msplit = max(1, min(msplit, (M + 31) // 32))
     if msplit >= 4:
         msplit &= ~3  # multiple of 4: per-channel partial rows stay 16B-aligned

@@ -342,17 +342,11 @@ k_conv_stem_wgrad(const uint16_t* __restrict__ x,
     const int rsc = d.R * d.S * d.C;
     const int j0 = blockIdx.y * 32;          // tap chunk base
     const int jn = min(32, rsc - j0);
-    __shared__ int tap_r[32], tap_s[32], tap_c[32];
-    if (threadIdx.x < 32) {
-        const int j = j0 + threadIdx.x;
-        if (j < rsc) {
-            tap_r[threadIdx.x] = j / (d.S * d.C);
-            const int sc = j - tap_r[threadIdx.x] * d.S * d.C;
-            tap_s[threadIdx.x] = sc / d.C;
-            tap_c[threadIdx.x] = sc - tap_s[threadIdx.x] * d.C;
-        }
-    }
-    __syncthreads();
+    // decode the chunk base once; walk (r,s,c) incrementally per tap
+    int r0 = j0 / (d.S * d.C);
+    const int sc0 = j0 - r0 * d.S * d.C;
+    int s0 = sc0 / d.C;
+    int c0 = sc0 - s0 * d.C;
 
     const int64_t M = (int64_t)d.N * d.Ho * d.Wo;
     const int64_t ms = (int64_t)blockIdx.z * m_per_block;
@@ -367,14 +361,18 @@ k_conv_stem_wgrad(const uint16_t* __restrict__ x,
         const int64_t n = m / ((int64_t)d.Ho * d.Wo);
         const int hb = ho * d.stride - d.pad;
         const int wb = wo * d.stride - d.pad;
+        int r = r0, sj = s0, c = c0;
         for (int j = 0; j < jn; ++j) {
-            const int hi = hb + tap_r[j];
-            const int wi = wb + tap_s[j];
+            const int hi = hb + r;
+            const int wi = wb + sj;
             if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W)
                 part[j] = fmaf(
-                    bf16_to_f32(x[((n * d.H + hi) * d.W + wi) * (int64_t)d.C +
-                                  tap_c[j]]),
+                    bf16_to_f32(x[((n * d.H + hi) * d.W + wi) * (int64_t)d.C + c]),
                     go, part[j]);
+            if (++c == d.C) {
+                c = 0;
+                if (++sj == d.S) { sj = 0; ++r; }
+            }
         }
     }
     __shared__ float red[8][32][32];
