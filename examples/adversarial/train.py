@@ -78,7 +78,10 @@ class Solver(BaseSolver):
         return distrib.average_metrics(metrics, self.cfg.steps_per_epoch)
 
     def run(self):
-        self.restore()
+        if self.restore():
+            for opt in (self.g_optim, self.adv.optimizer):
+                if hasattr(opt, "refresh_bf16"):
+                    opt.refresh_bf16()
         self.log_hyperparams(self.cfg)
         for epoch in range(self.epoch, self.cfg.epochs + 1):
             self.run_stage("train", self.train_stage)

@@ -119,7 +119,9 @@ class Solver(BaseSolver):
         return distrib.average_metrics(metrics, len(loader))
 
     def run(self):
-        self.restore()
+        if self.restore() and hasattr(self.optim, "refresh_bf16"):
+            # restored fp32 params -> re-sync the bf16 weight mirrors
+            self.optim.refresh_bf16()
         self.log_hyperparams(self.cfg)
         for epoch in range(self.epoch, self.cfg.epochs + 1):
             self.run_stage("train", self.do_train_valid, True)

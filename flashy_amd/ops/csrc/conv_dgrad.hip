@@ -14,7 +14,7 @@
 
 #include "conv_common.h"
 
-template <int BM, bool SPLITK, bool S1>
+template <int BM, bool SPLITK, bool S1, int SUBS = 2>
 __global__ void __launch_bounds__(CONV_THREADS)
 k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_rsck,
              uint16_t* __restrict__ dx, float* __restrict__ ws_out,
@@ -23,7 +23,7 @@ k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_r
     constexpr int WAVES_N = 4 / WAVES_M;
     constexpr int MF = BM / WAVES_M / 16;
     constexpr int NF = CONV_BN / WAVES_N / 16;
-    constexpr int BK2 = 2 * CONV_BK;
+    constexpr int BK2 = SUBS * CONV_BK;
     constexpr int CHUNKS = BM * (BK2 / 8);
     constexpr int CPT = (CHUNKS + CONV_THREADS - 1) / CONV_THREADS;
 
@@ -44,7 +44,7 @@ k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_r
 #pragma unroll
     for (int t = 0; t < CPT; ++t) {
         const int chunk = tid + t * CONV_THREADS;
-        const int row = chunk >> 3;
+        const int row = chunk / (BK2 / 8);
         const int64_t m = m0 + row;
         if (chunk < CHUNKS && m < M) {
             const int hw = d.H * d.W;
@@ -56,7 +56,7 @@ k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_r
             st_n[t] = -1;
         }
         const int kk0 = (SPLITK ? blockIdx.z * stages_per_split * BK2 : 0) +
-                        (chunk & 7) * 8;
+                        (chunk % (BK2 / 8)) * 8;
         st_r[t] = kk0 / (d.S * d.K);
         const int sk0 = kk0 - st_r[t] * d.S * d.K;
         st_s[t] = sk0 / d.K;
@@ -93,8 +93,8 @@ k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_r
         for (int t = 0; t < CPT; ++t) {
             const int chunk = tid + t * CONV_THREADS;
             if (chunk < CHUNKS) {
-                const int row = chunk >> 3;
-                const int koff = (chunk & 7) * 8;
+                const int row = chunk / (BK2 / 8);
+                const int koff = (chunk % (BK2 / 8)) * 8;
                 const int sub = koff >> 5;
                 *reinterpret_cast<short8*>(
                     &buf[(sub * BM + row) * CONV_APITCH + (koff & 31)]) = src[t];
@@ -120,7 +120,7 @@ k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_r
     }
     auto load_b = [&](short8 (*dst)[NF]) {
 #pragma unroll
-        for (int sub = 0; sub < 2; ++sub) {
+        for (int sub = 0; sub < SUBS; ++sub) {
 #pragma unroll
             for (int nf = 0; nf < NF; ++nf)
                 dst[sub][nf] = (b_r[sub] < d.R)
@@ -151,7 +151,7 @@ k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_r
     __syncthreads();
 
     auto step = [&](int i, const uint16_t* buf, uint16_t* nbuf) {
-        short8 b[2][NF];
+        short8 b[SUBS][NF];
         load_b(b);
         if (i + 1 < n_stages) {
             lds_write(nbuf, stage);
@@ -159,7 +159,7 @@ k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_r
         }
         const int kc = (s0 + i) * BK2;
 #pragma unroll
-        for (int sub = 0; sub < 2; ++sub) {
+        for (int sub = 0; sub < SUBS; ++sub) {
             if (kc + sub * CONV_BK >= rsk) break;
 #pragma unroll
             for (int mf = 0; mf < MF; ++mf) {
@@ -211,7 +211,12 @@ extern "C" void launch_conv_dgrad(const void* dout, const void* w_rsck,
     auto dd = (const uint16_t*)dout;
     auto ww = (const uint16_t*)w_rsck;
     auto xx = (uint16_t*)dx;
-    if (d.stride == 1) {
+    extern int conv_subs_dg();
+    if (conv_subs_dg() == 1 && d.stride == 1) {
+        if (bm == 128) k_conv_dgrad<128, false, true, 1><<<grid, CONV_THREADS, 0, stream>>>(dd, ww, xx, nullptr, d, 0);
+        else if (bm == 64) k_conv_dgrad<64, false, true, 1><<<grid, CONV_THREADS, 0, stream>>>(dd, ww, xx, nullptr, d, 0);
+        else k_conv_dgrad<32, false, true, 1><<<grid, CONV_THREADS, 0, stream>>>(dd, ww, xx, nullptr, d, 0);
+    } else if (d.stride == 1) {
         if (bm == 128) k_conv_dgrad<128, false, true><<<grid, CONV_THREADS, 0, stream>>>(dd, ww, xx, nullptr, d, 0);
         else if (bm == 64) k_conv_dgrad<64, false, true><<<grid, CONV_THREADS, 0, stream>>>(dd, ww, xx, nullptr, d, 0);
         else k_conv_dgrad<32, false, true><<<grid, CONV_THREADS, 0, stream>>>(dd, ww, xx, nullptr, d, 0);
@@ -238,6 +243,15 @@ extern "C" void launch_conv_dgrad_splitk(const void* dout, const void* w_rsck,
         k_conv_dgrad<64, true, false><<<grid, CONV_THREADS, 0, stream>>>(
             (const uint16_t*)dout, (const uint16_t*)w_rsck, nullptr, (float*)ws,
             d, spz);
+}
+
+#include <cstdlib>
+int conv_subs_dg() {
+    static int v = [] {
+        const char* e = getenv("FLASHY_CONV_SUBS");
+        return (e && e[0] == '1') ? 1 : 2;
+    }();
+    return v;
 }
 
 // ---------------------------------------------------------------------------

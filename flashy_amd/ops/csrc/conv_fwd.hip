@@ -20,7 +20,7 @@
 
 #include "conv_common.h"
 
-template <int BM, bool RELU, bool SPLITK>
+template <int BM, bool RELU, bool SPLITK, int SUBS = 2>
 __global__ void __launch_bounds__(CONV_THREADS)
 k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
            uint16_t* __restrict__ y, float* __restrict__ ws_out,
@@ -29,7 +29,7 @@ k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
     constexpr int WAVES_N = 4 / WAVES_M;
     constexpr int MF = BM / WAVES_M / 16;      // m fragments per wave
     constexpr int NF = CONV_BN / WAVES_N / 16; // n fragments per wave
-    constexpr int BK2 = 2 * CONV_BK;           // 64-deep stage
+    constexpr int BK2 = SUBS * CONV_BK;        // stage depth (SUBS x 32)
     constexpr int CHUNKS = BM * (BK2 / 8);
     constexpr int CPT = (CHUNKS + CONV_THREADS - 1) / CONV_THREADS;
 
@@ -52,7 +52,7 @@ k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
 #pragma unroll
     for (int t = 0; t < CPT; ++t) {
         const int chunk = tid + t * CONV_THREADS;
-        const int row = chunk >> 3;
+        const int row = chunk / (BK2 / 8);
         const int64_t m = m0 + row;
         if (chunk < CHUNKS && m < M) {
             const int hw = d.Ho * d.Wo;
@@ -64,7 +64,7 @@ k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
             st_n[t] = -1;
         }
         const int kk = (SPLITK ? blockIdx.z * stages_per_split * BK2 : 0) +
-                       (chunk & 7) * 8;   // tap offset of the first stage
+                       (chunk % (BK2 / 8)) * 8;   // tap offset of stage 0
         st_r[t] = kk / (d.S * d.C);
         const int sc = kk - st_r[t] * d.S * d.C;
         st_s[t] = sc / d.C;
@@ -98,8 +98,8 @@ k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
         for (int t = 0; t < CPT; ++t) {
             const int chunk = tid + t * CONV_THREADS;
             if (chunk < CHUNKS) {
-                const int row = chunk >> 3;
-                const int koff = (chunk & 7) * 8;
+                const int row = chunk / (BK2 / 8);
+                const int koff = (chunk % (BK2 / 8)) * 8;
                 const int sub = koff >> 5;
                 *reinterpret_cast<short8*>(
                     &buf[(sub * BM + row) * CONV_APITCH + (koff & 31)]) = src[t];
@@ -113,7 +113,7 @@ k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
 
     auto load_b = [&](short8 (*dst)[NF], int stage) {
 #pragma unroll
-        for (int sub = 0; sub < 2; ++sub)
+        for (int sub = 0; sub < SUBS; ++sub)
 #pragma unroll
             for (int nf = 0; nf < NF; ++nf)
                 dst[sub][nf] = *reinterpret_cast<const short8*>(
@@ -140,7 +140,7 @@ k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
     // B fragments load at use: the weight panel is L2-resident (re-read by
     // every M-block), and the freed 32 VGPRs buy a third wave per SIMD.
     auto step = [&](int i, const uint16_t* buf, uint16_t* nbuf) {
-        short8 b[2][NF];
+        short8 b[SUBS][NF];
         load_b(b, s0 + i);
         if (i + 1 < n_stages) {
             lds_write(nbuf, stage);
@@ -148,7 +148,7 @@ k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
         }
         const int kc = (s0 + i) * BK2;
 #pragma unroll
-        for (int sub = 0; sub < 2; ++sub) {
+        for (int sub = 0; sub < SUBS; ++sub) {
             if (kc + sub * CONV_BK >= rsc) break;
 #pragma unroll
             for (int mf = 0; mf < MF; ++mf) {
@@ -229,6 +229,15 @@ static int pick_bm(int64_t M, int K) {
     return 32;
 }
 
+#include <cstdlib>
+static int conv_subs() {   // A/B switch: FLASHY_CONV_SUBS=1 -> 32-deep stages
+    static int v = [] {
+        const char* e = getenv("FLASHY_CONV_SUBS");
+        return (e && e[0] == '1') ? 1 : 2;
+    }();
+    return v;
+}
+
 extern "C" void launch_conv_fwd(const void* x, const void* w, void* y,
                                 ConvDims d, int relu, hipStream_t stream) {
     const int64_t M = (int64_t)d.N * d.Ho * d.Wo;
@@ -237,6 +246,18 @@ extern "C" void launch_conv_fwd(const void* x, const void* w, void* y,
     auto xx = (const uint16_t*)x;
     auto ww = (const uint16_t*)w;
     auto yy = (uint16_t*)y;
+    if (conv_subs() == 1) {
+        if (relu) {
+            if (bm == 128) k_conv_fwd<128, true, false, 1><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, d, 0);
+            else if (bm == 64) k_conv_fwd<64, true, false, 1><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, d, 0);
+            else k_conv_fwd<32, true, false, 1><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, d, 0);
+        } else {
+            if (bm == 128) k_conv_fwd<128, false, false, 1><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, d, 0);
+            else if (bm == 64) k_conv_fwd<64, false, false, 1><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, d, 0);
+            else k_conv_fwd<32, false, false, 1><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, d, 0);
+        }
+        return;
+    }
     if (relu) {
         if (bm == 128) k_conv_fwd<128, true, false><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, d, 0);
         else if (bm == 64) k_conv_fwd<64, true, false><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, d, 0);

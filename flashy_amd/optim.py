@@ -153,6 +153,9 @@ class FlatOptimizer:
         self.defaults.update(state.get("defaults", {}))
         self.step_count = state.get("step_count", 0)
         self._load_extra_state(state.get("extra", {}))
+        # model params are usually restored in the same breath
+        # (StateManager restores the model first); re-sync the bf16 mirrors
+        self.refresh_bf16()
 
 
 class FusedSGD(FlatOptimizer):

@@ -99,3 +99,24 @@ def test_checkpoint_views_compacted(tmp_path):
     model2.load_state_dict(loaded)
     for p, q in zip(model.parameters(), model2.parameters()):
         assert torch.equal(p, q)
+
+
+def test_bf16_mirror_refresh_on_restore():
+    """Restoring fp32 params must re-sync the bf16 weight mirrors
+    (load_state_dict path; solver-level restore calls refresh_bf16)."""
+    import torch
+    model, _ = _models()
+    opt = FusedSGD(model.parameters(), lr=0.05, momentum=0.9, bf16_mirror=True)
+    p = next(model.parameters())
+    assert hasattr(p, "_bf16_mirror")
+    assert torch.allclose(p._bf16_mirror.float(), p.float(), atol=1e-2)
+    saved_params = [q.detach().clone() for q in model.parameters()]
+    saved_opt = opt.state_dict()
+    _train(model, opt, steps=2)
+    # emulate a restore: params back in place + optimizer state reload
+    with torch.no_grad():
+        for q, s in zip(model.parameters(), saved_params):
+            q.copy_(s)
+    opt.load_state_dict(saved_opt)
+    assert torch.allclose(p._bf16_mirror.float(),
+                          p.detach().to(torch.bfloat16).float())
