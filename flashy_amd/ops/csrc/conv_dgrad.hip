@@ -14,7 +14,7 @@
 
 #include "conv_common.h"
 
-template <int BM, bool SPLITK, bool S1, int SUBS = 2>
+template <int BM, bool SPLITK, bool S1, int SUBS = 2, int BN = CONV_BN>
 __global__ void __launch_bounds__(CONV_THREADS)
 k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_rsck,
              uint16_t* __restrict__ dx, float* __restrict__ ws_out,
@@ -22,7 +22,7 @@ k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_r
     constexpr int WAVES_M = BM >= 64 ? 2 : 1;
     constexpr int WAVES_N = 4 / WAVES_M;
     constexpr int MF = BM / WAVES_M / 16;
-    constexpr int NF = CONV_BN / WAVES_N / 16;
+    constexpr int NF = BN / WAVES_N / 16;
     constexpr int BK2 = SUBS * CONV_BK;
     constexpr int CHUNKS = BM * (BK2 / 8);
     constexpr int CPT = (CHUNKS + CONV_THREADS - 1) / CONV_THREADS;
@@ -35,7 +35,7 @@ k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_r
     const int wave_m = WAVES_M == 1 ? 0 : (wid >> 1);
     const int wave_n = WAVES_M == 1 ? wid : (wid & 1);
     const int64_t m0 = (int64_t)blockIdx.x * BM;
-    const int col0 = blockIdx.y * CONV_BN;
+    const int col0 = blockIdx.y * BN;
 
     __shared__ uint16_t A_lds[2][2 * BM * CONV_APITCH];
 
@@ -104,7 +104,7 @@ k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_r
 
     const int a_row = wave_m * (BM / WAVES_M) + (lane & 15);
     const int a_koff = (lane >> 4) * 8;
-    const int b_col = col0 + wave_n * (CONV_BN / WAVES_N) + (lane & 15);
+    const int b_col = col0 + wave_n * (BN / WAVES_N) + (lane & 15);
 
     // B tap state: (r, s, k) of (stage base + a_koff); (r,s) constant across
     // each 32-subchunk since K % 32 == 0, so track per sub ∈ {0,1}.
@@ -180,7 +180,7 @@ k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_r
     }
 
     const int64_t out_row0 = m0 + wave_m * (BM / WAVES_M) + (lane >> 4) * 4;
-    const int out_col0 = col0 + wave_n * (CONV_BN / WAVES_N) + (lane & 15);
+    const int out_col0 = col0 + wave_n * (BN / WAVES_N) + (lane & 15);
 #pragma unroll
     for (int mf = 0; mf < MF; ++mf)
 #pragma unroll
@@ -202,15 +202,23 @@ k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_r
 extern "C" void launch_conv_dgrad(const void* dout, const void* w_rsck,
                                   void* dx, ConvDims d, hipStream_t stream) {
     const int64_t M = (int64_t)d.N * d.H * d.W;
+    auto dd = (const uint16_t*)dout;
+    auto ww = (const uint16_t*)w_rsck;
+    auto xx = (uint16_t*)dx;
+    if (d.C % 128 == 0 && (M + 127) / 128 * (d.C / 128) >= 208) {
+        dim3 g((unsigned)((M + 127) / 128), (unsigned)(d.C / 128));
+        if (d.stride == 1)
+            k_conv_dgrad<128, false, true, 2, 128><<<g, CONV_THREADS, 0, stream>>>(dd, ww, xx, nullptr, d, 0);
+        else
+            k_conv_dgrad<128, false, false, 2, 128><<<g, CONV_THREADS, 0, stream>>>(dd, ww, xx, nullptr, d, 0);
+        return;
+    }
     const int ktiles = d.C / CONV_BN;
     int bm = 32;
     for (int cand : {128, 64}) {
         if ((M + cand - 1) / cand * ktiles >= 208) { bm = cand; break; }
     }
     dim3 grid((unsigned)((M + bm - 1) / bm), (unsigned)(d.C / CONV_BN));
-    auto dd = (const uint16_t*)dout;
-    auto ww = (const uint16_t*)w_rsck;
-    auto xx = (uint16_t*)dx;
     extern int conv_subs_dg();
     if (conv_subs_dg() == 1 && d.stride == 1) {
         if (bm == 128) k_conv_dgrad<128, false, true, 1><<<grid, CONV_THREADS, 0, stream>>>(dd, ww, xx, nullptr, d, 0);

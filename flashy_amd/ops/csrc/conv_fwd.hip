@@ -20,7 +20,7 @@
 
 #include "conv_common.h"
 
-template <int BM, bool RELU, bool SPLITK, int SUBS = 2>
+template <int BM, bool RELU, bool SPLITK, int SUBS = 2, int BN = CONV_BN>
 __global__ void __launch_bounds__(CONV_THREADS)
 k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
            uint16_t* __restrict__ y, float* __restrict__ ws_out,
@@ -28,7 +28,7 @@ k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
     constexpr int WAVES_M = BM >= 64 ? 2 : 1;
     constexpr int WAVES_N = 4 / WAVES_M;
     constexpr int MF = BM / WAVES_M / 16;      // m fragments per wave
-    constexpr int NF = CONV_BN / WAVES_N / 16; // n fragments per wave
+    constexpr int NF = BN / WAVES_N / 16;      // n fragments per wave
     constexpr int BK2 = SUBS * CONV_BK;        // stage depth (SUBS x 32)
     constexpr int CHUNKS = BM * (BK2 / 8);
     constexpr int CPT = (CHUNKS + CONV_THREADS - 1) / CONV_THREADS;
@@ -41,7 +41,7 @@ k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
     const int wave_m = WAVES_M == 1 ? 0 : (wid >> 1);
     const int wave_n = WAVES_M == 1 ? wid : (wid & 1);
     const int64_t m0 = (int64_t)blockIdx.x * BM;
-    const int col0 = blockIdx.y * CONV_BN;
+    const int col0 = blockIdx.y * BN;
 
     // [sub][row][APITCH] per buffer: row pitch 48 = conflict-free b128 groups
     __shared__ uint16_t A_lds[2][2 * BM * CONV_APITCH];
@@ -109,7 +109,7 @@ k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
 
     const int a_row = wave_m * (BM / WAVES_M) + (lane & 15);
     const int a_koff = (lane >> 4) * 8;
-    const int b_col = col0 + wave_n * (CONV_BN / WAVES_N) + (lane & 15);
+    const int b_col = col0 + wave_n * (BN / WAVES_N) + (lane & 15);
 
     auto load_b = [&](short8 (*dst)[NF], int stage) {
 #pragma unroll
@@ -169,7 +169,7 @@ k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
     }
 
     const int64_t out_row0 = m0 + wave_m * (BM / WAVES_M) + (lane >> 4) * 4;
-    const int out_col0 = col0 + wave_n * (CONV_BN / WAVES_N) + (lane & 15);
+    const int out_col0 = col0 + wave_n * (BN / WAVES_N) + (lane & 15);
 #pragma unroll
     for (int mf = 0; mf < MF; ++mf)
 #pragma unroll
@@ -220,13 +220,19 @@ extern "C" void launch_splitk_combine(const void* ws, void* out, int64_t total,
         (const float*)ws, (uint16_t*)out, total, zn, relu);
 }
 
-// pick BM so the grid keeps >= ~208 workgroups where possible (256 CUs)
-static int pick_bm(int64_t M, int K) {
-    const int ktiles = K / CONV_BN;
-    for (int bm : {128, 64}) {
-        if ((M + bm - 1) / bm * ktiles >= 208) return bm;
+// pick (BM, BN) so the grid keeps >= ~208 workgroups (256 CUs); BN=128
+// halves the A-tile re-reads across column tiles when K allows it.
+static void pick_tile(int64_t M, int K, int* bm_out, int* bn_out) {
+    if (K % 128 == 0 && (M + 127) / 128 * (K / 128) >= 208) {
+        *bm_out = 128;
+        *bn_out = 128;
+        return;
     }
-    return 32;
+    *bn_out = 64;
+    for (int bm : {128, 64}) {
+        if ((M + bm - 1) / bm * (K / 64) >= 208) { *bm_out = bm; return; }
+    }
+    *bm_out = 32;
 }
 
 #include <cstdlib>
@@ -241,8 +247,19 @@ static int conv_subs() {   // A/B switch: FLASHY_CONV_SUBS=1 -> 32-deep stages
 extern "C" void launch_conv_fwd(const void* x, const void* w, void* y,
                                 ConvDims d, int relu, hipStream_t stream) {
     const int64_t M = (int64_t)d.N * d.Ho * d.Wo;
-    const int bm = pick_bm(M, d.K);
-    dim3 grid((unsigned)((M + bm - 1) / bm), (unsigned)(d.K / CONV_BN));
+    int bm, bn;
+    pick_tile(M, d.K, &bm, &bn);
+    dim3 grid((unsigned)((M + bm - 1) / bm), (unsigned)(d.K / bn));
+    if (bn == 128) {
+        auto xx = (const uint16_t*)x;
+        auto ww = (const uint16_t*)w;
+        auto yy = (uint16_t*)y;
+        if (relu)
+            k_conv_fwd<128, true, false, 2, 128><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, d, 0);
+        else
+            k_conv_fwd<128, false, false, 2, 128><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, d, 0);
+        return;
+    }
     auto xx = (const uint16_t*)x;
     auto ww = (const uint16_t*)w;
     auto yy = (uint16_t*)y;
