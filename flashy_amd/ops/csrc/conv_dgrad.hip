@@ -199,12 +199,18 @@ k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_r
             }
 }
 
+extern "C" int launch_conv_dgrad_s2(const void* dout, const void* w_rsck,
+                                    void* dx, ConvDims d, hipStream_t stream);
+
 extern "C" void launch_conv_dgrad(const void* dout, const void* w_rsck,
                                   void* dx, ConvDims d, hipStream_t stream) {
     const int64_t M = (int64_t)d.N * d.H * d.W;
     auto dd = (const uint16_t*)dout;
     auto ww = (const uint16_t*)w_rsck;
     auto xx = (uint16_t*)dx;
+    if (d.stride == 2 &&
+        launch_conv_dgrad_s2(dout, w_rsck, dx, d, stream))
+        return;  // parity-class form (no zero-filled MFMA work)
     if (d.C % 128 == 0 && (M + 127) / 128 * (d.C / 128) >= 208) {
         dim3 g((unsigned)((M + 127) / 128), (unsigned)(d.C / 128));
         if (d.stride == 1)
@@ -251,6 +257,228 @@ extern "C" void launch_conv_dgrad_splitk(const void* dout, const void* w_rsck,
         k_conv_dgrad<64, true, false><<<grid, CONV_THREADS, 0, stream>>>(
             (const uint16_t*)dout, (const uint16_t*)w_rsck, nullptr, (float*)ws,
             d, spz);
+}
+
+// ---------------------------------------------------------------------------
+// Stride-2 dgrad, parity-class form: grid.z enumerates the 4 (hi%2, wi%2)
+// classes; within a class every tap (r, s) with r = ph + 2r', s = pw + 2s'
+// is VALID, so no MFMA work is zero-filled (the generic kernel wastes 3/4
+// on stride 2).  Same 64-deep double-buffered single-barrier schedule.
+// ---------------------------------------------------------------------------
+
+template <int BM>
+__global__ void __launch_bounds__(CONV_THREADS)
+k_conv_dgrad_s2(const uint16_t* __restrict__ dout,
+                const uint16_t* __restrict__ w_rsck,
+                uint16_t* __restrict__ dx, ConvDims d) {
+    constexpr int WAVES_M = BM >= 64 ? 2 : 1;
+    constexpr int WAVES_N = 4 / WAVES_M;
+    constexpr int MF = BM / WAVES_M / 16;
+    constexpr int NF = CONV_BN / WAVES_N / 16;
+    constexpr int BK2 = 64;
+    constexpr int CHUNKS = BM * (BK2 / 8);
+    constexpr int CPT = (CHUNKS + CONV_THREADS - 1) / CONV_THREADS;
+
+    const int ph = blockIdx.z >> 1;      // parity of (hi + pad)
+    const int pw = blockIdx.z & 1;
+    // first input coordinate in this class, and class extents
+    const int a0 = ((ph - d.pad) % 2 + 2) % 2;
+    const int b0 = ((pw - d.pad) % 2 + 2) % 2;
+    const int Hc = a0 < d.H ? (d.H - a0 + 1) / 2 : 0;
+    const int Wc = b0 < d.W ? (d.W - b0 + 1) / 2 : 0;
+    // valid tap counts
+    const int Rp = ph < d.R ? (d.R - ph + 1) / 2 : 0;
+    const int Sp = pw < d.S ? (d.S - pw + 1) / 2 : 0;
+    const int rsk = Rp * Sp * d.K;
+    const int64_t Mc = (int64_t)d.N * Hc * Wc;
+    if (rsk == 0 || Mc == 0) return;
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6;
+    const int wave_m = WAVES_M == 1 ? 0 : (wid >> 1);
+    const int wave_n = WAVES_M == 1 ? wid : (wid & 1);
+    const int64_t m0 = (int64_t)blockIdx.x * BM;
+    const int col0 = blockIdx.y * CONV_BN;
+
+    __shared__ uint16_t A_lds[2][2 * BM * CONV_APITCH];
+
+    // staging geometry: class row -> (n, ho base, wo base); all taps valid
+    // up to the usual Ho/Wo range checks at the borders.
+    int st_hb[CPT], st_wb[CPT], st_r[CPT], st_s[CPT], st_k[CPT];
+    int64_t st_n[CPT];
+#pragma unroll
+    for (int t = 0; t < CPT; ++t) {
+        const int chunk = tid + t * CONV_THREADS;
+        const int row = chunk / (BK2 / 8);
+        const int64_t m = m0 + row;
+        if (chunk < CHUNKS && m < Mc) {
+            const int hw = Hc * Wc;
+            st_n[t] = m / hw;
+            const int rem = (int)(m % hw);
+            const int hi = a0 + 2 * (rem / Wc);
+            const int wi = b0 + 2 * (rem % Wc);
+            st_hb[t] = (hi + d.pad - ph) >> 1;   // = ho when r' = 0
+            st_wb[t] = (wi + d.pad - pw) >> 1;
+        } else {
+            st_n[t] = -1;
+        }
+        const int kk0 = (chunk % (BK2 / 8)) * 8;
+        st_r[t] = kk0 / (Sp * d.K);
+        const int sk0 = kk0 - st_r[t] * Sp * d.K;
+        st_s[t] = sk0 / d.K;
+        st_k[t] = sk0 - st_s[t] * d.K;
+    }
+
+    auto load_stage = [&](short8* dst) {
+#pragma unroll
+        for (int t = 0; t < CPT; ++t) {
+            short8 v = {};
+            if (st_n[t] >= 0 && st_r[t] < Rp) {
+                const int ho = st_hb[t] - st_r[t];
+                const int wo = st_wb[t] - st_s[t];
+                if (ho >= 0 && ho < d.Ho && wo >= 0 && wo < d.Wo)
+                    v = *reinterpret_cast<const short8*>(
+                        dout + (((st_n[t] * d.Ho + ho) * d.Wo + wo) *
+                                (int64_t)d.K + st_k[t]));
+            }
+            dst[t] = v;
+            int k = st_k[t] + BK2;
+            while (k >= d.K) {
+                k -= d.K;
+                if (++st_s[t] == Sp) { st_s[t] = 0; ++st_r[t]; }
+            }
+            st_k[t] = k;
+        }
+    };
+    auto lds_write = [&](uint16_t* buf, const short8* src) {
+#pragma unroll
+        for (int t = 0; t < CPT; ++t) {
+            const int chunk = tid + t * CONV_THREADS;
+            if (chunk < CHUNKS) {
+                const int row = chunk / (BK2 / 8);
+                const int koff = (chunk % (BK2 / 8)) * 8;
+                const int sub = koff >> 5;
+                *reinterpret_cast<short8*>(
+                    &buf[(sub * BM + row) * CONV_APITCH + (koff & 31)]) = src[t];
+            }
+        }
+    };
+
+    const int a_row = wave_m * (BM / WAVES_M) + (lane & 15);
+    const int a_koff = (lane >> 4) * 8;
+    const int b_col = col0 + wave_n * (CONV_BN / WAVES_N) + (lane & 15);
+
+    int b_r[2], b_s[2], b_k[2];
+#pragma unroll
+    for (int sub = 0; sub < 2; ++sub) {
+        const int kk0 = sub * CONV_BK + a_koff;
+        b_r[sub] = kk0 / (Sp * d.K);
+        const int sk0 = kk0 - b_r[sub] * Sp * d.K;
+        b_s[sub] = sk0 / d.K;
+        b_k[sub] = sk0 - b_s[sub] * d.K;
+    }
+    auto load_b = [&](short8 (*dst)[NF]) {
+#pragma unroll
+        for (int sub = 0; sub < 2; ++sub) {
+#pragma unroll
+            for (int nf = 0; nf < NF; ++nf)
+                dst[sub][nf] = (b_r[sub] < Rp)
+                    ? *reinterpret_cast<const short8*>(
+                          w_rsck + ((int64_t)((ph + 2 * b_r[sub]) * d.S +
+                                              pw + 2 * b_s[sub]) * d.C +
+                                    b_col + nf * 16) * d.K + b_k[sub])
+                    : short8{};
+            int k = b_k[sub] + BK2;
+            while (k >= d.K) {
+                k -= d.K;
+                if (++b_s[sub] == Sp) { b_s[sub] = 0; ++b_r[sub]; }
+            }
+            b_k[sub] = k;
+        }
+    };
+
+    floatx4 acc[MF][NF] = {};
+    const int n_stages = (rsk + BK2 - 1) / BK2;
+    short8 stage[CPT];
+
+    load_stage(stage);
+    lds_write(A_lds[0], stage);
+    if (n_stages > 1) load_stage(stage);
+    __syncthreads();
+
+    auto step = [&](int i, const uint16_t* buf, uint16_t* nbuf) {
+        short8 b[2][NF];
+        load_b(b);
+        if (i + 1 < n_stages) {
+            lds_write(nbuf, stage);
+            if (i + 2 < n_stages) load_stage(stage);
+        }
+        const int kc = i * BK2;
+#pragma unroll
+        for (int sub = 0; sub < 2; ++sub) {
+            if (kc + sub * CONV_BK >= rsk) break;
+#pragma unroll
+            for (int mf = 0; mf < MF; ++mf) {
+                const short8 a = *reinterpret_cast<const short8*>(
+                    &buf[(sub * BM + a_row + mf * 16) * CONV_APITCH + a_koff]);
+#pragma unroll
+                for (int nf = 0; nf < NF; ++nf)
+                    acc[mf][nf] = MFMA_BF16(a, b[sub][nf], acc[mf][nf]);
+            }
+        }
+        __syncthreads();
+    };
+    for (int i = 0; i < n_stages;) {
+        step(i, A_lds[0], A_lds[1]);
+        if (++i >= n_stages) break;
+        step(i, A_lds[1], A_lds[0]);
+        ++i;
+    }
+
+    // epilogue: scatter rows back to stride-2 positions of dx
+    const int64_t out_row0 = m0 + wave_m * (BM / WAVES_M) + (lane >> 4) * 4;
+    const int out_col0 = col0 + wave_n * (CONV_BN / WAVES_N) + (lane & 15);
+#pragma unroll
+    for (int mf = 0; mf < MF; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < NF; ++nf)
+#pragma unroll
+            for (int rr = 0; rr < 4; ++rr) {
+                const int64_t m = out_row0 + mf * 16 + rr;
+                if (m < Mc) {
+                    const int hw = Hc * Wc;
+                    const int64_t n = m / hw;
+                    const int rem = (int)(m % hw);
+                    const int hi = a0 + 2 * (rem / Wc);
+                    const int wi = b0 + 2 * (rem % Wc);
+                    dx[((n * d.H + hi) * d.W + wi) * (int64_t)d.C +
+                       out_col0 + nf * 16] = f32_to_bf16(acc[mf][nf][rr]);
+                }
+            }
+}
+
+extern "C" int launch_conv_dgrad_s2(const void* dout, const void* w_rsck,
+                                    void* dx, ConvDims d, hipStream_t stream) {
+    if (d.stride != 2) return 0;
+    // class extents (max over classes) for grid sizing
+    const int Hc = (d.H + 1) / 2;
+    const int Wc = (d.W + 1) / 2;
+    const int64_t Mc = (int64_t)d.N * Hc * Wc;
+    const int ktiles = d.C / CONV_BN;
+    int bm = 32;
+    for (int cand : {128, 64}) {
+        if ((Mc + cand - 1) / cand * ktiles * 4 >= 208) { bm = cand; break; }
+    }
+    if ((Mc + 31) / 32 * ktiles * 4 < 104) return 0;  // too small: caller falls back
+    dim3 grid((unsigned)((Mc + bm - 1) / bm), (unsigned)ktiles, 4);
+    auto dd = (const uint16_t*)dout;
+    auto ww = (const uint16_t*)w_rsck;
+    auto xx = (uint16_t*)dx;
+    if (bm == 128) k_conv_dgrad_s2<128><<<grid, CONV_THREADS, 0, stream>>>(dd, ww, xx, d);
+    else if (bm == 64) k_conv_dgrad_s2<64><<<grid, CONV_THREADS, 0, stream>>>(dd, ww, xx, d);
+    else k_conv_dgrad_s2<32><<<grid, CONV_THREADS, 0, stream>>>(dd, ww, xx, d);
+    return 1;
 }
 
 #include <cstdlib>
