@@ -291,7 +291,37 @@ k_conv_dgrad_s2(const uint16_t* __restrict__ dout,
     const int Sp = pw < d.S ? (d.S - pw + 1) / 2 : 0;
     const int rsk = Rp * Sp * d.K;
     const int64_t Mc = (int64_t)d.N * Hc * Wc;
-    if (rsk == 0 || Mc == 0) return;
+    if (Mc == 0) return;
+    if (rsk == 0) {
+        // no tap reaches this parity class (e.g. 1x1 stride 2): its input
+        // pixels receive ZERO gradient — write it, don't skip it.
+        const int lane0 = threadIdx.x & 63;
+        const int wid0 = threadIdx.x >> 6;
+        const int wm = WAVES_M == 1 ? 0 : (wid0 >> 1);
+        const int wn = WAVES_M == 1 ? wid0 : (wid0 & 1);
+        const int64_t zr0 = (int64_t)blockIdx.x * BM +
+            wm * (BM / WAVES_M) + (lane0 >> 4) * 4;
+        const int zc0 = blockIdx.y * CONV_BN +
+            wn * (CONV_BN / WAVES_N) + (lane0 & 15);
+#pragma unroll
+        for (int mf = 0; mf < MF; ++mf)
+#pragma unroll
+            for (int nf = 0; nf < NF; ++nf)
+#pragma unroll
+                for (int rr = 0; rr < 4; ++rr) {
+                    const int64_t m = zr0 + mf * 16 + rr;
+                    if (m < Mc) {
+                        const int hw = Hc * Wc;
+                        const int64_t n = m / hw;
+                        const int rem = (int)(m % hw);
+                        const int hi = a0 + 2 * (rem / Wc);
+                        const int wi = b0 + 2 * (rem % Wc);
+                        dx[((n * d.H + hi) * d.W + wi) * (int64_t)d.C +
+                           zc0 + nf * 16] = 0;
+                    }
+                }
+        return;
+    }
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
