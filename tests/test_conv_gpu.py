@@ -19,6 +19,7 @@ SHAPES = [
     (2, 16, 16, 64, 128, 1, 2, 0),
     (2, 4, 4, 256, 512, 3, 2, 1),
     (3, 7, 5, 64, 64, 3, 1, 1),       # non-pow2 spatial, odd M tail
+    (2, 16, 16, 64, 128, 4, 2, 1),    # even kernel, stride 2 (GAN shapes)
 ]
 
 
@@ -53,7 +54,7 @@ def test_conv_fwd(shape):
 
 
 @requires_gpu
-@pytest.mark.parametrize("shape", SHAPES[:5])
+@pytest.mark.parametrize("shape", SHAPES[:5] + SHAPES[6:])
 def test_conv_dgrad(shape):
     from flashy_amd import ops
     N, H, W, C, K, R, stride, pad = shape
@@ -78,7 +79,7 @@ def test_conv_dgrad(shape):
 
 
 @requires_gpu
-@pytest.mark.parametrize("shape", SHAPES[:5])
+@pytest.mark.parametrize("shape", SHAPES[:5] + SHAPES[6:])
 def test_conv_wgrad(shape):
     from flashy_amd import ops
     N, H, W, C, K, R, stride, pad = shape
