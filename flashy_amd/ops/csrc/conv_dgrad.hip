@@ -143,19 +143,21 @@ k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_r
     const int n_stages = SPLITK
         ? (all_stages - s0 < stages_per_split ? all_stages - s0 : stages_per_split)
         : all_stages;
-    short8 stage[CPT];
+    short8 stageA[CPT], stageB[CPT];
 
-    load_stage(stage);
-    lds_write(A_lds[0], stage);
-    if (n_stages > 1) load_stage(stage);
+    load_stage(stageA);
+    lds_write(A_lds[0], stageA);
+    if (n_stages > 1) load_stage(stageB);
+    if (n_stages > 2) load_stage(stageA);
     __syncthreads();
 
-    auto step = [&](int i, const uint16_t* buf, uint16_t* nbuf) {
+    auto step = [&](int i, const uint16_t* buf, uint16_t* nbuf,
+                    short8 (&rset)[CPT]) {
         short8 b[SUBS][NF];
         load_b(b);
         if (i + 1 < n_stages) {
-            lds_write(nbuf, stage);
-            if (i + 2 < n_stages) load_stage(stage);
+            lds_write(nbuf, rset);
+            if (i + 3 < n_stages) load_stage(rset);
         }
         const int kc = (s0 + i) * BK2;
 #pragma unroll
@@ -173,9 +175,9 @@ k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_r
         __syncthreads();
     };
     for (int i = 0; i < n_stages;) {
-        step(i, A_lds[0], A_lds[1]);
+        step(i, A_lds[0], A_lds[1], stageB);
         if (++i >= n_stages) break;
-        step(i, A_lds[1], A_lds[0]);
+        step(i, A_lds[1], A_lds[0], stageA);
         ++i;
     }
 
@@ -430,19 +432,21 @@ k_conv_dgrad_s2(const uint16_t* __restrict__ dout,
 
     floatx4 acc[MF][NF] = {};
     const int n_stages = (rsk + BK2 - 1) / BK2;
-    short8 stage[CPT];
+    short8 stageA[CPT], stageB[CPT];
 
-    load_stage(stage);
-    lds_write(A_lds[0], stage);
-    if (n_stages > 1) load_stage(stage);
+    load_stage(stageA);
+    lds_write(A_lds[0], stageA);
+    if (n_stages > 1) load_stage(stageB);
+    if (n_stages > 2) load_stage(stageA);
     __syncthreads();
 
-    auto step = [&](int i, const uint16_t* buf, uint16_t* nbuf) {
+    auto step = [&](int i, const uint16_t* buf, uint16_t* nbuf,
+                    short8 (&rset)[CPT]) {
         short8 b[2][NF];
         load_b(b);
         if (i + 1 < n_stages) {
-            lds_write(nbuf, stage);
-            if (i + 2 < n_stages) load_stage(stage);
+            lds_write(nbuf, rset);
+            if (i + 3 < n_stages) load_stage(rset);
         }
         const int kc = i * BK2;
 #pragma unroll
@@ -460,9 +464,9 @@ k_conv_dgrad_s2(const uint16_t* __restrict__ dout,
         __syncthreads();
     };
     for (int i = 0; i < n_stages;) {
-        step(i, A_lds[0], A_lds[1]);
+        step(i, A_lds[0], A_lds[1], stageB);
         if (++i >= n_stages) break;
-        step(i, A_lds[1], A_lds[0]);
+        step(i, A_lds[1], A_lds[0], stageA);
         ++i;
     }
 

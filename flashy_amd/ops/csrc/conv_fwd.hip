@@ -127,24 +127,28 @@ k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
     const int n_stages = SPLITK
         ? (all_stages - s0 < stages_per_split ? all_stages - s0 : stages_per_split)
         : all_stages;
-    short8 stage[CPT];
+    // TWO staging register sets: loads for stage j land in set[j&1] and are
+    // written to LDS two steps later — global latency hides under 2 full
+    // stages instead of 1.
+    short8 stageA[CPT], stageB[CPT];
 
-    // prologue: first stage -> buf0; next -> regs
-    load_stage(stage);
-    lds_write(A_lds[0], stage);
-    if (n_stages > 1) load_stage(stage);
+    // prologue: s0 -> buf0 (via A); s1 -> B; s2 -> A
+    load_stage(stageA);
+    lds_write(A_lds[0], stageA);
+    if (n_stages > 1) load_stage(stageB);
+    if (n_stages > 2) load_stage(stageA);
     __syncthreads();
 
-    // even/odd bodies keep the LDS buffer parity COMPILE-TIME (a runtime
-    // [i&1] register index lowers to thousands of cndmask selects — rule 20).
-    // B fragments load at use: the weight panel is L2-resident (re-read by
-    // every M-block), and the freed 32 VGPRs buy a third wave per SIMD.
-    auto step = [&](int i, const uint16_t* buf, uint16_t* nbuf) {
+    // even/odd bodies keep LDS-buffer and register-set parity COMPILE-TIME
+    // (a runtime [i&1] index lowers to cndmask ladders — rule 20).
+    // B fragments load at use: the weight panel is L2-resident.
+    auto step = [&](int i, const uint16_t* buf, uint16_t* nbuf,
+                    short8 (&rset)[CPT]) {
         short8 b[SUBS][NF];
         load_b(b, s0 + i);
         if (i + 1 < n_stages) {
-            lds_write(nbuf, stage);
-            if (i + 2 < n_stages) load_stage(stage);
+            lds_write(nbuf, rset);           // stage i+1
+            if (i + 3 < n_stages) load_stage(rset);  // stage i+3, same set
         }
         const int kc = (s0 + i) * BK2;
 #pragma unroll
@@ -162,9 +166,9 @@ k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
         __syncthreads();
     };
     for (int i = 0; i < n_stages;) {
-        step(i, A_lds[0], A_lds[1]);
+        step(i, A_lds[0], A_lds[1], stageB);
         if (++i >= n_stages) break;
-        step(i, A_lds[1], A_lds[0]);
+        step(i, A_lds[1], A_lds[0], stageA);
         ++i;
     }
 
