@@ -94,21 +94,19 @@ k_conv_wgrad(const uint16_t* __restrict__ x, const uint16_t* __restrict__ dout,
     };
 
     const int64_t n_stages = (me - ms + 2 * CONV_BK - 1) / (2 * CONV_BK);
-    short8 dvA[2], xvA[2], dvB[2], xvB[2];
-    load_pair(ms, dvA, xvA);
-    stage_write(doutT[0], xT[0], dvA, xvA);
-    if (n_stages > 1) load_pair(ms + 2 * CONV_BK, dvB, xvB);
-    if (n_stages > 2) load_pair(ms + 4 * CONV_BK, dvA, xvA);
+    short8 dv[2], xv[2];
+    load_pair(ms, dv, xv);
+    stage_write(doutT[0], xT[0], dv, xv);
+    if (n_stages > 1) load_pair(ms + 2 * CONV_BK, dv, xv);
     __syncthreads();
 
     auto step = [&](int64_t i, const uint16_t (&dT)[2 * 64 * WG_MP],
                     const uint16_t (&xTb)[2 * 64 * WG_MP],
                     uint16_t (&ndT)[2 * 64 * WG_MP],
-                    uint16_t (&nxT)[2 * 64 * WG_MP],
-                    short8 (&dset)[2], short8 (&xset)[2]) {
+                    uint16_t (&nxT)[2 * 64 * WG_MP]) {
         if (i + 1 < n_stages) {
-            stage_write(ndT, nxT, dset, xset);
-            if (i + 3 < n_stages) load_pair(ms + (i + 3) * 2 * CONV_BK, dset, xset);
+            stage_write(ndT, nxT, dv, xv);
+            if (i + 2 < n_stages) load_pair(ms + (i + 2) * 2 * CONV_BK, dv, xv);
         }
 #pragma unroll
         for (int sc = 0; sc < 2; ++sc) {
@@ -133,9 +131,9 @@ k_conv_wgrad(const uint16_t* __restrict__ x, const uint16_t* __restrict__ dout,
         __syncthreads();
     };
     for (int64_t i = 0; i < n_stages;) {
-        step(i, doutT[0], xT[0], doutT[1], xT[1], dvB, xvB);
+        step(i, doutT[0], xT[0], doutT[1], xT[1]);
         if (++i >= n_stages) break;
-        step(i, doutT[1], xT[1], doutT[0], xT[0], dvA, xvA);
+        step(i, doutT[1], xT[1], doutT[0], xT[0]);
         ++i;
     }
 
