@@ -6,8 +6,16 @@
 //   reduction over M = N*Ho*Wo, split across grid.z blocks, fp32
 //   atomicAdd into the flat fp32 gradient view (accumulate semantics match
 //   autograd, and zero_grad() memsets the flat buffer).
-// Both operands arrive [m][channel]-contiguous, so 16 B loads stage them
-// into LDS *transposed* ([channel][m]) for the MFMA fragment reads.
+//
+// Both operands arrive [m][channel]-contiguous.  They are staged into LDS
+// AS-LOADED ([m][channel] rows, one b128 store per thread per tensor per
+// subchunk) and the MFMA fragments — which need [channel][m] — are read with
+// gfx950's ds_read_b64_tr_b16 hardware transpose (guide T10): each 16-lane
+// group gathers a [4 m][16 channel] block, two reads (offset:+4 rows) build
+// the 8-m-deep operand.  This replaces the previous transposed-store scheme
+// whose 32 scalar ds_write_b16 per thread per stage dominated issue.
+// Row pitch 72 elements: 144 B rows keep b128 stores 16 B-aligned and give
+// tr reads conflict-free banks (row stride 36 dwords -> {0,36,8,44} mod 64).
 //
 // 64-deep m-stages (two MFMA-K subchunks) in two LDS buffers, ONE barrier
 // per stage, global loads for the next stage issued under the MFMA cluster
@@ -16,9 +24,9 @@
 
 #include "conv_common.h"
 
-#define WG_MP 40  // LDS m-pitch per 32-m subchunk (80 B, 16B-aligned reads)
+#define WG_P 72  // LDS row pitch in bf16 elements (144 B = 9 * 16 B)
 
-__global__ void __launch_bounds__(CONV_THREADS)
+__global__ void __launch_bounds__(CONV_THREADS, 4)  // cap at 128 VGPR
 k_conv_wgrad(const uint16_t* __restrict__ x, const uint16_t* __restrict__ dout,
              float* __restrict__ dw, ConvDims d, int m_per_split) {
     const int rsc = d.R * d.S * d.C;
@@ -33,14 +41,11 @@ k_conv_wgrad(const uint16_t* __restrict__ x, const uint16_t* __restrict__ dout,
     const int64_t ms = (int64_t)blockIdx.z * m_per_split;
     const int64_t me = min(ms + (int64_t)m_per_split, M);
 
-    // [buffer][subchunk][64 channels][WG_MP m]
-    __shared__ uint16_t doutT[2][2 * 64 * WG_MP];
-    __shared__ uint16_t xT[2][2 * 64 * WG_MP];
+    // [buffer][subchunk 2][m 32][WG_P], rows as loaded from global
+    __shared__ __attribute__((aligned(16))) uint16_t doutT[2][2 * 32 * WG_P];
+    __shared__ __attribute__((aligned(16))) uint16_t xT[2][2 * 32 * WG_P];
 
     floatx4 acc[2][2] = {};
-    const int frag_row = wave_k * 32 + (lane & 15);   // + kf*16  (K dim)
-    const int frag_col = wave_j * 32 + (lane & 15);   // + jf*16  (rsc dim)
-    const int moff = (lane >> 4) * 8;
 
     // staging: thread -> (m row within subchunk, channel octet); each stage
     // covers 64 m = 2 subchunks of 32.
@@ -73,25 +78,23 @@ k_conv_wgrad(const uint16_t* __restrict__ x, const uint16_t* __restrict__ dout,
             xv[sc] = u;
         }
     };
-    // XOR the 8-m column group by a per-row pattern so the 8 lanes sharing
-    // an m_r (channel octets k8 = 0..56) land on distinct banks (was 8-way).
-    auto swz_col = [](int row, int col) {
-        const int g = ((row >> 2) ^ (row >> 3)) & 3;
-        return (col & 7) | ((((col >> 3) ^ g) & 3) << 3);
-    };
-    auto stage_write = [&](uint16_t (&dT)[2 * 64 * WG_MP],
-                           uint16_t (&xTb)[2 * 64 * WG_MP],
+    auto stage_write = [&](uint16_t (&dT)[2 * 32 * WG_P],
+                           uint16_t (&xTb)[2 * 32 * WG_P],
                            const short8* dv, const short8* xv) {
 #pragma unroll
-        for (int sc = 0; sc < 2; ++sc)
-#pragma unroll
-            for (int j = 0; j < 8; ++j) {
-                const int row = sc * 64 + k8 + j;
-                const int col = swz_col(k8 + j, m_r);
-                dT[row * WG_MP + col] = ((const uint16_t*)&dv[sc])[j];
-                xTb[row * WG_MP + col] = ((const uint16_t*)&xv[sc])[j];
-            }
+        for (int sc = 0; sc < 2; ++sc) {
+            const int at = (sc * 32 + m_r) * WG_P + k8;
+            *reinterpret_cast<short8*>(&dT[at]) = dv[sc];
+            *reinterpret_cast<short8*>(&xTb[at]) = xv[sc];
+        }
     };
+
+    // per-lane element offset of the tr-read gather within one subchunk
+    // image: group (lane>>4) covers m rows g*8..g*8+7; input lane (lane&15)
+    // supplies row (k>>2), channels ch0 + 4*(k&3) (4 contiguous bf16).
+    const int tr_lane = ((lane & 15) >> 2) * WG_P + 4 * (lane & 3) +
+                        (lane >> 4) * 8 * WG_P;
+    union U64x2 { struct { unsigned long long lo, hi; } q; short8 v; };
 
     const int64_t n_stages = (me - ms + 2 * CONV_BK - 1) / (2 * CONV_BK);
     short8 dv[2], xv[2];
@@ -100,33 +103,45 @@ k_conv_wgrad(const uint16_t* __restrict__ x, const uint16_t* __restrict__ dout,
     if (n_stages > 1) load_pair(ms + 2 * CONV_BK, dv, xv);
     __syncthreads();
 
-    auto step = [&](int64_t i, const uint16_t (&dT)[2 * 64 * WG_MP],
-                    const uint16_t (&xTb)[2 * 64 * WG_MP],
-                    uint16_t (&ndT)[2 * 64 * WG_MP],
-                    uint16_t (&nxT)[2 * 64 * WG_MP]) {
+    auto step = [&](int64_t i, const uint16_t (&dT)[2 * 32 * WG_P],
+                    const uint16_t (&xTb)[2 * 32 * WG_P],
+                    uint16_t (&ndT)[2 * 32 * WG_P],
+                    uint16_t (&nxT)[2 * 32 * WG_P]) {
         if (i + 1 < n_stages) {
             stage_write(ndT, nxT, dv, xv);
             if (i + 2 < n_stages) load_pair(ms + (i + 2) * 2 * CONV_BK, dv, xv);
         }
 #pragma unroll
         for (int sc = 0; sc < 2; ++sc) {
-            short8 a[2], b[2];
-#pragma unroll
-            for (int f = 0; f < 2; ++f) {
-                // reads stay contiguous: the swizzle only permutes which
-                // 8-m group sits at moff for this row
-                a[f] = *reinterpret_cast<const short8*>(
-                    &dT[(sc * 64 + frag_row + f * 16) * WG_MP +
-                        swz_col(frag_row + f * 16, moff)]);
-                b[f] = *reinterpret_cast<const short8*>(
-                    &xTb[(sc * 64 + frag_col + f * 16) * WG_MP +
-                         swz_col(frag_col + f * 16, moff)]);
-            }
+            const int base = sc * 32 * WG_P + tr_lane;
+            const unsigned a0 = (unsigned)(unsigned long long)(const void*)
+                &dT[base + wave_k * 32];
+            const unsigned a1 = a0 + 32;                    // +16 ch * 2 B
+            const unsigned b0 = (unsigned)(unsigned long long)(const void*)
+                &xTb[base + wave_j * 32];
+            const unsigned b1 = b0 + 32;
+            U64x2 af[2], bf[2];
+            // 8 transpose reads; offset:576 = +4 m rows (4 * WG_P * 2 B)
+            asm volatile(
+                "ds_read_b64_tr_b16 %0, %8\n\t"
+                "ds_read_b64_tr_b16 %1, %8 offset:576\n\t"
+                "ds_read_b64_tr_b16 %2, %9\n\t"
+                "ds_read_b64_tr_b16 %3, %9 offset:576\n\t"
+                "ds_read_b64_tr_b16 %4, %10\n\t"
+                "ds_read_b64_tr_b16 %5, %10 offset:576\n\t"
+                "ds_read_b64_tr_b16 %6, %11\n\t"
+                "ds_read_b64_tr_b16 %7, %11 offset:576\n\t"
+                "s_waitcnt lgkmcnt(0)"
+                : "=v"(af[0].q.lo), "=v"(af[0].q.hi),
+                  "=v"(af[1].q.lo), "=v"(af[1].q.hi),
+                  "=v"(bf[0].q.lo), "=v"(bf[0].q.hi),
+                  "=v"(bf[1].q.lo), "=v"(bf[1].q.hi)
+                : "v"(a0), "v"(a1), "v"(b0), "v"(b1));
 #pragma unroll
             for (int kf = 0; kf < 2; ++kf)
 #pragma unroll
                 for (int jf = 0; jf < 2; ++jf)
-                    acc[kf][jf] = MFMA_BF16(a[kf], b[jf], acc[kf][jf]);
+                    acc[kf][jf] = MFMA_BF16(af[kf].v, bf[jf].v, acc[kf][jf]);
         }
         __syncthreads();
     };
