@@ -404,10 +404,13 @@ k_conv_stem_wgrad(const uint16_t* __restrict__ x,
         const int hb = ho * d.stride - d.pad;
         const int wb = wo * d.stride - d.pad;
         int r = r0, sj = s0, c = c0;
-        for (int j = 0; j < jn; ++j) {
+        // compile-time trip count so part[j] stays register-indexed
+        // (guide rule 20); `j < jn` is wave-uniform predication.
+#pragma unroll
+        for (int j = 0; j < 32; ++j) {
             const int hi = hb + r;
             const int wi = wb + sj;
-            if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W)
+            if (j < jn && hi >= 0 && hi < d.H && wi >= 0 && wi < d.W)
                 part[j] = fmaf(
                     bf16_to_f32(x[((n * d.H + hi) * d.W + wi) * (int64_t)d.C + c]),
                     go, part[j]);
@@ -432,10 +435,194 @@ k_conv_stem_wgrad(const uint16_t* __restrict__ x,
     }
 }
 
+// MFMA stem wgrad:  dW[64][taps] = dout^T[64][M] * im2col[M][taps].
+// Key fact: for a FIXED filter row r, the im2col taps (s, c) of one output
+// pixel are CONTIGUOUS x memory (NHWC, c innermost), so the B operand
+// stages with plain loads — no gather.  Both operands land in LDS
+// [m][channel] row-major and the MFMA fragments are read with
+// ds_read_b64_tr_b16 (same recipe as k_conv_wgrad).  The tap tile TJ is
+// padded to 32/64 columns; a chunk covers rows_per_chunk = TJ / (S*C)
+// filter rows (grid.y chunks when R*S*C > TJ, e.g. the 7x7x3 stem).
+// Requires K == 64, S*C <= TJ.  Replaces the scalar per-thread reduction
+// (k_conv_stem_wgrad below, kept for the S*C > 64 fallback) which was
+// ~6x slower (scalar-load latency bound).
+template <int OFF>
+__device__ inline short8 stem_tr2(const uint16_t* p) {
+    union { struct { unsigned long long lo, hi; } q; short8 v; } r;
+    const unsigned a = (unsigned)(unsigned long long)(const void*)p;
+    asm volatile("ds_read_b64_tr_b16 %0, %2\n\t"
+                 "ds_read_b64_tr_b16 %1, %2 offset:%3\n\t"
+                 "s_waitcnt lgkmcnt(0)"
+                 : "=v"(r.q.lo), "=v"(r.q.hi) : "v"(a), "n"(OFF));
+    return r.v;
+}
+
+template <int TJ>
+__global__ void __launch_bounds__(256, 4)
+k_stem_wgrad_mm(const uint16_t* __restrict__ x,
+                const uint16_t* __restrict__ dout, float* __restrict__ dw,
+                ConvDims d, int m_per_block, int rows_per_chunk) {
+    constexpr int PJ = TJ == 32 ? 36 : 72;   // pitches keep tr reads
+    constexpr int PD = 72;                   // conflict-free (see wgrad)
+    constexpr int EPT = TJ / 8;              // x elems per thread per 32-m
+    constexpr int JF = TJ / 16;
+    const int rsc = d.R * d.S * d.C;
+    const int sC = d.S * d.C;
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6;                // wave = 16-K block
+    const int r0 = blockIdx.y * rows_per_chunk;
+    const int jn = (min(rows_per_chunk, d.R - r0)) * sC;
+    const int jbase = r0 * sC;
+    const int64_t M = (int64_t)d.N * d.Ho * d.Wo;
+    const int64_t ms = (int64_t)blockIdx.z * m_per_block;
+    const int64_t me = min(ms + (int64_t)m_per_block, M);
+
+    __shared__ __attribute__((aligned(16))) uint16_t dT[2][2 * 32 * PD];
+    __shared__ __attribute__((aligned(16))) uint16_t xJ[2][2 * 32 * PJ];
+
+    const int m_r = tid >> 3;                // m row within subchunk
+    const int oct = tid & 7;
+    // per-elem tap decode, packed (roff<<16 | s<<8 | s*C+c) to spare VGPRs
+    int e_pack[EPT];
+#pragma unroll
+    for (int e = 0; e < EPT; ++e) {
+        const int j = oct * EPT + e;
+        const int roff = j / sC;
+        const int rem = j - roff * sC;
+        const int ss = rem / d.C;
+        e_pack[e] = (roff << 16) | (ss << 8) | rem;
+    }
+
+    auto load_stage = [&](int64_t mc, short8 (&dv)[2],
+                          uint16_t (&xsv)[2][EPT]) {
+#pragma unroll
+        for (int sc = 0; sc < 2; ++sc) {
+            const int64_t m = mc + sc * 32 + m_r;
+            const bool ok = m < me;
+            short8 v = {};
+            if (ok)
+                v = *reinterpret_cast<const short8*>(dout + m * d.K + oct * 8);
+            dv[sc] = v;
+            int64_t n = 0; int hb = 0, wb = 0;
+            if (ok) {
+                const int wo = (int)(m % d.Wo);
+                const int ho = (int)((m / d.Wo) % d.Ho);
+                n = m / ((int64_t)d.Ho * d.Wo);
+                hb = ho * d.stride - d.pad;
+                wb = wo * d.stride - d.pad;
+            }
+#pragma unroll
+            for (int e = 0; e < EPT; ++e) {
+                uint16_t u = 0;
+                const int hi = hb + r0 + (e_pack[e] >> 16);
+                const int wi = wb + ((e_pack[e] >> 8) & 255);
+                // scoff = s*C + c, so the address reduces to row + wb*C + scoff
+                if (ok && oct * EPT + e < jn &&
+                    hi >= 0 && hi < d.H && wi >= 0 && wi < d.W)
+                    u = x[((n * d.H + hi) * d.W + wb) * (int64_t)d.C +
+                          (e_pack[e] & 255)];
+                xsv[sc][e] = u;
+            }
+        }
+    };
+    auto stage_write = [&](uint16_t (&dTb)[2 * 32 * PD],
+                           uint16_t (&xJb)[2 * 32 * PJ],
+                           const short8 (&dv)[2],
+                           const uint16_t (&xsv)[2][EPT]) {
+#pragma unroll
+        for (int sc = 0; sc < 2; ++sc) {
+            *reinterpret_cast<short8*>(
+                &dTb[(sc * 32 + m_r) * PD + oct * 8]) = dv[sc];
+#pragma unroll
+            for (int e = 0; e < EPT; ++e)
+                xJb[(sc * 32 + m_r) * PJ + oct * EPT + e] = xsv[sc][e];
+        }
+    };
+
+    // tr-read per-lane gather offsets (see k_conv_wgrad for the derivation)
+    const int trd = ((lane & 15) >> 2) * PD + 4 * (lane & 3) +
+                    (lane >> 4) * 8 * PD;
+    const int trx = ((lane & 15) >> 2) * PJ + 4 * (lane & 3) +
+                    (lane >> 4) * 8 * PJ;
+    floatx4 acc[JF] = {};
+
+    const int64_t n_stages = (me - ms + 63) / 64;
+    short8 dv[2];
+    uint16_t xsv[2][EPT];
+    load_stage(ms, dv, xsv);
+    stage_write(dT[0], xJ[0], dv, xsv);
+    if (n_stages > 1) load_stage(ms + 64, dv, xsv);
+    __syncthreads();
+
+    auto step = [&](int64_t i, const uint16_t (&dTb)[2 * 32 * PD],
+                    const uint16_t (&xJb)[2 * 32 * PJ],
+                    uint16_t (&ndT)[2 * 32 * PD],
+                    uint16_t (&nxJ)[2 * 32 * PJ]) {
+        if (i + 1 < n_stages) {
+            stage_write(ndT, nxJ, dv, xsv);
+            if (i + 2 < n_stages) load_stage(ms + (i + 2) * 64, dv, xsv);
+        }
+#pragma unroll
+        for (int sc = 0; sc < 2; ++sc) {
+            const short8 a = stem_tr2<4 * PD * 2>(
+                &dTb[sc * 32 * PD + trd + wid * 16]);
+#pragma unroll
+            for (int jf = 0; jf < JF; ++jf) {
+                const short8 b = stem_tr2<4 * PJ * 2>(
+                    &xJb[sc * 32 * PJ + trx + jf * 16]);
+                acc[jf] = MFMA_BF16(a, b, acc[jf]);
+            }
+        }
+        __syncthreads();
+    };
+    for (int64_t i = 0; i < n_stages;) {
+        step(i, dT[0], xJ[0], dT[1], xJ[1]);
+        if (++i >= n_stages) break;
+        step(i, dT[1], xJ[1], dT[0], xJ[0]);
+        ++i;
+    }
+
+    const int k_out = wid * 16 + (lane >> 4) * 4;
+    const int j_lane = lane & 15;
+#pragma unroll
+    for (int jf = 0; jf < JF; ++jf) {
+        const int j = jf * 16 + j_lane;
+        if (j < jn)
+#pragma unroll
+            for (int rr = 0; rr < 4; ++rr)
+                atomicAdd(&dw[(int64_t)(k_out + rr) * rsc + jbase + j],
+                          acc[jf][rr]);
+    }
+}
+
 extern "C" void launch_conv_stem_wgrad(const void* x, const void* dout, void* dw,
                                        ConvDims d, hipStream_t stream) {
     const int64_t M = (int64_t)d.N * d.Ho * d.Wo;
     const int rsc = d.R * d.S * d.C;
+    const int sC = d.S * d.C;
+    if (d.K == 64 && sC <= 64) {             // MFMA path
+        const int TJ = rsc <= 32 ? 32 : 64;
+        const int rpc = rsc <= 32 ? d.R : (TJ / sC < d.R ? TJ / sC : d.R);
+        const int chunks = (d.R + rpc - 1) / rpc;
+        int64_t msplit = (M + 127) / 128;
+        const int64_t cap = 384 / chunks > 0 ? 384 / chunks : 1;
+        if (msplit > cap) msplit = cap;
+        if (msplit < 1) msplit = 1;
+        int m_per_block = (int)((M + msplit - 1) / msplit);
+        m_per_block = (m_per_block + 63) / 64 * 64;
+        const int zn = (int)((M + m_per_block - 1) / m_per_block);
+        dim3 grid(1, (unsigned)chunks, (unsigned)zn);
+        if (TJ == 32)
+            k_stem_wgrad_mm<32><<<grid, 256, 0, stream>>>(
+                (const uint16_t*)x, (const uint16_t*)dout, (float*)dw, d,
+                m_per_block, rpc);
+        else
+            k_stem_wgrad_mm<64><<<grid, 256, 0, stream>>>(
+                (const uint16_t*)x, (const uint16_t*)dout, (float*)dw, d,
+                m_per_block, rpc);
+        return;
+    }
     const int jchunks = (rsc + 31) / 32;
     int64_t msplit = (M + 511) / 512;
     const int64_t cap = 256 / jchunks > 0 ? 256 / jchunks : 1;
@@ -484,13 +671,19 @@ k_conv_stem_dgrad(const uint16_t* __restrict__ dout,
                 const int base = (r * d.S + s) * d.C;
                 for (int k = 0; k < 64; ++k) {
                     const float g = bf16_to_f32(gp[k]);
-                    for (int c = 0; c < d.C; ++c)
-                        acc[c] = fmaf(g, w_lds[k * rsc + base + c], acc[c]);
+                    // compile-time trip count keeps acc[] in registers
+                    // (guide rule 20); C <= 8 by the dispatch condition.
+#pragma unroll
+                    for (int c = 0; c < 8; ++c)
+                        if (c < d.C)
+                            acc[c] = fmaf(g, w_lds[k * rsc + base + c], acc[c]);
                 }
             }
         }
-        for (int c = 0; c < d.C; ++c)
-            dx[m * d.C + c] = f32_to_bf16(acc[c]);
+#pragma unroll
+        for (int c = 0; c < 8; ++c)
+            if (c < d.C)
+                dx[m * d.C + c] = f32_to_bf16(acc[c]);
     }
 }
 

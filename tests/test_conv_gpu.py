@@ -20,6 +20,7 @@ SHAPES = [
     (2, 4, 4, 256, 512, 3, 2, 1),
     (3, 7, 5, 64, 64, 3, 1, 1),       # non-pow2 spatial, odd M tail
     (2, 16, 16, 64, 128, 4, 2, 1),    # even kernel, stride 2 (GAN shapes)
+    (2, 32, 32, 3, 64, 7, 2, 3),      # ImageNet-style stem (C=3, chunked taps)
 ]
 
 
@@ -63,10 +64,13 @@ def test_conv_dgrad(shape):
     g = torch.Generator(device="cuda").manual_seed(1)
     dy = torch.randn(N, d.Ho, d.Wo, K, device="cuda", generator=g).to(torch.bfloat16)
 
-    wt = w.new_empty((d.R, d.S, d.C, d.K))
-    ops.weight_transpose(w, wt)
     dx = x.new_empty(x.shape)
-    ops.conv_dgrad(dy, wt, dx, d)
+    if C < 8:   # small-C edge path takes KRSC weights (as flashy_amd/nn.py)
+        ops.conv_stem_dgrad(dy, w, dx, d)
+    else:
+        wt = w.new_empty((d.R, d.S, d.C, d.K))
+        ops.weight_transpose(w, wt)
+        ops.conv_dgrad(dy, wt, dx, d)
     torch.cuda.synchronize()
 
     xr = x.float().permute(0, 3, 1, 2).requires_grad_(True)
