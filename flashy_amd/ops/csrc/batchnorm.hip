@@ -67,16 +67,20 @@ k_bn_stats(const uint16_t* __restrict__ x, float* __restrict__ partials,
 // threads = 64 channels x 4 split-lanes; contiguous per-channel reads.
 // ---------------------------------------------------------------------------
 
-__device__ __forceinline__ void combine_partials(
+// Block = 8 channels x 32 split-lanes (grid C/8): at C=64 that is 8 blocks
+// instead of 1, and each thread issues only ~msplit/16 float4 loads, so the
+// single-digit-block latency wall of a C/64 mapping disappears.  Threads
+// tid<8 leave with the channel's (s, s2); they are also the writers.
+__device__ __forceinline__ void combine_partials8(
         const float* __restrict__ partials, int msplit, int C,
         float* s_out, float* s2_out) {
-    const int c = blockIdx.x * 64 + (threadIdx.x & 63);
-    const int slane = threadIdx.x >> 6;  // 0..3
+    const int ch = threadIdx.x & 7;               // channel within block
+    const int slane = threadIdx.x >> 3;           // 0..31
+    const int c = blockIdx.x * 8 + ch;
     const float* row0 = partials + (int64_t)c * msplit;
     const float* row1 = partials + ((int64_t)C + c) * msplit;
     float s = 0.f, s2 = 0.f;
-    int i = slane * 4;
-    for (; i + 4 <= msplit; i += 16) {   // float4 rows: 4x fewer, wider loads
+    for (int i = slane * 4; i + 4 <= msplit; i += 128) {
         const float4 a = *reinterpret_cast<const float4*>(row0 + i);
         const float4 b = *reinterpret_cast<const float4*>(row1 + i);
         s += a.x + a.y + a.z + a.w;
@@ -87,14 +91,21 @@ __device__ __forceinline__ void combine_partials(
             s += row0[j];
             s2 += row1[j];
         }
-    __shared__ float red[2][4][64];
-    red[0][slane][threadIdx.x & 63] = s;
-    red[1][slane][threadIdx.x & 63] = s2;
+    __shared__ float red[2][32][8];
+    red[0][slane][ch] = s;
+    red[1][slane][ch] = s2;
     __syncthreads();
-    *s_out = red[0][0][threadIdx.x & 63] + red[0][1][threadIdx.x & 63] +
-             red[0][2][threadIdx.x & 63] + red[0][3][threadIdx.x & 63];
-    *s2_out = red[1][0][threadIdx.x & 63] + red[1][1][threadIdx.x & 63] +
-              red[1][2][threadIdx.x & 63] + red[1][3][threadIdx.x & 63];
+    if (threadIdx.x < 16) {
+        const int w = threadIdx.x >> 3;
+        const int cc = threadIdx.x & 7;
+        float acc = 0.f;
+#pragma unroll 8
+        for (int i = 0; i < 32; ++i) acc += red[w][i][cc];
+        red[w][0][cc] = acc;
+    }
+    __syncthreads();
+    *s_out = red[0][0][ch];
+    *s2_out = red[1][0][ch];
 }
 
 // work[0..C) = mean, [C..2C) = invstd, [2C..3C) = scale, [3C..4C) = shift
@@ -107,9 +118,9 @@ k_bn_finalize(const float* __restrict__ partials, int msplit,
               float* __restrict__ work, int64_t M, int C,
               float eps, float momentum, int update_running) {
     float s, s2;
-    combine_partials(partials, msplit, C, &s, &s2);
-    if (threadIdx.x >= 64) return;
-    const int c = blockIdx.x * 64 + threadIdx.x;
+    combine_partials8(partials, msplit, C, &s, &s2);
+    if (threadIdx.x >= 8) return;
+    const int c = blockIdx.x * 8 + threadIdx.x;
     const float mean = s / (float)M;
     float var = s2 / (float)M - mean * mean;
     var = fmaxf(var, 0.f);
@@ -226,9 +237,9 @@ k_bn_bwd_grads(const float* __restrict__ partials, int msplit,
                float* __restrict__ dgamma,
                float* __restrict__ dbeta, int C) {
     float s, sx;
-    combine_partials(partials, msplit, C, &s, &sx);
-    if (threadIdx.x >= 64) return;
-    const int c = blockIdx.x * 64 + threadIdx.x;
+    combine_partials8(partials, msplit, C, &s, &sx);
+    if (threadIdx.x >= 8) return;
+    const int c = blockIdx.x * 8 + threadIdx.x;
     bsums[c] = s;
     bsums[C + c] = sx;
     dbeta[c] += s;
@@ -293,7 +304,7 @@ extern "C" void launch_bn_finalize(const void* partials, int msplit,
                                    void* work, int64_t M, int C, float eps,
                                    float momentum, int update_running,
                                    hipStream_t stream) {
-    k_bn_finalize<<<C / 64, 256, 0, stream>>>(
+    k_bn_finalize<<<C / 8, 256, 0, stream>>>(
         (const float*)partials, msplit, (const float*)gamma,
         (const float*)beta, (float*)running_mean, (float*)running_var,
         (float*)work, M, C, eps, momentum, update_running);
@@ -341,7 +352,7 @@ extern "C" void launch_bn_bwd_reduce(const void* dy, const void* y,
 extern "C" void launch_bn_bwd_grads(const void* partials, int msplit,
                                     void* bsums, void* dgamma, void* dbeta,
                                     int C, hipStream_t stream) {
-    k_bn_bwd_grads<<<C / 64, 256, 0, stream>>>(
+    k_bn_bwd_grads<<<C / 8, 256, 0, stream>>>(
         (const float*)partials, msplit, (float*)bsums, (float*)dgamma,
         (float*)dbeta, C);
 }

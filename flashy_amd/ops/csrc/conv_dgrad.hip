@@ -529,28 +529,76 @@ int conv_subs_dg() {
 // Weight transpose  [K][R*S*C] -> [R*S*C][K]  (bf16).
 // ---------------------------------------------------------------------------
 
+__device__ __forceinline__ void transpose_tile64(
+        const uint16_t* __restrict__ w, uint16_t* __restrict__ wt,
+        int K, int rsc, int k0, int j0);
+
 __global__ void __launch_bounds__(256)
 k_weight_transpose(const uint16_t* __restrict__ w, uint16_t* __restrict__ wt,
                    int K, int rsc) {
-    const int64_t total = (int64_t)K * rsc;
-    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
-         i += stride) {
-        const int k = (int)(i / rsc);
-        const int j = (int)(i - (int64_t)k * rsc);
-        wt[(int64_t)j * K + k] = w[i];
-    }
+    const int tiles_j = (rsc + 63) / 64;
+    const int tk = blockIdx.x / tiles_j;
+    const int tj = blockIdx.x - tk * tiles_j;
+    transpose_tile64(w, wt, K, rsc, tk * 64, tj * 64);
 }
 
 extern "C" void launch_weight_transpose(const void* w, void* wt, int K, int rsc,
                                         hipStream_t stream) {
-    k_weight_transpose<<<ew_grid((int64_t)K * rsc, 256, 1), 256, 0, stream>>>(
+    const int tiles = ((K + 63) / 64) * ((rsc + 63) / 64);
+    k_weight_transpose<<<tiles, 256, 0, stream>>>(
         (const uint16_t*)w, (uint16_t*)wt, K, rsc);
+}
+
+// LDS-tiled 64x64 transpose step: both global sides are short8 (the naive
+// element loop scatters 2 B across 64 cachelines per wave -> ~3% write
+// efficiency; this version is bandwidth-bound).  Guarded for edge tiles.
+__device__ __forceinline__ void transpose_tile64(
+        const uint16_t* __restrict__ w, uint16_t* __restrict__ wt,
+        int K, int rsc, int k0, int j0) {
+    __shared__ __attribute__((aligned(16))) uint16_t t[64][72];
+    const int tid = threadIdx.x;
+    // read [k][j]: 64 rows x 8 short8; thread -> (row pair, j octet)
+    const int rr = tid >> 3;            // 0..31
+    const int oc = (tid & 7) * 8;       // j octet
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+        const int k = k0 + h * 32 + rr;
+        short8 v = {};
+        if (k < K && j0 + oc < rsc) {
+            if (j0 + oc + 8 <= rsc)
+                v = *reinterpret_cast<const short8*>(w + (int64_t)k * rsc + j0 + oc);
+            else
+                for (int e = 0; e < 8; ++e)
+                    if (j0 + oc + e < rsc)
+                        ((uint16_t*)&v)[e] = w[(int64_t)k * rsc + j0 + oc + e];
+        }
+        // transposed placement: element e lands at [j][k]
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+            t[oc + e][h * 32 + rr] = ((const uint16_t*)&v)[e];
+    }
+    __syncthreads();
+    // write [j][k]: thread -> (j row pair, k octet); short8 rows
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+        const int j = j0 + h * 32 + rr;
+        if (j < rsc && k0 + oc < K) {
+            const short8 v = *reinterpret_cast<const short8*>(&t[h * 32 + rr][oc]);
+            if (k0 + oc + 8 <= K)
+                *reinterpret_cast<short8*>(wt + (int64_t)j * K + k0 + oc) = v;
+            else
+                for (int e = 0; e < 8; ++e)
+                    if (k0 + oc + e < K)
+                        wt[(int64_t)j * K + k0 + oc + e] = ((const uint16_t*)&v)[e];
+        }
+    }
+    __syncthreads();   // LDS reused by the next tile of this block
 }
 
 // Batched transpose of EVERY conv weight in one launch: meta[i] =
 // {src_off, dst_off, K, rsc} (elements), src = the optimizer's flat bf16
-// mirror, dst = the shared RSCK arena.  grid.y = conv index.
+// mirror, dst = the shared RSCK arena.  grid.y = conv index; grid.x tiles
+// the largest conv and loops (grid-stride) over this conv's 64x64 tiles.
 __global__ void __launch_bounds__(256)
 k_weight_transpose_batched(const uint16_t* __restrict__ src_base,
                            uint16_t* __restrict__ dst_base,
@@ -560,13 +608,12 @@ k_weight_transpose_batched(const uint16_t* __restrict__ src_base,
     uint16_t* wt = dst_base + mi[1];
     const int K = mi[2];
     const int rsc = mi[3];
-    const int64_t total = (int64_t)K * rsc;
-    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
-         i += stride) {
-        const int k = (int)(i / rsc);
-        const int j = (int)(i - (int64_t)k * rsc);
-        wt[(int64_t)j * K + k] = w[i];
+    const int tiles_j = (rsc + 63) / 64;
+    const int tiles = ((K + 63) / 64) * tiles_j;
+    for (int t = blockIdx.x; t < tiles; t += gridDim.x) {
+        const int tk = t / tiles_j;
+        const int tj = t - tk * tiles_j;
+        transpose_tile64(w, wt, K, rsc, tk * 64, tj * 64);
     }
 }
 
@@ -574,7 +621,8 @@ extern "C" void launch_weight_transpose_batched(const void* src, void* dst,
                                                 const void* meta, int n_convs,
                                                 int64_t max_elems,
                                                 hipStream_t stream) {
-    dim3 grid((unsigned)ew_grid(max_elems, 256, 4), (unsigned)n_convs);
+    const int max_tiles = (int)((max_elems + 4095) / 4096);
+    dim3 grid((unsigned)(max_tiles < 1 ? 1 : max_tiles), (unsigned)n_convs);
     k_weight_transpose_batched<<<grid, 256, 0, stream>>>(
         (const uint16_t*)src, (uint16_t*)dst, (const int*)meta);
 }
