@@ -52,11 +52,11 @@ class NativeDCGANDiscriminator(nn.Module):
     def __init__(self, ndf: int = 64):
         super().__init__()
         self.conv1 = fnn.Conv2d(3, ndf, 4, 2, 1)            # RGB edge kernel
-        self.conv2 = fnn.Conv2d(ndf, ndf * 2, 4, 2, 1)
+        self.conv2 = fnn.Conv2d(ndf, ndf * 2, 4, 2, 1, feeds_bn=True)
         self.bn2 = fnn.BatchNorm2d(ndf * 2)
-        self.conv3 = fnn.Conv2d(ndf * 2, ndf * 4, 4, 2, 1)
+        self.conv3 = fnn.Conv2d(ndf * 2, ndf * 4, 4, 2, 1, feeds_bn=True)
         self.bn3 = fnn.BatchNorm2d(ndf * 4)
-        self.conv4 = fnn.Conv2d(ndf * 4, ndf * 8, 4, 2, 1)
+        self.conv4 = fnn.Conv2d(ndf * 4, ndf * 8, 4, 2, 1, feeds_bn=True)
         self.bn4 = fnn.BatchNorm2d(ndf * 8)
         self.head = nn.Linear(ndf * 8 * 4 * 4, 1)  # = the final 4x4 conv
 

@@ -25,12 +25,12 @@ class NativeBasicBlock(nn.Module):
     def __init__(self, cin: int, planes: int, stride: int = 1,
                  downsample: bool = False):
         super().__init__()
-        self.conv1 = fnn.Conv2d(cin, planes, 3, stride, 1)
+        self.conv1 = fnn.Conv2d(cin, planes, 3, stride, 1, feeds_bn=True)
         self.bn1 = fnn.BatchNorm2d(planes)
-        self.conv2 = fnn.Conv2d(planes, planes, 3, 1, 1)
+        self.conv2 = fnn.Conv2d(planes, planes, 3, 1, 1, feeds_bn=True)
         self.bn2 = fnn.BatchNorm2d(planes)
         if downsample:
-            self.dconv = fnn.Conv2d(cin, planes, 1, stride, 0)
+            self.dconv = fnn.Conv2d(cin, planes, 1, stride, 0, feeds_bn=True)
             self.dbn = fnn.BatchNorm2d(planes)
         else:
             self.dconv = None
@@ -51,14 +51,14 @@ class NativeBottleneck(nn.Module):
     def __init__(self, cin: int, planes: int, stride: int = 1,
                  downsample: bool = False):
         super().__init__()
-        self.conv1 = fnn.Conv2d(cin, planes, 1)
+        self.conv1 = fnn.Conv2d(cin, planes, 1, feeds_bn=True)
         self.bn1 = fnn.BatchNorm2d(planes)
-        self.conv2 = fnn.Conv2d(planes, planes, 3, stride, 1)
+        self.conv2 = fnn.Conv2d(planes, planes, 3, stride, 1, feeds_bn=True)
         self.bn2 = fnn.BatchNorm2d(planes)
-        self.conv3 = fnn.Conv2d(planes, planes * 4, 1)
+        self.conv3 = fnn.Conv2d(planes, planes * 4, 1, feeds_bn=True)
         self.bn3 = fnn.BatchNorm2d(planes * 4)
         if downsample:
-            self.dconv = fnn.Conv2d(cin, planes * 4, 1, stride, 0)
+            self.dconv = fnn.Conv2d(cin, planes * 4, 1, stride, 0, feeds_bn=True)
             self.dbn = fnn.BatchNorm2d(planes * 4)
         else:
             self.dconv = None

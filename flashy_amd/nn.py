@@ -47,11 +47,15 @@ def _grad_target(p: torch.Tensor) -> tp.Tuple[torch.Tensor, bool]:
 class _ConvFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x: torch.Tensor, w: torch.Tensor, stride: int, pad: int,
-                input_grad: bool, wt_cached: tp.Optional[torch.Tensor]):
+                input_grad: bool, wt_cached: tp.Optional[torch.Tensor],
+                want_stats: bool = False):
         w16 = _weight_bf16(w)
         d = ops.ConvDims.infer(x, w16, stride, pad)
         y = x.new_empty((d.N, d.Ho, d.Wo, d.K))
-        ops.conv_fwd(x, w16, y, d)
+        stats = ops.conv_fwd(x, w16, y, d, want_stats=want_stats)
+        if stats is not None:
+            # consumed by the following BatchNorm2d (skips its stats pass)
+            y._bn_stats = stats
         ctx.save_for_backward(x, w16)
         ctx.dims = d
         ctx.input_grad = input_grad
@@ -77,18 +81,22 @@ class _ConvFn(torch.autograd.Function):
                 ops.conv_dgrad(dy, wt, dx, d)
             else:  # small-C edge conv (e.g. a discriminator RGB stem)
                 ops.conv_stem_dgrad(dy, w16, dx, d)
-        return dx, None if direct else dw_buf, None, None, None, None
+        return dx, None if direct else dw_buf, None, None, None, None, None
 
 
 class Conv2d(nn.Module):
     """NHWC bf16 conv (no bias, as in ResNet).  Weight: [K, R, S, C] fp32."""
 
     def __init__(self, in_channels: int, out_channels: int, kernel_size: int,
-                 stride: int = 1, padding: int = 0, input_grad: bool = True):
+                 stride: int = 1, padding: int = 0, input_grad: bool = True,
+                 feeds_bn: bool = False):
         super().__init__()
         self.stride = stride
         self.padding = padding
         self.input_grad = input_grad
+        # when the conv output goes straight into a training BatchNorm2d,
+        # the conv epilogue emits the BN sum/sumsq partials for free
+        self.feeds_bn = feeds_bn
         self._wt_view: tp.Optional[torch.Tensor] = None
         k = kernel_size
         self.weight = nn.Parameter(
@@ -99,7 +107,7 @@ class Conv2d(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         return _ConvFn.apply(x, self.weight, self.stride, self.padding,
                              self.input_grad and x.requires_grad,
-                             self._wt_view)
+                             self._wt_view, self.feeds_bn and self.training)
 
 
 class WtCache:
@@ -161,11 +169,15 @@ class _BnFn(torch.autograd.Function):
         M = N * H * W
         y = torch.empty_like(x)
         if module.training:
-            msplit = ops.bn_msplit(M, C)
-            partials = torch.empty(msplit * 2 * C, dtype=torch.float32,
-                                   device=x.device)
+            fused = getattr(x, "_bn_stats", None)
+            if fused is not None:   # producing conv already emitted partials
+                partials, msplit = fused
+            else:
+                msplit = ops.bn_msplit(M, C)
+                partials = torch.empty(msplit * 2 * C, dtype=torch.float32,
+                                       device=x.device)
+                ops.bn_stats(x, partials, M, C, msplit)
             work = torch.empty(4 * C, dtype=torch.float32, device=x.device)
-            ops.bn_stats(x, partials, M, C, msplit)
             ops.bn_finalize(partials, msplit, gamma, beta, module.running_mean,
                             module.running_var, work, M, C, module.eps,
                             module.momentum, update_running=True)
@@ -259,11 +271,15 @@ class ConvTranspose2d(nn.Module):
     kernels."""
 
     def __init__(self, in_channels: int, out_channels: int, kernel_size: int,
-                 stride: int = 1, padding: int = 0, input_grad: bool = True):
+                 stride: int = 1, padding: int = 0, input_grad: bool = True,
+                 feeds_bn: bool = False):
         super().__init__()
         self.stride = stride
         self.padding = padding
         self.input_grad = input_grad
+        # when the conv output goes straight into a training BatchNorm2d,
+        # the conv epilogue emits the BN sum/sumsq partials for free
+        self.feeds_bn = feeds_bn
         k = kernel_size
         self.weight = nn.Parameter(
             torch.empty(in_channels, k, k, out_channels))

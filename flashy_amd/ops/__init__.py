@@ -157,7 +157,12 @@ def _splitk_plan(M: int, ktiles: int, red_stages: int):
 
 
 def conv_fwd(x: torch.Tensor, w: torch.Tensor, y: torch.Tensor, d: ConvDims,
-             relu: bool = False) -> None:
+             relu: bool = False,
+             want_stats: bool = False) -> tp.Optional[tp.Tuple[torch.Tensor, int]]:
+    """Forward conv.  With ``want_stats`` (non-split-K implicit-GEMM path
+    only) the epilogue also emits BatchNorm partials [2][K][msplit] summed
+    over each block's output rows; returns (partials, msplit) for
+    :func:`bn_finalize`, else None."""
     ext = require()
     if d.C % 8 == 0:
         rsc = d.R * d.S * d.C
@@ -174,12 +179,22 @@ def conv_fwd(x: torch.Tensor, w: torch.Tensor, y: torch.Tensor, d: ConvDims,
             ext.splitk_combine(ws.data_ptr(), y.data_ptr(), M * d.K, zeff, relu,
                                _stream())
         else:
+            stats = None
+            bn_ptr = 0
+            if want_stats:
+                msplit = ext.conv_fwd_msplit(*d)
+                partials = torch.empty(2 * d.K * msplit, dtype=torch.float32,
+                                       device=x.device)
+                stats = (partials, msplit)
+                bn_ptr = partials.data_ptr()
             ext.conv_fwd(x.data_ptr(), w.data_ptr(), y.data_ptr(), *d, relu,
-                         _stream())
+                         bn_ptr, _stream())
+            return stats
     else:
         assert not relu
         assert d.R * d.S * d.C <= 160 and d.K == 64, d
         ext.conv_stem_fwd(x.data_ptr(), w.data_ptr(), y.data_ptr(), *d, _stream())
+    return None
 
 
 def conv_stem_dgrad(dout: torch.Tensor, w_krsc: torch.Tensor,

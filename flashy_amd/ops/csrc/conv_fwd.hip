@@ -24,7 +24,7 @@ template <int BM, bool RELU, bool SPLITK, int SUBS = 2, int BN = CONV_BN>
 __global__ void __launch_bounds__(CONV_THREADS)
 k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
            uint16_t* __restrict__ y, float* __restrict__ ws_out,
-           ConvDims d, int stages_per_split) {
+           float* __restrict__ bn_ws, ConvDims d, int stages_per_split) {
     constexpr int WAVES_M = BM >= 64 ? 2 : 1;
     constexpr int WAVES_N = 4 / WAVES_M;
     constexpr int MF = BM / WAVES_M / 16;      // m fragments per wave
@@ -192,6 +192,48 @@ k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
                     }
                 }
             }
+
+    // Fused BatchNorm statistics: per-column sum/sumsq of this block's
+    // output tile -> partials[2][K][gridDim.x] at slice blockIdx.x (the
+    // same layout bn_finalize combines; out-of-range rows stage zeros so
+    // they contribute nothing).  Saves the bn_stats re-read of y.
+    if (!SPLITK && bn_ws != nullptr) {
+        float* sred = reinterpret_cast<float*>(A_lds);  // tile done: reuse
+#pragma unroll
+        for (int nf = 0; nf < NF; ++nf) {
+            float s = 0.f, s2 = 0.f;
+#pragma unroll
+            for (int mf = 0; mf < MF; ++mf)
+#pragma unroll
+                for (int rr = 0; rr < 4; ++rr) {
+                    float v = acc[mf][nf][rr];
+                    if (RELU) v = fmaxf(v, 0.f);
+                    s += v;
+                    s2 = fmaf(v, v, s2);
+                }
+            s += __shfl_xor(s, 16, 64);
+            s += __shfl_xor(s, 32, 64);
+            s2 += __shfl_xor(s2, 16, 64);
+            s2 += __shfl_xor(s2, 32, 64);
+            if (lane < 16) {
+                const int colL = wave_n * (BN / WAVES_N) + nf * 16 + lane;
+                sred[wave_m * BN + colL] = s;
+                sred[(WAVES_M + wave_m) * BN + colL] = s2;
+            }
+        }
+        __syncthreads();
+        if (tid < BN) {
+            float s = 0.f, s2 = 0.f;
+#pragma unroll
+            for (int wm = 0; wm < WAVES_M; ++wm) {
+                s += sred[wm * BN + tid];
+                s2 += sred[(WAVES_M + wm) * BN + tid];
+            }
+            const int c = col0 + tid;
+            bn_ws[((int64_t)c) * gridDim.x + blockIdx.x] = s;
+            bn_ws[((int64_t)d.K + c) * gridDim.x + blockIdx.x] = s2;
+        }
+    }
 }
 
 // combine split-K fp32 partials -> bf16 (+optional relu)
@@ -248,8 +290,18 @@ static int conv_subs() {   // A/B switch: FLASHY_CONV_SUBS=1 -> 32-deep stages
     return v;
 }
 
+// grid.x the fwd launcher will use for these dims (= the msplit of the
+// fused BN-stats partials); Python sizes the partials buffer with this.
+extern "C" int conv_fwd_msplit(ConvDims d) {
+    const int64_t M = (int64_t)d.N * d.Ho * d.Wo;
+    int bm, bn;
+    pick_tile(M, d.K, &bm, &bn);
+    return (int)((M + bm - 1) / bm);
+}
+
 extern "C" void launch_conv_fwd(const void* x, const void* w, void* y,
-                                ConvDims d, int relu, hipStream_t stream) {
+                                ConvDims d, int relu, void* bn_ws,
+                                hipStream_t stream) {
     const int64_t M = (int64_t)d.N * d.Ho * d.Wo;
     int bm, bn;
     pick_tile(M, d.K, &bm, &bn);
@@ -259,9 +311,9 @@ extern "C" void launch_conv_fwd(const void* x, const void* w, void* y,
         auto ww = (const uint16_t*)w;
         auto yy = (uint16_t*)y;
         if (relu)
-            k_conv_fwd<128, true, false, 2, 128><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, d, 0);
+            k_conv_fwd<128, true, false, 2, 128><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, (float*)bn_ws, d, 0);
         else
-            k_conv_fwd<128, false, false, 2, 128><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, d, 0);
+            k_conv_fwd<128, false, false, 2, 128><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, (float*)bn_ws, d, 0);
         return;
     }
     auto xx = (const uint16_t*)x;
@@ -269,24 +321,24 @@ extern "C" void launch_conv_fwd(const void* x, const void* w, void* y,
     auto yy = (uint16_t*)y;
     if (conv_subs() == 1) {
         if (relu) {
-            if (bm == 128) k_conv_fwd<128, true, false, 1><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, d, 0);
-            else if (bm == 64) k_conv_fwd<64, true, false, 1><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, d, 0);
-            else k_conv_fwd<32, true, false, 1><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, d, 0);
+            if (bm == 128) k_conv_fwd<128, true, false, 1><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, (float*)bn_ws, d, 0);
+            else if (bm == 64) k_conv_fwd<64, true, false, 1><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, (float*)bn_ws, d, 0);
+            else k_conv_fwd<32, true, false, 1><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, (float*)bn_ws, d, 0);
         } else {
-            if (bm == 128) k_conv_fwd<128, false, false, 1><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, d, 0);
-            else if (bm == 64) k_conv_fwd<64, false, false, 1><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, d, 0);
-            else k_conv_fwd<32, false, false, 1><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, d, 0);
+            if (bm == 128) k_conv_fwd<128, false, false, 1><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, (float*)bn_ws, d, 0);
+            else if (bm == 64) k_conv_fwd<64, false, false, 1><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, (float*)bn_ws, d, 0);
+            else k_conv_fwd<32, false, false, 1><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, (float*)bn_ws, d, 0);
         }
         return;
     }
     if (relu) {
-        if (bm == 128) k_conv_fwd<128, true, false><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, d, 0);
-        else if (bm == 64) k_conv_fwd<64, true, false><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, d, 0);
-        else k_conv_fwd<32, true, false><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, d, 0);
+        if (bm == 128) k_conv_fwd<128, true, false><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, (float*)bn_ws, d, 0);
+        else if (bm == 64) k_conv_fwd<64, true, false><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, (float*)bn_ws, d, 0);
+        else k_conv_fwd<32, true, false><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, (float*)bn_ws, d, 0);
     } else {
-        if (bm == 128) k_conv_fwd<128, false, false><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, d, 0);
-        else if (bm == 64) k_conv_fwd<64, false, false><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, d, 0);
-        else k_conv_fwd<32, false, false><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, d, 0);
+        if (bm == 128) k_conv_fwd<128, false, false><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, (float*)bn_ws, d, 0);
+        else if (bm == 64) k_conv_fwd<64, false, false><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, (float*)bn_ws, d, 0);
+        else k_conv_fwd<32, false, false><<<grid, CONV_THREADS, 0, stream>>>(xx, ww, yy, nullptr, (float*)bn_ws, d, 0);
     }
 }
 
@@ -300,7 +352,8 @@ extern "C" void launch_conv_fwd_splitk(const void* x, const void* w, void* ws,
     const int zeff = (all_stages + spz - 1) / spz;
     dim3 grid((unsigned)((M + 63) / 64), (unsigned)(d.K / CONV_BN), (unsigned)zeff);
     k_conv_fwd<64, false, true><<<grid, CONV_THREADS, 0, stream>>>(
-        (const uint16_t*)x, (const uint16_t*)w, nullptr, (float*)ws, d, spz);
+        (const uint16_t*)x, (const uint16_t*)w, nullptr, (float*)ws, nullptr,
+        d, spz);
 }
 
 // ---------------------------------------------------------------------------

@@ -22,7 +22,8 @@ struct ConvDims {
 
 extern "C" {
 void launch_conv_fwd(const void* x, const void* w, void* y, ConvDims d,
-                     int relu, hipStream_t stream);
+                     int relu, void* bn_ws, hipStream_t stream);
+int conv_fwd_msplit(ConvDims d);
 void launch_conv_stem_fwd(const void* x, const void* w, void* y, ConvDims d,
                           hipStream_t stream);
 void launch_conv_stem_wgrad(const void* x, const void* dout, void* dw,
@@ -105,11 +106,17 @@ PYBIND11_MODULE(_hip_ops, m) {
     m.def("conv_fwd",
           [](uintptr_t x, uintptr_t w, uintptr_t y, int N, int H, int W, int C,
              int K, int R, int S, int Ho, int Wo, int stride, int pad,
-             bool relu, uintptr_t stream) {
+             bool relu, uintptr_t bn_ws, uintptr_t stream) {
               launch_conv_fwd((const void*)x, (const void*)w, (void*)y,
                               make_dims(N, H, W, C, K, R, S, Ho, Wo, stride, pad),
-                              relu ? 1 : 0, as_stream(stream));
+                              relu ? 1 : 0, (void*)bn_ws, as_stream(stream));
               check_last();
+          });
+    m.def("conv_fwd_msplit",
+          [](int N, int H, int W, int C, int K, int R, int S, int Ho, int Wo,
+             int stride, int pad) {
+              return conv_fwd_msplit(
+                  make_dims(N, H, W, C, K, R, S, Ho, Wo, stride, pad));
           });
     m.def("conv_stem_fwd",
           [](uintptr_t x, uintptr_t w, uintptr_t y, int N, int H, int W, int C,

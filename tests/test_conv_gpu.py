@@ -244,3 +244,25 @@ def test_native_resnet_trains():
         opt.step()
         losses.append(loss.item())
     assert losses[-1] < losses[0] * 0.7, losses  # overfits a fixed batch
+
+
+@requires_gpu
+def test_conv_fused_bn_stats():
+    """The conv epilogue's fused BN partials must reproduce the standalone
+    bn_stats sums (finalize consumes either)."""
+    from flashy_amd import ops
+    N, H, W, C, K = 64, 16, 16, 64, 128   # M large: non-split-K path
+    x, w = _mk(N, H, W, C, K, 3)
+    d = ops.ConvDims.infer(x, w, 1, 1)
+    y = x.new_empty((d.N, d.Ho, d.Wo, d.K))
+    stats = ops.conv_fwd(x, w, y, d, want_stats=True)
+    assert stats is not None
+    partials, msplit = stats
+    torch.cuda.synchronize()
+    s = partials[:K * msplit].view(K, msplit).sum(1)
+    s2 = partials[K * msplit:].view(K, msplit).sum(1)
+    yf = y.float().reshape(-1, K)
+    assert torch.allclose(s, yf.sum(0), rtol=1e-3, atol=1.0), \
+        (s - yf.sum(0)).abs().max()
+    assert torch.allclose(s2, (yf * yf).sum(0), rtol=1e-3, atol=1.0), \
+        (s2 - (yf * yf).sum(0)).abs().max()
