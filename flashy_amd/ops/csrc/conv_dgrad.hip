@@ -34,7 +34,9 @@ k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_r
     const int wid = tid >> 6;
     const int wave_m = WAVES_M == 1 ? 0 : (wid >> 1);
     const int wave_n = WAVES_M == 1 ? wid : (wid & 1);
-    const int64_t m0 = (int64_t)blockIdx.x * BM;
+    unsigned bx = blockIdx.x;   // XCD-aware remap (see conv_fwd.hip)
+    if ((gridDim.x & 7) == 0) bx = (bx & 7) * (gridDim.x >> 3) + (bx >> 3);
+    const int64_t m0 = (int64_t)bx * BM;
     const int col0 = blockIdx.y * BN;
 
     __shared__ uint16_t A_lds[2][2 * BM * CONV_APITCH];

@@ -40,7 +40,13 @@ k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
     const int wid = tid >> 6;
     const int wave_m = WAVES_M == 1 ? 0 : (wid >> 1);
     const int wave_n = WAVES_M == 1 ? wid : (wid & 1);
-    const int64_t m0 = (int64_t)blockIdx.x * BM;
+    // XCD-aware tile order: blocks dispatch round-robin over the 8 XCDs,
+    // so giving block b the tile (b&7)*(grid/8)+(b>>3) makes each XCD's
+    // blocks work on CONSECUTIVE m-tiles — neighbouring tiles share input
+    // halo rows, which then hit that XCD's own L2.
+    unsigned bx = blockIdx.x;
+    if ((gridDim.x & 7) == 0) bx = (bx & 7) * (gridDim.x >> 3) + (bx >> 3);
+    const int64_t m0 = (int64_t)bx * BM;
     const int col0 = blockIdx.y * BN;
 
     // [sub][row][APITCH] per buffer: row pitch 48 = conflict-free b128 groups

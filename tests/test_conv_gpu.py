@@ -261,8 +261,14 @@ def test_conv_fused_bn_stats():
     torch.cuda.synchronize()
     s = partials[:K * msplit].view(K, msplit).sum(1)
     s2 = partials[K * msplit:].view(K, msplit).sum(1)
+    # the fused path sums the fp32 accumulators BEFORE the bf16 store, so
+    # it differs from sums of the rounded y by accumulated rounding noise:
+    # tolerance ~ sqrt(M) * bf16_eps * max|y|
     yf = y.float().reshape(-1, K)
-    assert torch.allclose(s, yf.sum(0), rtol=1e-3, atol=1.0), \
-        (s - yf.sum(0)).abs().max()
-    assert torch.allclose(s2, (yf * yf).sum(0), rtol=1e-3, atol=1.0), \
-        (s2 - (yf * yf).sum(0)).abs().max()
+    M = yf.shape[0]
+    tol = 0.02 * M ** 0.5 * yf.abs().max().item()
+    assert (s - yf.sum(0)).abs().max().item() < tol, \
+        ((s - yf.sum(0)).abs().max(), tol)
+    tol2 = 0.02 * M ** 0.5 * (yf * yf).max().item()
+    assert (s2 - (yf * yf).sum(0)).abs().max().item() < tol2, \
+        ((s2 - (yf * yf).sum(0)).abs().max(), tol2)
