@@ -51,6 +51,13 @@ k_conv_wgrad(const uint16_t* __restrict__ x, const uint16_t* __restrict__ dout,
     // covers 64 m = 2 subchunks of 32.
     const int m_r = tid >> 3;            // 0..31
     const int k8 = (tid & 7) * 8;        // channel octet
+    // this thread's im2col tap is FIXED (j0 + k8): decode it once
+    const int t_r = (j0 + k8) / (d.S * d.C);
+    const int t_scc = (j0 + k8) - t_r * d.S * d.C;
+    const int t_s = t_scc / d.C;
+    const int t_c = t_scc - t_s * d.C;
+    const int t_hoff = t_r - d.pad;      // hi = ho*stride + t_hoff
+    const int t_woff = t_s - d.pad;
     auto load_pair = [&](int64_t mc, short8* dv, short8* xv) {
 #pragma unroll
         for (int sc = 0; sc < 2; ++sc) {
@@ -61,19 +68,14 @@ k_conv_wgrad(const uint16_t* __restrict__ x, const uint16_t* __restrict__ dout,
             dv[sc] = v;
             short8 u = {};
             if (m < me) {
-                const int jj = j0 + k8;
-                const int r = jj / (d.S * d.C);
-                const int scc = jj - r * d.S * d.C;
-                const int s = scc / d.C;
-                const int c = scc - s * d.C;
                 const int wo = (int)(m % d.Wo);
                 const int ho = (int)((m / d.Wo) % d.Ho);
                 const int64_t n = m / ((int64_t)d.Ho * d.Wo);
-                const int hi = ho * d.stride + r - d.pad;
-                const int wi = wo * d.stride + s - d.pad;
+                const int hi = ho * d.stride + t_hoff;
+                const int wi = wo * d.stride + t_woff;
                 if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W)
                     u = *reinterpret_cast<const short8*>(
-                        x + (((n * d.H + hi) * d.W + wi) * (int64_t)d.C + c));
+                        x + (((n * d.H + hi) * d.W + wi) * (int64_t)d.C + t_c));
             }
             xv[sc] = u;
         }
