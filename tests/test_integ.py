@@ -69,3 +69,18 @@ def test_bench_distributed_contract(tmp_path):
     assert result["config"]["parallelism"] == "dp2"
     assert result["config"]["global_batch"] == 16
     assert result["value"] > 0
+
+
+def test_elastic_restart(tmp_path):
+    """The elastic wrapper relaunches a failing command until it succeeds."""
+    import subprocess, sys
+    marker = tmp_path / "tries"
+    prog = (
+        "import pathlib, sys; p = pathlib.Path(%r); "
+        "n = int(p.read_text()) if p.exists() else 0; "
+        "p.write_text(str(n + 1)); sys.exit(0 if n >= 2 else 3)" % str(marker))
+    rc = subprocess.call([sys.executable, "examples/elastic/run_elastic.py",
+                          "--backoff", "0.01", "--",
+                          sys.executable, "-c", prog])
+    assert rc == 0
+    assert marker.read_text() == "3"  # failed twice, succeeded third

@@ -45,3 +45,23 @@ def test_dcgan_shapes():
     assert img.shape == (2, 3, 64, 64)
     logits = d(img)
     assert logits.shape == (2,)
+
+
+def test_from_torch_imagenet_stem():
+    """The 7x7/s2 ImageNet stem maps from a torch twin: every native tensor
+    must equal its NCHW->KRSC-permuted source (pure tensor plumbing, CPU)."""
+    import torch
+    from flashy_amd.models import native_resnet50, resnet50
+    twin = resnet50(num_classes=100, small_input=False)
+    model = native_resnet50(100, imagenet_stem=True).from_torch(twin)
+    w = model.stem_conv.weight
+    assert w.shape == (64, 7, 7, 3)
+    ref = twin.stem[0].weight.permute(0, 2, 3, 1)
+    assert torch.equal(w.detach(), ref)
+    # spot-check a bottleneck conv and a BN pair deep in the net
+    assert torch.equal(model.layer3[0].conv2.weight.detach(),
+                       twin.layer3[0].conv2.weight.permute(0, 2, 3, 1))
+    assert torch.equal(model.layer4[1].bn3.weight.detach(),
+                       twin.layer4[1].bn3.weight)
+    assert torch.equal(model.layer1[0].dbn.running_var,
+                       twin.layer1[0].downsample[1].running_var)
