@@ -13,6 +13,9 @@ bench:
 	$(PYTHON) bench.py --steps 50 --warmup 20
 
 linter:
-	$(PYTHON) -m flake8 --max-line-length 100 flashy_amd tests examples bench.py || true
+	$(PYTHON) -m flake8 flashy_amd tests examples bench.py || true
 
-.PHONY: build tests tests_gpu bench linter
+docs:
+	$(PYTHON) -m pdoc flashy_amd -o docs/  # pdoc not in the offline image; CI target
+
+.PHONY: build tests tests_gpu bench linter docs
