@@ -32,4 +32,6 @@ struct ConvDims {
 #define CONV_THREADS 256 // 4 waves: 2 (m) x 2 (n)
 // LDS A-tile pitch in bf16 elements: 48 keeps every 16-lane ds_read_b128
 // group on distinct banks (see design notes) and 16B alignment.
-#define CONV_APITCH 48
+#define CONV_APITCH 40  // 80 B rows: 16 B-aligned b128, stride 20 dwords (16 distinct
+                        // banks mod 64), and the BM128 double buffer fits 4
+                        // blocks/CU (49 KB at pitch 48 capped it at 3)
