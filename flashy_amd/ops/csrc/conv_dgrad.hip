@@ -9,7 +9,8 @@
 //   (k_weight_transpose below) so each MFMA B-fragment lane reads 8
 //   consecutive k — contiguous 16 B.
 // Same pipelined 64-deep double-buffered single-barrier structure as the
-// forward kernel, with incremental (r,s,k) tap walk and B prefetch.
+// forward kernel (two staging register sets, loads 3 steps ahead), with
+// an incremental (r,s,k) tap walk; B fragments load at use.
 // Requires: K % 64 == 0, C % 64 == 0.
 
 #include "conv_common.h"

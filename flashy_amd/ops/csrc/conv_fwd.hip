@@ -8,13 +8,17 @@
 //       reads 8 consecutive rsc for its column K — 16 B contiguous, L2).
 //
 // Schedule per 64-deep stage (two MFMA-K subchunks), double-buffered LDS,
-// ONE barrier per stage: write regs(stage i+1) -> other buffer, issue
-// global loads for stage i+2 and the B fragments of stage i+1, then the
-// MFMA cluster over stage i.  Tap indices (r,s,c) advance INCREMENTALLY
-// (+64 with carry) — no integer divisions in the steady state.
+// ONE barrier per stage, TWO staging register sets: stage j's global
+// loads are issued at step j-3 and written to LDS at step j-1, so HBM
+// latency hides under two full stages; B fragments load at use (the
+// weight panel is L2-resident).  Tap indices (r,s,c) advance
+// INCREMENTALLY (+64 with carry) — no integer divisions steady-state.
 //
-// Tile template: BM in {128, 64, 32} x BN=64, 4 waves.  Smaller BM keeps
-// the deep ResNet layers (M as small as 1024) above ~208 workgroups.
+// Tile template: BM in {128, 64, 32} x BN in {64, 128}, 4 waves.  Smaller
+// BM keeps the deep ResNet layers (M as small as 1024) above ~208
+// workgroups; BN=128 halves A re-reads when K % 128 == 0 and the grid
+// stays full.  Split-K (grid.z stage slices + fp32 combine) kicks in for
+// underfilled grids and for long reductions (ops/__init__.py).
 // Requires: C % 8 == 0, K % 64 == 0, rsc % 32 == 0 (ResNet bodies; the
 // C=3 stem has its own direct kernels below).
 
