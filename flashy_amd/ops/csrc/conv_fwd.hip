@@ -53,7 +53,8 @@ k_conv_fwd(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
     const int64_t m0 = (int64_t)bx * BM;
     const int col0 = blockIdx.y * BN;
 
-    // [sub][row][APITCH] per buffer: row pitch 48 = conflict-free b128 groups
+    // [sub][row][APITCH] per buffer: 80 B rows (pitch 40) = aligned,
+    // conflict-free b128 groups
     __shared__ uint16_t A_lds[2][2 * BM * CONV_APITCH];
 
     // --- staging state: row geometry + incremental (r,s,c) tap walk ------
