@@ -172,6 +172,12 @@ class BaseSolver(ABC):
             self._current_formatter = None
 
     # -- commit / restore --------------------------------------------------
+    #: opt-in: overlap the pickle+disk half of the checkpoint with the next
+    #: epoch (device-to-host staging still happens inside commit, so the
+    #: state snapshot is consistent; durability is deferred one commit).
+    async_checkpoint: bool = False
+    _async_ckpt: tp.Optional[_checkpoint.AsyncCheckpointer] = None
+
     def commit(self, save_checkpoint: bool = True) -> None:
         """End the epoch: push pending metrics to history (every rank — the
         epoch counter must advance identically everywhere), then on rank 0
@@ -182,8 +188,18 @@ class BaseSolver(ABC):
             self.xp.link.update_history(self.history)
             if save_checkpoint:
                 state = self.stateful.state_dict()
-                _checkpoint.save_state(state, self.checkpoint_path)
+                if self.async_checkpoint:
+                    if self._async_ckpt is None:
+                        self._async_ckpt = _checkpoint.AsyncCheckpointer()
+                    self._async_ckpt.save(state, self.checkpoint_path)
+                else:
+                    _checkpoint.save_state(state, self.checkpoint_path)
                 self.logger.debug("checkpoint saved to %s", self.checkpoint_path)
+
+    def finalize_checkpoint(self) -> None:
+        """Join any in-flight async checkpoint write (call at run end)."""
+        if self._async_ckpt is not None:
+            self._async_ckpt.wait()
 
     def restore(self) -> bool:
         """Load + restore the checkpoint if one exists.  Returns True when a

@@ -106,3 +106,25 @@ def test_commit_restore_roundtrip(xp_root):
 
 def test_restore_without_checkpoint(solver):
     assert solver.restore() is False
+
+
+def test_async_checkpoint(xp_root):
+    """async_checkpoint=True defers the write but produces the identical
+    artifact; finalize_checkpoint() makes it durable."""
+    from flashy_amd import checkpoint as fckpt
+    fxp.create_xp(Config.wrap({"lr": 0.01})).enter()
+    s = TinySolver()
+    s.async_checkpoint = True
+    s.run(epochs=2)
+    s.finalize_checkpoint()
+    assert s.checkpoint_path.exists()
+    state = fckpt.load_state(s.checkpoint_path)
+    assert len(state["history"]) == 2
+
+    # a fresh sync solver restores the async-written artifact identically
+    fxp._current_xp = None
+    fxp.create_xp(Config.wrap({"lr": 0.01})).enter()
+    s2 = TinySolver()
+    assert s2.restore()
+    assert s2.epoch == 3
+    assert torch.equal(s2.model.weight, s.model.weight)
