@@ -295,6 +295,14 @@ def main_gan(args):
         g_optim.step()
         return g_loss
 
+    # single GPU: the whole G+D step (including the in-graph randn — HIP
+    # graphs replay graph-safe Philox offsets) is captured and replayed;
+    # the eager path's ~hundred launches per step were both slower and
+    # run-to-run noisy.
+    use_graph = native and ws == 1 and not args.no_graph
+    if use_graph:
+        step = CapturedStep(step, warmup=3).capture()
+
     for _ in range(args.warmup):
         step()
     distrib.barrier()
@@ -321,7 +329,7 @@ def main_gan(args):
             "dtype": "bf16" if native else "fp32", "data": "synthetic",
             "config": {"model": "dcgan64", "dataset": "synthetic-64x64",
                        "global_batch": ws * batch, "img_size": 64, "nz": nz,
-                       "parallelism": f"dp{ws}",
+                       "parallelism": f"dp{ws}", "graph": use_graph,
                        "mode": "native-kernels" if native else "reference-torch-ops"},
         }))
 
