@@ -30,7 +30,6 @@ import functools
 import io
 import logging
 import os
-import pickle
 import typing as tp
 from contextlib import contextmanager
 

@@ -67,7 +67,8 @@ class LocalFSLogger(ExperimentLogger):
         try:
             import torchaudio
             path = self._media_path(prefix, key, step, "wav")
-            torchaudio.save(str(path), audio.reshape(max(1, audio.shape[0] if audio.dim() > 1 else 1), -1),
+            n_ch = max(1, audio.shape[0] if audio.dim() > 1 else 1)
+            torchaudio.save(str(path), audio.reshape(n_ch, -1),
                             sample_rate)
         except ImportError:
             path = self._media_path(prefix, key, step, "pt")
