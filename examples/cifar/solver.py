@@ -41,6 +41,7 @@ class Solver(BaseSolver):
     def __init__(self, cfg, model, loaders, optim):
         super().__init__()
         self.cfg = cfg
+        self.async_checkpoint = bool(cfg.get('async_checkpoint', False))
         self.model = model
         self.loaders = loaders
         self.optim = optim
@@ -127,3 +128,4 @@ class Solver(BaseSolver):
             self.run_stage("train", self.do_train_valid, True)
             self.run_stage("valid", self.do_train_valid, False)
             self.commit()
+        self.finalize_checkpoint()   # join an in-flight async write, if any

@@ -128,3 +128,17 @@ def test_async_checkpoint(xp_root):
     assert s2.restore()
     assert s2.epoch == 3
     assert torch.equal(s2.model.weight, s.model.weight)
+
+
+def test_async_checkpoint_error_propagates(tmp_path):
+    """A failed background write must raise at the next wait()/save()."""
+    from flashy_amd.checkpoint import AsyncCheckpointer
+    ck = AsyncCheckpointer()
+    (tmp_path / "blocker").write_text("")   # a FILE where a dir is needed
+    ck.save({"x": torch.ones(4)}, tmp_path / "blocker" / "ck.th")
+    with pytest.raises(RuntimeError, match="async checkpoint"):
+        ck.wait()
+    # the checkpointer is reusable after the error
+    ck.save({"x": torch.ones(4)}, tmp_path / "ck.th")
+    ck.wait()
+    assert (tmp_path / "ck.th").exists()
