@@ -43,9 +43,13 @@ def _compact_cpu(t: torch.Tensor) -> torch.Tensor:
     return t
 
 
-def _stage_to_host(state: tp.Any) -> tp.Any:
+def _stage_to_host(state: tp.Any, snapshot_cpu: bool = False) -> tp.Any:
     """Deep-copy ``state`` with every CUDA tensor replaced by an async pinned
-    host copy; all copies are in flight before the final sync."""
+    host copy; all copies are in flight before the final sync.
+
+    ``snapshot_cpu`` additionally CLONES every CPU tensor: required when the
+    result outlives the call (async writer) — otherwise the background
+    pickle would read live tensors the next epoch is mutating."""
     pending: tp.List[tp.Tuple[torch.Tensor, torch.Tensor]] = []
 
     def _walk(obj: tp.Any) -> tp.Any:
@@ -54,6 +58,8 @@ def _stage_to_host(state: tp.Any) -> tp.Any:
                 host = torch.empty(obj.shape, dtype=obj.dtype, pin_memory=True)
                 pending.append((host, obj))
                 return host
+            if snapshot_cpu:
+                return obj.detach().clone()
             return _compact_cpu(obj)
         if isinstance(obj, dict):
             return {k: _walk(v) for k, v in obj.items()}
@@ -104,7 +110,7 @@ class AsyncCheckpointer:
 
     def save(self, state: tp.Any, path: tp.Union[str, Path]) -> None:
         self.wait()
-        host_state = _stage_to_host(state)
+        host_state = _stage_to_host(state, snapshot_cpu=True)
 
         def _write() -> None:
             try:

@@ -142,3 +142,22 @@ def test_async_checkpoint_error_propagates(tmp_path):
     ck.save({"x": torch.ones(4)}, tmp_path / "ck.th")
     ck.wait()
     assert (tmp_path / "ck.th").exists()
+
+
+def test_async_checkpoint_snapshot_isolation(xp_root):
+    """The async writer must snapshot CPU state at commit time: training
+    that continues during the background write may not leak into it."""
+    from flashy_amd import checkpoint as fckpt
+    fxp.create_xp(Config.wrap({"lr": 0.01})).enter()
+    s = TinySolver()
+    s.async_checkpoint = True
+    s.log_metrics("train", {"loss": 1.0})
+    s.commit()
+    snap = {k: v.detach().clone() for k, v in s.model.state_dict().items()}
+    with torch.no_grad():           # mutate AFTER commit, before the join
+        for p in s.model.parameters():
+            p.add_(123.0)
+    s.finalize_checkpoint()
+    state = fckpt.load_state(s.checkpoint_path)
+    for k, v in snap.items():
+        assert torch.equal(state["model"][k], v), k
