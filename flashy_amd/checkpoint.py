@@ -96,6 +96,12 @@ def save_state(state: tp.Any, path: tp.Union[str, Path]) -> None:
 class AsyncCheckpointer:
     """Overlap the pickle+disk half of a checkpoint with training.
 
+    EXPERIMENTAL: state-vs-file equality is verified
+    (scripts/async_ckpt_check.py, scripts/async_restore_check.py), but the
+    GPU cifar resume soak occasionally shows a first-epoch-after-resume
+    loss transient with this enabled that the sync path has not shown —
+    keep the default (sync) for production until that is root-caused.
+
     ``save()`` blocks only for the device-to-host staging (tens of ms),
     then serializes and atomically renames on a background thread;
     ``wait()`` joins the in-flight write (called automatically by the next

@@ -9,7 +9,9 @@ root = pathlib.Path("/tmp/soak_xp")
 env = dict(os.environ, _FLASHY_AMD_DIR=str(root), PYTHONPATH=".")
 base = [sys.executable, "-m", "examples.cifar.train",
         "epochs=2", "dataset_size=2048", "valid_size=512", "batch_size=64",
-        "run.exclude=[device,use_graph,epochs]", "async_checkpoint=true"]
+        "run.exclude=[device,use_graph,epochs]"]
+if "--async" in sys.argv:   # exercise the experimental async writer
+    base.append("async_checkpoint=true")
 rc = subprocess.call(base, env=env)
 assert rc == 0, rc
 rc = subprocess.call([a if a != "epochs=2" else "epochs=3" for a in base], env=env)
