@@ -155,7 +155,11 @@ def _splitk_plan(M: int, ktiles: int, red_stages: int):
     if red_stages < 4:
         return 0
     blocks64 = ((M + 63) // 64) * ktiles
-    fill = 0 if blocks64 >= 208 else         max(2, min((256 + blocks64 - 1) // max(1, blocks64), red_stages // 2))
+    if blocks64 >= 208:
+        fill = 0
+    else:
+        fill = max(2, min((256 + blocks64 - 1) // max(1, blocks64),
+                          red_stages // 2))
     longk = red_stages // 14 if red_stages >= 48 else 0
     zn = min(max(fill, longk), 8)
     return zn if zn >= 2 else 0
