@@ -65,3 +65,36 @@ def test_from_torch_imagenet_stem():
                        twin.layer4[1].bn3.weight)
     assert torch.equal(model.layer1[0].dbn.running_var,
                        twin.layer1[0].downsample[1].running_var)
+
+
+def test_wt_cache_arena_layout():
+    """WtCache packs every eligible conv's RSCK view into one arena with
+    offsets derived from the flat bf16 mirror — pure construction logic,
+    checkable on CPU (only refresh() needs the GPU kernel)."""
+    import torch
+    from flashy_amd import nn as fnn
+    from flashy_amd.models import native_resnet18
+    from flashy_amd.optim import FusedSGD
+    model = native_resnet18(10)
+    FusedSGD(model.parameters(), lr=0.1, bf16_mirror=True)
+    cache = fnn.WtCache(model)
+    assert cache.active
+    convs = [m for m in model.modules()
+             if isinstance(m, fnn.Conv2d) and m.input_grad
+             and m.weight.shape[-1] % 64 == 0]
+    assert cache.n == len(convs) and cache.n >= 15   # resnet18 body convs
+    total = sum(m.weight.numel() for m in convs)
+    assert cache.arena.numel() == total
+    meta = cache.meta.view(cache.n, 4)
+    seen_dst = set()
+    for c, (src_off, dst_off, K, rsc) in zip(convs, meta.tolist()):
+        Kw, R, S, C = c.weight.shape
+        assert (K, rsc) == (Kw, R * S * C)
+        # src offset points at this conv's mirror inside the flat buffer
+        mir = c.weight._bf16_mirror
+        assert (mir.data_ptr() - cache.src.data_ptr()) // 2 == src_off
+        # dst views alias the arena at dst_off, transposed to RSCK
+        assert c._wt_view.shape == (R, S, C, Kw)
+        assert c._wt_view.data_ptr() == cache.arena.data_ptr() + dst_off * 2
+        assert dst_off not in seen_dst
+        seen_dst.add(dst_off)
