@@ -124,6 +124,11 @@ def main():
     distrib.init()
     ws = distrib.world_size()
     rank = distrib.rank()
+    if args.gpus != ws:
+        raise SystemExit(
+            f"--gpus {args.gpus} but world size is {ws}: launch with "
+            f"torch.distributed.run --nproc-per-node {args.gpus} (a result "
+            f"labeled {args.gpus} GPUs must actually run on {args.gpus})")
     use_cuda = torch.cuda.is_available()
     device = torch.device("cuda", torch.cuda.current_device()) if use_cuda \
         else torch.device("cpu")
@@ -190,19 +195,28 @@ def main():
         static_y.copy_(ys[i % pool_n], non_blocking=True)
         return runner()
 
+    loss = None
     for i in range(args.warmup):
-        one_step(i)
+        loss = one_step(i)
+    # graph-mode numerical validity check: the loss the warmup trained with
+    # must be finite BEFORE we publish a throughput number
+    if use_cuda:
+        torch.cuda.synchronize()
+    if loss is not None and not bool(torch.isfinite(loss.detach()).all()):
+        raise SystemExit(f"non-finite loss after warmup: {loss}")
 
     distrib.barrier()
     if use_cuda:
         torch.cuda.synchronize()
     t0 = time.perf_counter()
     for i in range(args.steps):
-        one_step(i)
+        loss = one_step(i)
     distrib.barrier()
     if use_cuda:
         torch.cuda.synchronize()
     elapsed = time.perf_counter() - t0
+    if loss is not None and not bool(torch.isfinite(loss.detach()).all()):
+        raise SystemExit(f"non-finite loss after timed steps: {loss}")
 
     # max over ranks decides the whole-job time
     t = torch.tensor([elapsed], device=distrib.device(), dtype=torch.float64)
@@ -303,18 +317,25 @@ def main_gan(args):
     if use_graph:
         step = CapturedStep(step, warmup=3).capture()
 
+    g_loss = None
     for _ in range(args.warmup):
-        step()
+        g_loss = step()
+    if use_cuda:
+        torch.cuda.synchronize()
+    if g_loss is not None and not bool(torch.isfinite(g_loss.detach()).all()):
+        raise SystemExit(f"non-finite G loss after warmup: {g_loss}")
     distrib.barrier()
     if use_cuda:
         torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        step()
+        g_loss = step()
     distrib.barrier()
     if use_cuda:
         torch.cuda.synchronize()
     elapsed = time.perf_counter() - t0
+    if g_loss is not None and not bool(torch.isfinite(g_loss.detach()).all()):
+        raise SystemExit(f"non-finite G loss after timed steps: {g_loss}")
     t = torch.tensor([elapsed], device=distrib.device(), dtype=torch.float64)
     if ws > 1:
         torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)

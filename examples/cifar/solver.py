@@ -8,6 +8,8 @@ network; shapes and the training loop structure are the same).
 """
 from __future__ import annotations
 
+import math
+
 import torch
 from torch.nn import functional as F
 
@@ -117,6 +119,11 @@ class Solver(BaseSolver):
                 loss, acc = self._step(img, label, train)
                 metrics = avg({"loss": loss.item(), "acc": acc.item()})
                 lp.update(**metrics)
+        if not math.isfinite(metrics["loss"]):
+            # a NaN must halt the run loudly, not be silently averaged —
+            # especially with graph capture in the loop
+            raise RuntimeError(
+                f"non-finite {stage} loss at epoch {self.epoch}: {metrics}")
         return distrib.average_metrics(metrics, len(loader))
 
     def run(self):

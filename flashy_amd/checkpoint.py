@@ -26,9 +26,48 @@ import torch
 from .utils import write_and_rename
 
 _copy_stream: tp.Optional["torch.cuda.Stream"] = None
-# Reused pinned staging buffers keyed by (dtype) — grown on demand, kept for
-# the life of the process so repeated commits pay no allocation cost.
-_pinned_pool: tp.Dict[torch.dtype, torch.Tensor] = {}
+
+
+class _PinnedPool:
+    """Reused pinned staging buffers: an exact-byte-size free list.
+
+    Consecutive checkpoints stage the same tensor set, so after the first
+    save every ``acquire`` is a free-list pop — zero pinned allocation
+    (pinned allocation is synchronous and expensive; at 288 GB-class states
+    it would dominate the save).  Each buffer is its OWN storage (not a view
+    of a shared arena) so ``torch.save`` stays compact and mixed-dtype-safe.
+    ``recycle()`` marks every previously acquired buffer reusable — call it
+    only once the previous save's bytes are on disk.  Not thread-safe: each
+    concurrent writer owns its own pool (the module-level one backs
+    ``save_state``; every :class:`AsyncCheckpointer` owns a private one so
+    its background pickle can never be clobbered by a later save reusing
+    the buffers)."""
+
+    def __init__(self) -> None:
+        self._free: tp.Dict[int, tp.List[torch.Tensor]] = {}
+        self._inuse: tp.List[torch.Tensor] = []
+
+    def recycle(self) -> None:
+        for buf in self._inuse:
+            self._free.setdefault(buf.numel(), []).append(buf)
+        self._inuse = []
+
+    def acquire(self, shape: tp.Sequence[int], dtype: torch.dtype) -> torch.Tensor:
+        shape = tuple(shape)
+        nbytes = int(torch.Size(shape).numel()) * torch._utils._element_size(dtype)
+        if nbytes == 0:
+            return torch.empty(shape, dtype=dtype)
+        lst = self._free.get(nbytes)
+        if lst:
+            buf = lst.pop()
+        else:  # pinned on GPU boxes; plain host memory in CPU-only CI
+            buf = torch.empty(nbytes, dtype=torch.uint8,
+                              pin_memory=torch.cuda.is_available())
+        self._inuse.append(buf)
+        return buf.view(dtype).view(shape)
+
+
+_pinned_pool = _PinnedPool()
 
 
 def _compact_cpu(t: torch.Tensor) -> torch.Tensor:
@@ -43,19 +82,26 @@ def _compact_cpu(t: torch.Tensor) -> torch.Tensor:
     return t
 
 
-def _stage_to_host(state: tp.Any, snapshot_cpu: bool = False) -> tp.Any:
+def _stage_to_host(state: tp.Any, snapshot_cpu: bool = False,
+                   pool: tp.Optional[_PinnedPool] = None) -> tp.Any:
     """Deep-copy ``state`` with every CUDA tensor replaced by an async pinned
     host copy; all copies are in flight before the final sync.
 
     ``snapshot_cpu`` additionally CLONES every CPU tensor: required when the
     result outlives the call (async writer) — otherwise the background
-    pickle would read live tensors the next epoch is mutating."""
+    pickle would read live tensors the next epoch is mutating.
+    ``pool`` reuses pinned staging buffers across saves (second save of the
+    same state allocates zero pinned memory)."""
     pending: tp.List[tp.Tuple[torch.Tensor, torch.Tensor]] = []
 
     def _walk(obj: tp.Any) -> tp.Any:
         if torch.is_tensor(obj):
             if obj.is_cuda:
-                host = torch.empty(obj.shape, dtype=obj.dtype, pin_memory=True)
+                if pool is not None:
+                    host = pool.acquire(obj.shape, obj.dtype)
+                else:
+                    host = torch.empty(obj.shape, dtype=obj.dtype,
+                                       pin_memory=True)
                 pending.append((host, obj))
                 return host
             if snapshot_cpu:
@@ -88,7 +134,8 @@ def save_state(state: tp.Any, path: tp.Union[str, Path]) -> None:
     torch.save framing: no per-record CRC32 pass, ~40% less serialize time
     and ~2x faster loads at identical ``torch.load`` compatibility.
     """
-    state = _stage_to_host(state)
+    _pinned_pool.recycle()  # previous save_state is on disk — buffers reusable
+    state = _stage_to_host(state, pool=_pinned_pool)
     with write_and_rename(path) as fh:
         torch.save(state, fh, _use_new_zipfile_serialization=False)
 
@@ -113,10 +160,12 @@ class AsyncCheckpointer:
     def __init__(self) -> None:
         self._thread: tp.Optional[threading.Thread] = None
         self._error: tp.Optional[BaseException] = None
+        self._pool = _PinnedPool()  # private: background pickle reads from it
 
     def save(self, state: tp.Any, path: tp.Union[str, Path]) -> None:
-        self.wait()
-        host_state = _stage_to_host(state, snapshot_cpu=True)
+        self.wait()  # previous write durable -> pool buffers reusable
+        self._pool.recycle()
+        host_state = _stage_to_host(state, snapshot_cpu=True, pool=self._pool)
 
         def _write() -> None:
             try:

@@ -422,7 +422,21 @@ class _EagerSync:
         for h in self._hooks:
             h.remove()
         self._hooks.clear()
-        if not is_distributed() or exc_type is not None:
+        if not is_distributed():
+            return
+        if exc_type is not None:
+            # Error path: DRAIN in-flight async all-reduces before propagating.
+            # Leaving them orphaned lets the next collective on this
+            # communicator interleave with them and deadlock ranks that did
+            # not throw.  Grads are garbage anyway — just wait.
+            for _params, _flat, handle, _done in self._inflight:
+                try:
+                    handle.wait()
+                except Exception:  # noqa: BLE001 — already propagating exc
+                    pass
+            self._inflight.clear()
+            self._pending = []
+            self._pending_bytes = 0
             return
         # Deadlock guard on the un-fired set: all ranks must agree on how many
         # params never produced a grad (reference distrib.py:186).

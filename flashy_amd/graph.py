@@ -59,16 +59,30 @@ class CapturedStep:
     def capture(self) -> "CapturedStep":
         if not self.enabled:
             return self
-        side = torch.cuda.Stream()
-        side.wait_stream(torch.cuda.current_stream())
-        with torch.cuda.stream(side):
-            for _ in range(self.warmup):
-                self.fn()
-        torch.cuda.current_stream().wait_stream(side)
-        torch.cuda.synchronize()
-        self.graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.graph, pool=self.pool):
-            self.output = self.fn()
+        # The autocast weight cache is incompatible with graph capture
+        # (same rule as torch.cuda.make_graphed_callables): cached casts
+        # allocated during capture are freed into the graph's private pool
+        # at autocast exit, and later reuse of those blocks can race with
+        # multi-stream library kernels on replay — observed as an
+        # intermittent, box-dependent NaN after a few replays (GPUTEST_r01).
+        # With the cache off, every cast is a recorded kernel whose output
+        # block stays live for the whole graph.
+        prev_cache = torch.is_autocast_cache_enabled()
+        torch.set_autocast_cache_enabled(False)
+        torch.clear_autocast_cache()
+        try:
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                for _ in range(self.warmup):
+                    self.fn()
+            torch.cuda.current_stream().wait_stream(side)
+            torch.cuda.synchronize()
+            self.graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self.graph, pool=self.pool):
+                self.output = self.fn()
+        finally:
+            torch.set_autocast_cache_enabled(prev_cache)
         return self
 
     def replay(self) -> tp.Any:

@@ -68,8 +68,11 @@ class _ConvFn(torch.autograd.Function):
         x, w16 = ctx.saved_tensors
         d: ops.ConvDims = ctx.dims
         dy = dy.contiguous()
-        dw_buf, direct = _grad_target(ctx.w_ref)
-        ops.conv_wgrad(x, dy, dw_buf, d)
+        dw_buf = None
+        direct = True
+        if ctx.needs_input_grad[1]:  # skip wgrad entirely for frozen weights
+            dw_buf, direct = _grad_target(ctx.w_ref)
+            ops.conv_wgrad(x, dy, dw_buf, d)
         dx = None
         if ctx.input_grad:
             dx = x.new_empty(x.shape)
@@ -81,7 +84,8 @@ class _ConvFn(torch.autograd.Function):
                 ops.conv_dgrad(dy, wt, dx, d)
             else:  # small-C edge conv (e.g. a discriminator RGB stem)
                 ops.conv_stem_dgrad(dy, w16, dx, d)
-        return dx, None if direct else dw_buf, None, None, None, None, None
+        return (dx, dw_buf if dw_buf is not None and not direct else None,
+                None, None, None, None, None)
 
 
 class Conv2d(nn.Module):
@@ -208,14 +212,20 @@ class _BnFn(torch.autograd.Function):
         dz = torch.empty_like(dy)
         ops.bn_bwd_reduce(dy, y, x, work, dz, partials, M, C, msplit,
                           ctx.relu, ctx.slope)
-        dgamma, g_direct = _grad_target(gamma)
-        dbeta, b_direct = _grad_target(beta)
+        # bn_bwd_grads also produces bsums (needed for dx), so the kernel
+        # always runs; frozen gamma/beta just accumulate into a discarded
+        # temp instead of param.grad (never pollute frozen params' grads).
+        need_g, need_b = ctx.needs_input_grad[1], ctx.needs_input_grad[2]
+        dgamma, g_direct = _grad_target(gamma) if need_g \
+            else (torch.empty_like(gamma), True)
+        dbeta, b_direct = _grad_target(beta) if need_b \
+            else (torch.empty_like(beta), True)
         ops.bn_bwd_grads(partials, msplit, bsums, dgamma, dbeta, C)
         dx = torch.empty_like(x)
         ops.bn_bwd_apply(dz, x, work, bsums, dx, M, C)
         return (dx,
-                None if g_direct else dgamma,
-                None if b_direct else dbeta,
+                dgamma if need_g and not g_direct else None,
+                dbeta if need_b and not b_direct else None,
                 dz if ctx.has_res else None,
                 None, None, None)
 
@@ -254,13 +264,17 @@ class _ConvTransposeFn(torch.autograd.Function):
         x, w16 = ctx.saved_tensors
         d: ops.ConvDims = ctx.dims
         dy = dy.contiguous()
-        dw_buf, direct = _grad_target(ctx.w_ref)
-        ops.conv_wgrad(dy, x, dw_buf, d)   # roles swapped vs regular conv
+        dw_buf = None
+        direct = True
+        if ctx.needs_input_grad[1]:  # skip wgrad entirely for frozen weights
+            dw_buf, direct = _grad_target(ctx.w_ref)
+            ops.conv_wgrad(dy, x, dw_buf, d)   # roles swapped vs regular conv
         dx = None
         if ctx.input_grad:
             dx = x.new_empty(x.shape)
             ops.conv_fwd(dy, w16, dx, d)
-        return dx, None if direct else dw_buf, None, None, None
+        return (dx, dw_buf if dw_buf is not None and not direct else None,
+                None, None, None)
 
 
 class ConvTranspose2d(nn.Module):

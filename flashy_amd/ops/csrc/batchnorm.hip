@@ -80,17 +80,22 @@ __device__ __forceinline__ void combine_partials8(
     const float* row0 = partials + (int64_t)c * msplit;
     const float* row1 = partials + ((int64_t)C + c) * msplit;
     float s = 0.f, s2 = 0.f;
-    for (int i = slane * 4; i + 4 <= msplit; i += 128) {
-        const float4 a = *reinterpret_cast<const float4*>(row0 + i);
-        const float4 b = *reinterpret_cast<const float4*>(row1 + i);
-        s += a.x + a.y + a.z + a.w;
-        s2 += b.x + b.y + b.z + b.w;
-    }
-    if (slane == 0)  // scalar tail when msplit % 4 != 0
-        for (int j = msplit & ~3; j < msplit; ++j) {
-            s += row0[j];
-            s2 += row1[j];
+    if (msplit & 3) {
+        // row base c*msplit is 16B-aligned only when msplit % 4 == 0 (the
+        // fused conv-epilogue path passes msplit = conv grid.x, which is
+        // not forced to a multiple of 4) -> scalar loads for odd msplit.
+        for (int i = slane; i < msplit; i += 32) {
+            s += row0[i];
+            s2 += row1[i];
         }
+    } else {
+        for (int i = slane * 4; i + 4 <= msplit; i += 128) {
+            const float4 a = *reinterpret_cast<const float4*>(row0 + i);
+            const float4 b = *reinterpret_cast<const float4*>(row1 + i);
+            s += a.x + a.y + a.z + a.w;
+            s2 += b.x + b.y + b.z + b.w;
+        }
+    }
     __shared__ float red[2][32][8];
     red[0][slane][ch] = s;
     red[1][slane][ch] = s2;

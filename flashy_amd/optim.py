@@ -254,6 +254,16 @@ class FusedAdam(FlatOptimizer):
             ops.adam_step_inc(dev_step)
         return out
 
+    def state_dict(self):
+        if self._step_dev:
+            # The device counter is authoritative: under HIP-graph capture
+            # step() runs only at capture time, so the host count freezes
+            # while the device counter advances per replay.  It holds the
+            # NEXT step number (starts at 1 before step #1).
+            self.step_count = max(
+                int(s.item()) for s in self._step_dev.values()) - 1
+        return super().state_dict()
+
     def _extra_state(self):
         return {"exp_avg": self._exp_avg, "exp_avg_sq": self._exp_avg_sq}
 

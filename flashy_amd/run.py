@@ -83,9 +83,10 @@ def main(argv: tp.Optional[tp.Sequence[str]] = None) -> int:
         return run_workers(ns.package, workers, forwarded)
 
     entry = _resolve_main(ns.package)
-    # Only rank 0 clears; other ranks wait for the folder via init/barrier.
-    clear = ns.clear and int(os.environ.get("RANK", 0)) == 0
-    entry.run(ns.overrides, clear=clear)
+    # All ranks pass clear=True: EntryPoint.run rendezvous-gates the wipe
+    # (rank 0 clears between two barriers) so no rank observes a half-
+    # cleared XP.
+    entry.run(ns.overrides, clear=ns.clear)
     return 0
 
 

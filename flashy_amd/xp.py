@@ -168,7 +168,18 @@ class EntryPoint:
         xp = self.get_xp(overrides)
         if clear:
             import shutil
-            shutil.rmtree(xp.folder, ignore_errors=True)
+            from . import distrib
+            if int(os.environ.get("WORLD_SIZE", "1")) > 1:
+                # Multi-worker --clear: rendezvous so no rank can be inside
+                # the XP (mkdir/history/restore) while rank 0 wipes it, and
+                # no rank enters before the wipe is complete.
+                distrib.init()
+                distrib.barrier()
+                if distrib.is_rank_zero():
+                    shutil.rmtree(xp.folder, ignore_errors=True)
+                distrib.barrier()
+            else:
+                shutil.rmtree(xp.folder, ignore_errors=True)
         xp.enter()
         # Persist the resolved config so get_xp_from_sig can rebuild the XP.
         # pid-suffixed temp: concurrent DDP workers of one run write the same
