@@ -20,6 +20,7 @@ from __future__ import annotations
 import argparse
 import json
 import os
+import sys
 import time
 
 os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
@@ -169,19 +170,41 @@ def main():
     autocast = use_cuda and not native  # the native model is bf16 internally
     use_graph = use_cuda and not args.no_graph and not args.ref
 
+    dp_sync = f"dp{ws}"
     if use_graph and ws > 1 and isinstance(optim, FusedSGD):
-        # multi-GPU: capture fwd+bwd as one graph (grads land in the static
-        # flat buffers), then RCCL all-reduce + fused optimizer step eagerly
-        # — collectives stay outside the graph.
-        fwd_bwd = CapturedStep(
-            build_fwd_bwd(model, optim, static_x, static_y, autocast),
-            warmup=3).capture()
+        # multi-GPU: ONE graph for the whole step — fwd + bwd with the
+        # chunked flat all-reduce overlapping backward (RCCL collectives
+        # are recorded into the graph; validated by scripts/rccl_probe.py)
+        # + the fused optimizer step.  Fallback: fwd+bwd graph with the
+        # post-hoc single all-reduce outside the graph.
+        fwd_bwd = build_fwd_bwd(model, optim, static_x, static_y, autocast)
+        runner = None
+        if os.environ.get("FLASHY_AMD_DP_MODE", "graph-overlap") == "graph-overlap":
+            sync = distrib.OverlappedFlatSync(optim)
 
-        def runner():
-            loss = fwd_bwd()
-            distrib.sync_flat_gradients(optim)
-            optim.step()
-            return loss
+            def overlapped_step():
+                loss = fwd_bwd()
+                sync.finish()
+                optim.step()
+                return loss
+
+            try:
+                runner = CapturedStep(overlapped_step, warmup=3).capture()
+                dp_sync = f"dp{ws}-overlap{sync.n_chunks}ch"
+            except Exception as exc:  # noqa: BLE001 — e.g. RCCL refuses capture
+                print(f"[bench] in-graph overlapped sync failed ({exc!r}); "
+                      "falling back to post-hoc all-reduce", file=sys.stderr)
+                sync.remove()
+                runner = None
+        if runner is None:
+            graphed_fwd_bwd = CapturedStep(fwd_bwd, warmup=3).capture()
+            dp_sync = f"dp{ws}-posthoc"
+
+            def runner():
+                loss = graphed_fwd_bwd()
+                distrib.sync_flat_gradients(optim)
+                optim.step()
+                return loss
     elif use_graph and ws == 1:
         runner = CapturedStep(
             build_step(model, optim, static_x, static_y, autocast, False),
@@ -249,7 +272,7 @@ def main():
                 "global_batch": ws * args.batch,
                 "img_size": args.img,
                 "num_classes": args.classes,
-                "parallelism": f"dp{ws}",
+                "parallelism": dp_sync,
                 "mode": ("reference-torch-ops" if args.ref
                          else "native-kernels" if native else "torch-model"),
                 "graph": use_graph,

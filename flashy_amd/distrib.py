@@ -323,6 +323,123 @@ def sync_flat_gradients(optimizer) -> None:
         g.div_(ws)
 
 
+class OverlappedFlatSync:
+    """Backward-overlapped DP gradient sync for a
+    :class:`flashy_amd.optim.FlatOptimizer`.
+
+    The flat gradient buffer is partitioned into contiguous chunks of
+    ``chunk_bytes`` following parameter registration order.  Each parameter
+    gets a persistent ``register_post_accumulate_grad_hook``; the moment the
+    last parameter of a chunk has accumulated its gradient inside
+    ``backward()``, that chunk's slice of the flat buffer is all-reduced
+    asynchronously (RCCL launches on its own stream, ordered after the
+    compute stream at the issue point) — so communication overlaps the
+    remainder of backward.  ``finish()`` waits the in-flight reduces and
+    divides the whole buffer by the world size (sum-then-divide: numerics
+    identical to the post-hoc path and the reference oracle).
+
+    Because chunks are contiguous slices of the already-flat buffer there is
+    **zero packing copy** — the all-reduce reads/writes the gradient storage
+    in place.
+
+    The whole step (backward with its chunk flushes + ``finish`` +
+    ``optimizer.step``) is HIP-graph-capturable: hooks run at capture time
+    and the recorded graph replays the overlapped schedule, collectives
+    included (validated by scripts/rccl_probe.py: RCCL all-reduce inside
+    ``torch.cuda.graph`` capture works on this stack).
+
+    ResNet registration order puts ~75% of the bytes in the deep layers
+    whose grads complete EARLY in backward, so most bytes are in flight
+    while the wide early-layer backward still runs.
+
+    Usage (per step, capturable)::
+
+        sync = OverlappedFlatSync(optim)      # once; installs hooks
+        ...
+        optim.zero_grad(set_to_none=False)
+        loss.backward()                       # chunks flush as they complete
+        sync.finish()                         # wait + /world_size
+        optim.step()
+
+    Everything is a no-op at world_size == 1 (hooks are not installed).
+    """
+
+    def __init__(self, optimizer, chunk_bytes: tp.Optional[int] = None):
+        if chunk_bytes is None:
+            chunk_bytes = int(os.environ.get("FLASHY_AMD_CHUNK_MB", "8")) << 20
+        self.optimizer = optimizer
+        self.chunk_bytes = chunk_bytes
+        self._hooks: tp.List[tp.Any] = []
+        # segment: [flat_g, start, length, n_params]; countdowns reset per step
+        self._segments: tp.List[tp.List[tp.Any]] = []
+        self._seg_of: tp.Dict[int, int] = {}
+        self._remaining: tp.List[int] = []
+        self._handles: tp.List[tp.Any] = []
+        if not is_distributed():
+            return
+        for group in optimizer.groups:
+            offset = 0
+            seg_start, seg_params, seg_bytes = 0, 0, 0
+            for p in group.params:
+                n = p.numel()
+                seg_params += 1
+                seg_bytes += n * p.grad.element_size()
+                self._seg_of[id(p)] = len(self._segments)
+                offset += n
+                if seg_bytes >= self.chunk_bytes:
+                    self._segments.append(
+                        [group.flat_g, seg_start, offset - seg_start, seg_params])
+                    seg_start, seg_params, seg_bytes = offset, 0, 0
+            if seg_params:
+                self._segments.append(
+                    [group.flat_g, seg_start, offset - seg_start, seg_params])
+        self._remaining = [s[3] for s in self._segments]
+        for group in optimizer.groups:
+            for p in group.params:
+                self._hooks.append(
+                    p.register_post_accumulate_grad_hook(self._on_grad))
+
+    @property
+    def n_chunks(self) -> int:
+        return len(self._segments)
+
+    def _on_grad(self, param: torch.nn.Parameter) -> None:
+        i = self._seg_of[id(param)]
+        self._remaining[i] -= 1
+        if self._remaining[i] == 0:
+            flat_g, start, length, _ = self._segments[i]
+            self._handles.append(dist.all_reduce(
+                flat_g.narrow(0, start, length), op=dist.ReduceOp.SUM,
+                async_op=True))
+
+    def finish(self) -> None:
+        """Wait in-flight chunk reduces, divide by world size, reset."""
+        if not is_distributed():
+            return
+        # flush stragglers: a param that never produced a grad leaves its
+        # segment incomplete — reduce it anyway (the grad bytes are the
+        # zeros zero_grad wrote) so the collective schedule stays aligned
+        # across ranks regardless of which params got grads
+        for i, rem in enumerate(self._remaining):
+            if rem > 0:
+                flat_g, start, length, _ = self._segments[i]
+                self._handles.append(dist.all_reduce(
+                    flat_g.narrow(0, start, length), op=dist.ReduceOp.SUM,
+                    async_op=True))
+        for h in self._handles:
+            h.wait()
+        self._handles.clear()
+        ws = world_size()
+        for group in self.optimizer.groups:
+            group.flat_g.div_(ws)
+        self._remaining = [s[3] for s in self._segments]
+
+    def remove(self) -> None:
+        for h in self._hooks:
+            h.remove()
+        self._hooks.clear()
+
+
 # ---------------------------------------------------------------------------
 # Eager (overlapped) gradient sync — comm on a side HIP stream during backward
 # ---------------------------------------------------------------------------

@@ -162,6 +162,80 @@ def _check_flat_optimizer_equivalence(rank: int, ws: int):
         assert torch.allclose(p, q, atol=1e-5), (p - q).abs().max()
 
 
+def _check_overlapped_flat_sync(rank: int, ws: int):
+    """OverlappedFlatSync (chunked, backward-overlapped flat all-reduce)
+    matches torch SGD on the virtual batch — with chunk_bytes small enough
+    that every chunk flushes from inside backward."""
+    from flashy_amd.optim import FusedSGD
+    model = _make_model()
+    opt = FusedSGD(model.parameters(), lr=0.05, momentum=0.9)
+    sync = distrib.OverlappedFlatSync(opt, chunk_bytes=64)
+    assert sync.n_chunks >= 3  # the point is multiple in-backward flushes
+    x, y = _virtual_batch(ws)
+    xs, ys = x[rank * 4:(rank + 1) * 4], y[rank * 4:(rank + 1) * 4]
+    for _ in range(3):
+        opt.zero_grad()
+        torch.nn.functional.mse_loss(model(xs), ys).backward()
+        sync.finish()
+        opt.step()
+    ref = _make_model()
+    ref_opt = torch.optim.SGD(ref.parameters(), lr=0.05, momentum=0.9)
+    for _ in range(3):
+        ref_opt.zero_grad()
+        torch.nn.functional.mse_loss(ref(x), y).backward()
+        ref_opt.step()
+    for p, q in zip(model.parameters(), ref.parameters()):
+        assert torch.allclose(p, q, atol=1e-5), (p - q).abs().max()
+    sync.remove()
+
+
+def _check_overlapped_sync_unused_param(rank: int, ws: int):
+    """A param that never gets a grad leaves its chunk incomplete; finish()
+    must still reduce it (straggler flush) so ranks stay aligned."""
+    from flashy_amd.optim import FusedSGD
+    model = _make_model()
+    model.extra = nn.Parameter(torch.ones(4))  # never used in forward
+    opt = FusedSGD(model.parameters(), lr=0.05)
+    sync = distrib.OverlappedFlatSync(opt)  # default chunking: 1 big chunk
+    x, y = _virtual_batch(ws)
+    xs, ys = x[rank * 4:(rank + 1) * 4], y[rank * 4:(rank + 1) * 4]
+    opt.zero_grad()
+    torch.nn.functional.mse_loss(model(xs), ys).backward()
+    sync.finish()
+    # grads averaged, unused param's grad stays exactly zero
+    assert torch.equal(model.extra.grad, torch.zeros(4))
+    grads = [p.grad for n, p in model.named_parameters() if n != "extra"]
+    for got, ref in zip(grads, _reference_grads(ws)):
+        assert torch.allclose(got, ref, atol=1e-6)
+    sync.remove()
+
+
+def _check_eager_sync_error_drain(rank: int, ws: int):
+    """An exception mid-backward inside the eager context must drain the
+    in-flight async all-reduces: the next collective on the communicator
+    still completes correctly on every rank (no interleave / deadlock)."""
+    model = _make_model()
+    x, y = _virtual_batch(ws)
+    xs, ys = x[rank * 4:(rank + 1) * 4], y[rank * 4:(rank + 1) * 4]
+    hidden = torch.relu(model[0](xs))
+
+    def _boom(_grad):
+        raise RuntimeError("boom")
+
+    hidden.register_hook(_boom)  # fires AFTER layer-2 grads flushed
+    raised = False
+    try:
+        # bucket_bytes=8: every early grad launches its own async all-reduce
+        with distrib.eager_sync_gradients(model.parameters(), bucket_bytes=8):
+            torch.nn.functional.mse_loss(model[2](hidden), ys).backward()
+    except RuntimeError:
+        raised = True
+    assert raised
+    t = torch.full((4,), float(rank))
+    distrib.average_tensors([t])
+    assert torch.allclose(t, torch.full((4,), (ws - 1) / 2)), t
+
+
 def _check_broadcast_object(rank: int, ws: int):
     import collections
     if rank == 0:
@@ -220,6 +294,9 @@ ALL_CHECKS = [
     "_check_eager_sync_equivalence",
     "_check_eager_small_buckets",
     "_check_flat_optimizer_equivalence",
+    "_check_overlapped_flat_sync",
+    "_check_overlapped_sync_unused_param",
+    "_check_eager_sync_error_drain",
     "_check_broadcast_object",
     "_check_average_metrics",
     "_check_broadcast_model_and_barrier",
