@@ -14,7 +14,7 @@ import torch
 from torch.nn import functional as F
 
 from flashy_amd import BaseSolver, Formatter, distrib
-from flashy_amd.functional import cross_entropy
+from flashy_amd.functional import accuracy, cross_entropy
 from flashy_amd.graph import CapturedStep
 from flashy_amd.models.resnet_native import NativeResNet
 from flashy_amd.utils import averager
@@ -82,7 +82,7 @@ class Solver(BaseSolver):
         self._static_img.copy_(img, non_blocking=True)
         self._static_label.copy_(label, non_blocking=True)
         loss, est = self._graph()
-        acc = (est.argmax(1) == self._static_label).float().mean()
+        acc = accuracy(est, self._static_label)
         return loss, acc
 
     def _step(self, img, label, train: bool):
@@ -97,7 +97,7 @@ class Solver(BaseSolver):
             loss = cross_entropy(est, label)  # fused fwd+grad kernel
         else:
             loss = F.cross_entropy(est, label)
-        acc = (est.argmax(1) == label).float().mean()
+        acc = accuracy(est, label)
         if train:
             self.optim.zero_grad()
             loss.backward()

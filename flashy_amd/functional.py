@@ -64,3 +64,41 @@ def bce_with_logits_const(x: torch.Tensor, target_value: float) -> torch.Tensor:
         target = torch.full_like(x, target_value)
         return torch.nn.functional.binary_cross_entropy_with_logits(x, target)
     return _BCEWithLogits.apply(x, target_value)
+
+
+class _MSE(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, t: torch.Tensor) -> torch.Tensor:
+        n = x.numel()
+        dx = torch.empty_like(x)
+        loss = torch.zeros((), dtype=torch.float32, device=x.device)
+        ops.mse_fwd_bwd(x.contiguous(), t.contiguous(), dx, loss,
+                        loss_scale=1.0 / n, grad_scale=1.0 / n)
+        ctx.save_for_backward(dx)
+        return loss
+
+    @staticmethod
+    def backward(ctx, grad_out: torch.Tensor):
+        (dx,) = ctx.saved_tensors
+        return dx * grad_out, None
+
+
+def mse_loss(x: torch.Tensor, target: torch.Tensor) -> torch.Tensor:
+    """Mean squared error (fused fwd+grad kernel on GPU; torch on CPU).
+
+    Replaces F.mse_loss of the teacher-student workload
+    (/root/reference/tests/dummy/train.py:93)."""
+    if x.device.type != "cuda":
+        return torch.nn.functional.mse_loss(x, target)
+    return _MSE.apply(x, target.detach())
+
+
+def accuracy(logits: torch.Tensor, target: torch.Tensor) -> torch.Tensor:
+    """mean(argmax(logits, 1) == target) as a device scalar — one kernel on
+    GPU (vs the argmax+eq+mean 3-launch chain), torch ops on CPU."""
+    if logits.device.type != "cuda":
+        return (logits.argmax(1) == target).float().mean()
+    B = logits.shape[0]
+    out = torch.zeros((), dtype=torch.float32, device=logits.device)
+    ops.accuracy_count(logits.detach().contiguous(), target, out)
+    return out / B

@@ -95,7 +95,7 @@ class NativeResNet(nn.Module):
         self.layer2 = self._make_layer(block, 128, layers[1], stride=2)
         self.layer3 = self._make_layer(block, 256, layers[2], stride=2)
         self.layer4 = self._make_layer(block, 512, layers[3], stride=2)
-        self.fc = nn.Linear(512 * block.expansion, num_classes)
+        self.fc = fnn.Linear(512 * block.expansion, num_classes)
         self._wt_cache = None
 
     def _make_layer(self, block: type, planes: int, n: int, stride: int = 1):

@@ -306,6 +306,62 @@ class ConvTranspose2d(nn.Module):
                                       self.input_grad and x.requires_grad)
 
 
+class _LinearFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, w: torch.Tensor,
+                b: tp.Optional[torch.Tensor]):
+        B, I = x.shape
+        O = w.shape[0]
+        y = x.new_empty((B, O))
+        ops.linear_fwd(x.contiguous(), w, b, y)
+        ctx.save_for_backward(x)
+        ctx.refs = (w, b)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy: torch.Tensor):
+        (x,) = ctx.saved_tensors
+        w, b = ctx.refs
+        dy = dy.contiguous()
+        dw = db = None
+        need_w = ctx.needs_input_grad[1]
+        need_b = b is not None and ctx.needs_input_grad[2]
+        if need_w or need_b:
+            # one kernel produces both; frozen side goes to a discarded temp
+            dw, w_direct = _grad_target(w) if need_w \
+                else (torch.empty_like(w), True)
+            db, b_direct = (_grad_target(b) if need_b else (None, True))
+            ops.linear_dw(x, dy, dw, db)
+            dw = dw if need_w and not w_direct else None
+            db = db if need_b and not b_direct else None
+        dx = None
+        if ctx.needs_input_grad[0]:
+            dx = x.new_empty(x.shape)
+            ops.linear_dx(dy, w, dx)
+        return dx, dw, db
+
+
+class Linear(nn.Module):
+    """fp32 fully-connected layer on the native fc kernels (the ResNet head
+    and the tiny example MLPs — launch-bound shapes, one kernel per product).
+    Weight [O, I], bias [O]; CPU falls back to torch.nn.functional.linear."""
+
+    def __init__(self, in_features: int, out_features: int, bias: bool = True):
+        super().__init__()
+        self.weight = nn.Parameter(torch.empty(out_features, in_features))
+        bound = 1.0 / math.sqrt(in_features)
+        nn.init.uniform_(self.weight, -bound, bound)
+        self.bias: tp.Optional[nn.Parameter] = None
+        if bias:
+            self.bias = nn.Parameter(torch.empty(out_features))
+            nn.init.uniform_(self.bias, -bound, bound)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if x.device.type != "cuda":
+            return torch.nn.functional.linear(x, self.weight, self.bias)
+        return _LinearFn.apply(x, self.weight, self.bias)
+
+
 class _MaxPoolFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x: torch.Tensor, kernel: int, stride: int, pad: int):

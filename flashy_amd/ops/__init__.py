@@ -347,3 +347,55 @@ def bce_logits_fwd_bwd(x: torch.Tensor, dx: torch.Tensor,
     assert x.is_contiguous()
     ext.bce_logits(x.data_ptr(), dx.data_ptr(), loss_sum.data_ptr(), x.numel(),
                    target, loss_scale, grad_scale, _is_bf16(x), _stream())
+
+
+def mse_fwd_bwd(x: torch.Tensor, t: torch.Tensor, dx: torch.Tensor,
+                loss_sum: torch.Tensor, loss_scale: float,
+                grad_scale: float) -> None:
+    ext = require()
+    assert x.is_contiguous() and t.is_contiguous() and x.dtype == t.dtype
+    ext.mse(x.data_ptr(), t.data_ptr(), dx.data_ptr(), loss_sum.data_ptr(),
+            x.numel(), loss_scale, grad_scale, _is_bf16(x), _stream())
+
+
+def accuracy_count(logits: torch.Tensor, target: torch.Tensor,
+                   out: torch.Tensor) -> None:
+    """out[0] += number of rows whose argmax == target (caller zeroes out)."""
+    ext = require()
+    B, C = logits.shape
+    assert logits.is_contiguous() and target.dtype == torch.int64
+    ext.accuracy(logits.data_ptr(), target.data_ptr(), out.data_ptr(), B, C,
+                 _is_bf16(logits), _stream())
+
+
+# ---------------------------------------------------------------------------
+# linear (fc) kernels — fp32, small shapes (see csrc/linear.hip)
+# ---------------------------------------------------------------------------
+
+def linear_fwd(x: torch.Tensor, w: torch.Tensor,
+               b: tp.Optional[torch.Tensor], y: torch.Tensor) -> None:
+    ext = require()
+    B, I = x.shape
+    O = w.shape[0]
+    assert x.dtype == torch.float32 and w.shape == (O, I)
+    ext.linear_fwd(x.data_ptr(), w.data_ptr(),
+                   b.data_ptr() if b is not None else 0,
+                   y.data_ptr(), B, I, O, _stream())
+
+
+def linear_dx(dy: torch.Tensor, w: torch.Tensor, dx: torch.Tensor) -> None:
+    ext = require()
+    B, O = dy.shape
+    I = w.shape[1]
+    ext.linear_dx(dy.data_ptr(), w.data_ptr(), dx.data_ptr(), B, I, O,
+                  _stream())
+
+
+def linear_dw(x: torch.Tensor, dy: torch.Tensor, dw: torch.Tensor,
+              db: tp.Optional[torch.Tensor]) -> None:
+    """Accumulates (+=) into dw / db (autograd accumulate semantics)."""
+    ext = require()
+    B, I = x.shape
+    O = dy.shape[1]
+    ext.linear_dw(x.data_ptr(), dy.data_ptr(), dw.data_ptr(),
+                  db.data_ptr() if db is not None else 0, B, I, O, _stream())

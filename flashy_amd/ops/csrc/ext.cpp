@@ -80,6 +80,17 @@ void launch_cross_entropy(const void* logits, const void* target, void* dlogits,
                           void* loss_sum, int64_t B, int64_t C,
                           float loss_scale, float grad_scale, int is_bf16,
                           hipStream_t stream);
+void launch_mse(const void* x, const void* t, void* dx, void* loss_sum,
+                int64_t n, float loss_scale, float grad_scale, int is_bf16,
+                hipStream_t stream);
+void launch_accuracy(const void* logits, const void* target, void* out,
+                     int64_t B, int64_t C, int is_bf16, hipStream_t stream);
+void launch_linear_fwd(const void* x, const void* w, const void* b, void* y,
+                       int64_t B, int64_t I, int64_t O, hipStream_t stream);
+void launch_linear_dx(const void* dy, const void* w, void* dx, int64_t B,
+                      int64_t I, int64_t O, hipStream_t stream);
+void launch_linear_dw(const void* x, const void* dy, void* dw, void* db,
+                      int64_t B, int64_t I, int64_t O, hipStream_t stream);
 void launch_bce_logits(const void* x, void* dx, void* loss_sum, int64_t n,
                        float target, float loss_scale, float grad_scale,
                        int is_bf16, hipStream_t stream);
@@ -323,6 +334,50 @@ PYBIND11_MODULE(_hip_ops, m) {
               launch_bce_logits((const void*)x, (void*)dx, (void*)loss_sum, n,
                                 target, loss_scale, grad_scale,
                                 is_bf16 ? 1 : 0, as_stream(stream));
+              check_last();
+          });
+
+    m.def("mse",
+          [](uintptr_t x, uintptr_t t, uintptr_t dx, uintptr_t loss_sum,
+             int64_t n, float loss_scale, float grad_scale, bool is_bf16,
+             uintptr_t stream) {
+              launch_mse((const void*)x, (const void*)t, (void*)dx,
+                         (void*)loss_sum, n, loss_scale, grad_scale,
+                         is_bf16 ? 1 : 0, as_stream(stream));
+              check_last();
+          });
+
+    m.def("accuracy",
+          [](uintptr_t logits, uintptr_t target, uintptr_t out, int64_t B,
+             int64_t C, bool is_bf16, uintptr_t stream) {
+              launch_accuracy((const void*)logits, (const void*)target,
+                              (void*)out, B, C, is_bf16 ? 1 : 0,
+                              as_stream(stream));
+              check_last();
+          });
+
+    m.def("linear_fwd",
+          [](uintptr_t x, uintptr_t w, uintptr_t b, uintptr_t y, int64_t B,
+             int64_t I, int64_t O, uintptr_t stream) {
+              launch_linear_fwd((const void*)x, (const void*)w,
+                                (const void*)b, (void*)y, B, I, O,
+                                as_stream(stream));
+              check_last();
+          });
+
+    m.def("linear_dx",
+          [](uintptr_t dy, uintptr_t w, uintptr_t dx, int64_t B, int64_t I,
+             int64_t O, uintptr_t stream) {
+              launch_linear_dx((const void*)dy, (const void*)w, (void*)dx, B,
+                               I, O, as_stream(stream));
+              check_last();
+          });
+
+    m.def("linear_dw",
+          [](uintptr_t x, uintptr_t dy, uintptr_t dw, uintptr_t db, int64_t B,
+             int64_t I, int64_t O, uintptr_t stream) {
+              launch_linear_dw((const void*)x, (const void*)dy, (void*)dw,
+                               (void*)db, B, I, O, as_stream(stream));
               check_last();
           });
 }

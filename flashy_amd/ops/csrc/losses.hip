@@ -135,3 +135,85 @@ extern "C" void launch_bce_logits(const void* x, void* dx, void* loss_sum,
             (const float*)x, (float*)dx, (float*)loss_sum, n, target,
             loss_scale, grad_scale);
 }
+
+// ---------------------------------------------------------------------------
+// MSE loss, fused fwd+grad:  loss += (x-t)^2 * loss_scale,
+// dx = 2(x-t) * grad_scale.  Elementwise + one atomic per wave.
+// Replaces F.mse_loss in the teacher-student workload
+// (/root/reference/tests/dummy/train.py:93).
+// ---------------------------------------------------------------------------
+
+template <typename T>
+__global__ void __launch_bounds__(256)
+k_mse(const T* __restrict__ x, const T* __restrict__ t, T* __restrict__ dx,
+      float* __restrict__ loss_sum, int64_t n, float loss_scale,
+      float grad_scale) {
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    float local = 0.f;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        const float d = load_f32(x, i) - load_f32(t, i);
+        local = fmaf(d, d, local);
+        store_f32(dx, i, 2.f * d * grad_scale);
+    }
+    local = wave_sum(local);
+    if ((threadIdx.x % WAVE_SIZE) == 0 && local != 0.f)
+        atomicAdd(loss_sum, local * loss_scale);
+}
+
+extern "C" void launch_mse(const void* x, const void* t, void* dx,
+                           void* loss_sum, int64_t n, float loss_scale,
+                           float grad_scale, int is_bf16, hipStream_t stream) {
+    int grid = ew_grid(n, 256, 4);
+    if (is_bf16)
+        k_mse<uint16_t><<<grid, 256, 0, stream>>>(
+            (const uint16_t*)x, (const uint16_t*)t, (uint16_t*)dx,
+            (float*)loss_sum, n, loss_scale, grad_scale);
+    else
+        k_mse<float><<<grid, 256, 0, stream>>>(
+            (const float*)x, (const float*)t, (float*)dx, (float*)loss_sum, n,
+            loss_scale, grad_scale);
+}
+
+// ---------------------------------------------------------------------------
+// Accuracy: mean(argmax(logits, 1) == target).  One wave per row, atomic
+// count of correct rows into out[0] (caller pre-zeroes and divides by B).
+// Replaces the torch argmax+eq+mean chain (3 launches -> 1,
+// examples/cifar/solver.py accuracy metric).
+// ---------------------------------------------------------------------------
+
+template <typename T>
+__global__ void __launch_bounds__(256)
+k_accuracy(const T* __restrict__ logits, const int64_t* __restrict__ target,
+           float* __restrict__ out, int64_t B, int64_t C) {
+    const int wave = threadIdx.x / WAVE_SIZE;
+    const int lane = threadIdx.x % WAVE_SIZE;
+    const int64_t row = (int64_t)blockIdx.x * 4 + wave;
+    if (row >= B) return;
+    const T* x = logits + row * C;
+    float best = -INFINITY;
+    int64_t best_c = 0;
+    for (int64_t c = lane; c < C; c += WAVE_SIZE) {
+        const float v = load_f32(x, c);
+        if (v > best || (v == best && c < best_c)) { best = v; best_c = c; }
+    }
+    // wave-reduce (value, index), ties -> smallest index (torch argmax)
+    for (int off = WAVE_SIZE / 2; off; off >>= 1) {
+        const float ov = __shfl_down(best, off, WAVE_SIZE);
+        const int64_t oc = __shfl_down(best_c, off, WAVE_SIZE);
+        if (ov > best || (ov == best && oc < best_c)) { best = ov; best_c = oc; }
+    }
+    if (lane == 0 && best_c == target[row]) atomicAdd(out, 1.f);
+}
+
+extern "C" void launch_accuracy(const void* logits, const void* target,
+                                void* out, int64_t B, int64_t C, int is_bf16,
+                                hipStream_t stream) {
+    int grid = (int)((B + 3) / 4);
+    if (is_bf16)
+        k_accuracy<uint16_t><<<grid, 256, 0, stream>>>(
+            (const uint16_t*)logits, (const int64_t*)target, (float*)out, B, C);
+    else
+        k_accuracy<float><<<grid, 256, 0, stream>>>(
+            (const float*)logits, (const int64_t*)target, (float*)out, B, C);
+}

@@ -16,6 +16,26 @@ SHAPES = [
     ("l3conv", 8, 8, 256, 256, 3, 1),
     ("l4conv", 4, 4, 512, 512, 3, 1),
 ]
+
+# ResNet-50 bottleneck shapes at 224px input (BASELINE config 5), batch 64
+SHAPES_R50 = [
+    ("r1_1x1a", 56, 56, 64, 64, 1, 1),
+    ("r1_3x3", 56, 56, 64, 64, 3, 1),
+    ("r1_1x1b", 56, 56, 64, 256, 1, 1),
+    ("r1_1x1c", 56, 56, 256, 64, 1, 1),
+    ("r2_down", 56, 56, 256, 512, 1, 2),
+    ("r2_1x1a", 28, 28, 512, 128, 1, 1),
+    ("r2_3x3", 28, 28, 128, 128, 3, 1),
+    ("r2_1x1b", 28, 28, 128, 512, 1, 1),
+    ("r3_1x1a", 14, 14, 1024, 256, 1, 1),
+    ("r3_3x3", 14, 14, 256, 256, 3, 1),
+    ("r3_1x1b", 14, 14, 256, 1024, 1, 1),
+    ("r4_1x1a", 7, 7, 2048, 512, 1, 1),
+    ("r4_3x3", 7, 7, 512, 512, 3, 1),
+    ("r4_1x1b", 7, 7, 512, 2048, 1, 1),
+]
+if "--r50" in sys.argv:
+    SHAPES = SHAPES_R50
 N = 64
 REPS = 30
 

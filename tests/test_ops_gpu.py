@@ -103,6 +103,79 @@ def test_bce_logits_gpu(dtype, target):
 
 
 @requires_gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_mse_gpu(dtype):
+    from flashy_amd.functional import mse_loss
+    torch.manual_seed(4)
+    x = (torch.randn(64, 33, device="cuda")).to(dtype).requires_grad_(True)
+    t = torch.randn(64, 33, device="cuda").to(dtype)
+    loss = mse_loss(x, t)
+    loss.backward()
+    ref_in = x.detach().float().clone().requires_grad_(True)
+    ref = torch.nn.functional.mse_loss(ref_in, t.float())
+    ref.backward()
+    tol = 1e-5 if dtype == torch.float32 else 2e-2
+    assert torch.allclose(loss.float(), ref, atol=tol, rtol=tol)
+    assert torch.allclose(x.grad.float(), ref_in.grad, atol=tol, rtol=tol)
+
+
+@requires_gpu
+@pytest.mark.parametrize("dtype,B,C", [
+    (torch.float32, 64, 10), (torch.bfloat16, 64, 10),
+    (torch.float32, 37, 1000)])
+def test_accuracy_gpu(dtype, B, C):
+    from flashy_amd.functional import accuracy
+    torch.manual_seed(5)
+    logits = (torch.randn(B, C, device="cuda") * 2).to(dtype)
+    target = torch.randint(C, (B,), device="cuda")
+    got = accuracy(logits, target)
+    ref = (logits.float().argmax(1) == target).float().mean()
+    assert torch.allclose(got, ref), (got.item(), ref.item())
+
+
+@requires_gpu
+@pytest.mark.parametrize("B,I,O,bias", [
+    (64, 512, 10, True), (7, 32, 1, True), (16, 33, 5, False)])
+def test_linear_gpu(B, I, O, bias):
+    from flashy_amd.nn import Linear
+    torch.manual_seed(6)
+    lin = Linear(I, O, bias=bias).cuda()
+    ref = torch.nn.Linear(I, O, bias=bias).cuda()
+    with torch.no_grad():
+        ref.weight.copy_(lin.weight)
+        if bias:
+            ref.bias.copy_(lin.bias)
+    x = torch.randn(B, I, device="cuda", requires_grad=True)
+    xr = x.detach().clone().requires_grad_(True)
+    y = lin(x)
+    yr = ref(xr)
+    assert torch.allclose(y, yr, atol=1e-5, rtol=1e-5)
+    g = torch.randn_like(y)
+    y.backward(g)
+    yr.backward(g)
+    assert torch.allclose(x.grad, xr.grad, atol=1e-5, rtol=1e-5)
+    assert torch.allclose(lin.weight.grad, ref.weight.grad, atol=1e-4, rtol=1e-4)
+    if bias:
+        assert torch.allclose(lin.bias.grad, ref.bias.grad, atol=1e-4, rtol=1e-4)
+    # grad accumulation across two backwards matches torch
+    lin(x).backward(g)
+    ref(xr).backward(g)
+    assert torch.allclose(lin.weight.grad, ref.weight.grad, atol=1e-4, rtol=1e-4)
+
+
+@requires_gpu
+def test_linear_frozen_weight_gpu():
+    from flashy_amd.nn import Linear
+    lin = Linear(16, 4).cuda()
+    lin.weight.requires_grad_(False)
+    x = torch.randn(8, 16, device="cuda", requires_grad=True)
+    lin(x).sum().backward()
+    assert lin.weight.grad is None
+    assert lin.bias.grad is not None
+    assert x.grad is not None
+
+
+@requires_gpu
 def test_graph_captured_step():
     from flashy_amd.graph import CapturedStep
     from flashy_amd.models import resnet18
