@@ -301,9 +301,18 @@ static int conv_subs() {   // A/B switch: FLASHY_CONV_SUBS=1 -> 32-deep stages
     return v;
 }
 
+// 8-wave 256-row kernel (conv_fwd8.hip) — used for the large-M layers.
+extern "C" int conv_fwd8_plan(ConvDims d, int* bn_out);
+extern "C" void launch_conv_fwd8(const void* x, const void* w, void* y,
+                                 ConvDims d, int relu, void* bn_ws, int bn,
+                                 int mtiles, hipStream_t stream);
+
 // grid.x the fwd launcher will use for these dims (= the msplit of the
 // fused BN-stats partials); Python sizes the partials buffer with this.
 extern "C" int conv_fwd_msplit(ConvDims d) {
+    int bn8;
+    const int mt8 = conv_fwd8_plan(d, &bn8);
+    if (mt8) return mt8;
     const int64_t M = (int64_t)d.N * d.Ho * d.Wo;
     int bm, bn;
     pick_tile(M, d.K, &bm, &bn);
@@ -313,6 +322,12 @@ extern "C" int conv_fwd_msplit(ConvDims d) {
 extern "C" void launch_conv_fwd(const void* x, const void* w, void* y,
                                 ConvDims d, int relu, void* bn_ws,
                                 hipStream_t stream) {
+    int bn8;
+    const int mt8 = conv_fwd8_plan(d, &bn8);
+    if (mt8) {
+        launch_conv_fwd8(x, w, y, d, relu, bn_ws, bn8, mt8, stream);
+        return;
+    }
     const int64_t M = (int64_t)d.N * d.Ho * d.Wo;
     int bm, bn;
     pick_tile(M, d.K, &bm, &bn);

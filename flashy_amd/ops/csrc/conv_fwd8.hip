@@ -1,0 +1,300 @@
+// Copyright (c) Flashy-AMD authors.
+// 8-wave 256-row implicit-GEMM conv FORWARD for gfx950 — the deep-pipeline
+// schedule (CDNA4 guide §5 "256² 8-phase template" adapted to conv):
+//
+//   * 512 threads (8 waves, 2M x 4N), tile BM=256 x BN in {64,128}, BK=64;
+//   * BOTH operands staged to LDS by async global->LDS DMA
+//     (`raw_ptr_buffer_load_lds` 16 B/lane): no staging registers, no
+//     ds_write pass, and the A (im2col) padding is handled by the buffer
+//     descriptor's bounds check — out-of-range taps load hardware zeros
+//     (voffset sentinel far past num_bytes), so the inner loop has ZERO
+//     branches;
+//   * 3 LDS buffers, ONE raw `s_barrier` per 64-deep stage, counted
+//     `s_waitcnt vmcnt(G)` so the next stage's DMA stays in flight across
+//     the barrier (never vmcnt(0) in the main loop);
+//   * LDS images [rows][64] bf16 (128 B rows) XOR-swizzled by
+//     `byte ^= (row&7)<<4` — applied on the glds SOURCE chunk index and the
+//     `ds_read_b128` byte address (guide §5.4 rule 21), so the 16-lane b128
+//     groups stay <=2-way bank-conflicted;
+//   * `s_setprio(1)` around each MFMA cluster (T5: pays on phase-split
+//     8-wave schedules).
+//
+// Used for the large-M ResNet-50/224-class layers (C%64==0, K%64==0,
+// tensor < 2 GB); small-M / long-K shapes stay on the 4-wave split-K
+// kernel in conv_fwd.hip.  GEMM view and layouts as in conv_fwd.hip.
+
+#include "conv_common.h"
+
+// out-of-range sentinel: voffset far past any real tensor (tensors routed
+// here are < 2^31 - 2^28 bytes); the buffer bounds check returns zeros.
+#define OOB_SENTINEL 0xF0000000u
+
+template <int BN, bool RELU>
+__global__ void __launch_bounds__(512, 2)
+k_conv_fwd8(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
+            uint16_t* __restrict__ y, float* __restrict__ bn_ws,
+            ConvDims d, unsigned x_nbytes) {
+    constexpr int BM = 256;
+    constexpr int BK = 64;
+    constexpr int NF = BN / 64;          // n fragments per wave (4 N-waves)
+    constexpr int MF = 8;                // m fragments per wave (2 M-waves)
+    constexpr int A_ELEMS = BM * BK;     // 16384 bf16 = 32 KiB
+    constexpr int B_ELEMS = BN * BK;     // 4096/8192 bf16 = 8/16 KiB
+    constexpr int BUF_ELEMS = A_ELEMS + B_ELEMS;
+    constexpr int G = 4 + BN / 64;       // glds per wave per stage (A + B)
+
+    const int rsc = d.R * d.S * d.C;
+    const int64_t M = (int64_t)d.N * d.Ho * d.Wo;
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid_u = __builtin_amdgcn_readfirstlane(tid >> 6);
+    const int wave_m = wid_u >> 2;       // 0..1
+    const int wave_n = wid_u & 3;        // 0..3
+
+    unsigned bx = blockIdx.x;
+    if ((gridDim.x & 7) == 0)            // XCD-aware m-tile order (T1)
+        bx = (bx & 7) * (gridDim.x >> 3) + (bx >> 3);
+    const int64_t m0 = (int64_t)bx * BM;
+    const int col0 = blockIdx.y * BN;
+
+    __shared__ uint16_t lds[3 * BUF_ELEMS];
+
+    const auto xrsrc = __builtin_amdgcn_make_buffer_rsrc(
+        (void*)x, 0 /*stride*/, x_nbytes, 0x00020000 /*flags*/);
+
+    // --- A staging state: 4 dest chunks per thread, geometry fixed, tap
+    // (r,s,c) walks +BK per stage with carries (no divides steady-state).
+    int a_r[4], a_s[4], a_c[4], a_hi0[4], a_wi0[4];
+    int64_t a_n[4];
+#pragma unroll
+    for (int g = 0; g < 4; ++g) {
+        const int chunk = g * 512 + tid;          // dest chunk 0..2047
+        const int row = chunk >> 3;
+        const int kc_s = (chunk & 7) ^ (row & 7); // source k-chunk (swizzle)
+        const int64_t m = m0 + row;
+        if (m < M) {
+            const int hw = d.Ho * d.Wo;
+            a_n[g] = m / hw;
+            const int rem = (int)(m % hw);
+            a_hi0[g] = (rem / d.Wo) * d.stride - d.pad;
+            a_wi0[g] = (rem % d.Wo) * d.stride - d.pad;
+        } else {
+            a_n[g] = -1;
+        }
+        const int kk = kc_s * 8;                  // tap offset at stage 0
+        a_r[g] = kk / (d.S * d.C);
+        const int sc = kk - a_r[g] * d.S * d.C;
+        a_s[g] = sc / d.C;
+        a_c[g] = sc - a_s[g] * d.C;
+    }
+    // B staging: source byte offset advances by BK*2 per stage.
+    unsigned b_src[2];
+#pragma unroll
+    for (int g = 0; g < NF; ++g) {
+        const int chunk = g * 512 + tid;          // dest chunk 0..BN*8-1
+        const int col = chunk >> 3;
+        const int kc_s = (chunk & 7) ^ (col & 7);
+        b_src[g] = (unsigned)(((int64_t)(col0 + col) * rsc + kc_s * 8) * 2);
+    }
+
+    auto issue_stage = [&](int stage, int buf) {
+        uint16_t* base = lds + buf * BUF_ELEMS;
+#pragma unroll
+        for (int g = 0; g < 4; ++g) {
+            unsigned voff = OOB_SENTINEL;
+            if (a_n[g] >= 0 && a_r[g] < d.R) {
+                const int hi = a_hi0[g] + a_r[g];
+                const int wi = a_wi0[g] + a_s[g];
+                if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W)
+                    voff = (unsigned)((((a_n[g] * d.H + hi) * d.W + wi) *
+                                       (int64_t)d.C + a_c[g]) * 2);
+            }
+            // advance tap by one stage (+BK) with carries
+            int c = a_c[g] + BK;
+            while (c >= d.C) {
+                c -= d.C;
+                if (++a_s[g] == d.S) { a_s[g] = 0; ++a_r[g]; }
+            }
+            a_c[g] = c;
+            __builtin_amdgcn_raw_ptr_buffer_load_lds(
+                xrsrc,
+                (__attribute__((address_space(3))) void*)
+                    (base + (g * 512 + wid_u * 64) * 8),
+                16, voff, 0, 0, 0);
+        }
+        uint16_t* bbase = base + A_ELEMS;
+        const unsigned kb = (unsigned)(stage * BK * 2);
+#pragma unroll
+        for (int g = 0; g < NF; ++g) {
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) unsigned int*)
+                    ((const char*)w + b_src[g] + kb),
+                (__attribute__((address_space(3))) unsigned int*)
+                    (bbase + (g * 512 + wid_u * 64) * 8),
+                16, 0, 0);
+        }
+    };
+
+    // fragment LDS byte offsets (swizzled), fixed per lane
+    const int a_row_l = wave_m * 128 + (lane & 15);   // + mf*16
+    const int frag_kb = (lane >> 4) * 16;             // byte within 64B half
+    const int b_col_l = wave_n * (BN / 4) + (lane & 15);  // + nf*16
+
+    floatx4 acc[MF][NF] = {};
+    const int n_stages = rsc / BK;
+
+    auto compute_stage = [&](int buf) {
+        const uint16_t* base = lds + buf * BUF_ELEMS;
+        const uint16_t* bbase = base + A_ELEMS;
+#pragma unroll
+        for (int sub = 0; sub < 2; ++sub) {
+            short8 a[MF], b[NF];
+#pragma unroll
+            for (int mf = 0; mf < MF; ++mf) {
+                const int row = a_row_l + mf * 16;
+                const int byte = (row * 128 + sub * 64 + frag_kb) ^
+                                 ((row & 7) << 4);
+                a[mf] = *reinterpret_cast<const short8*>(
+                    (const char*)base + byte);
+            }
+#pragma unroll
+            for (int nf = 0; nf < NF; ++nf) {
+                const int col = b_col_l + nf * 16;
+                const int byte = (col * 128 + sub * 64 + frag_kb) ^
+                                 ((col & 7) << 4);
+                b[nf] = *reinterpret_cast<const short8*>(
+                    (const char*)bbase + byte);
+            }
+            __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+            for (int mf = 0; mf < MF; ++mf)
+#pragma unroll
+                for (int nf = 0; nf < NF; ++nf)
+                    acc[mf][nf] = MFMA_BF16(a[mf], b[nf], acc[mf][nf]);
+            __builtin_amdgcn_s_setprio(0);
+        }
+        // no tail barrier: wave skew is bounded by the next iteration's
+        // barrier, and the in-flight writes always target b[(i+2)%3] while
+        // laggards read b[i%3] — never the same buffer.
+    };
+
+    issue_stage(0, 0);
+    for (int i = 0; i + 1 < n_stages; ++i) {
+        issue_stage(i + 1, (i + 1) % 3);
+        // stage i landed; own G loads of stage i+1 stay in flight across
+        // the barrier (counted wait — never vmcnt(0) in the hot loop)
+        asm volatile("s_waitcnt vmcnt(%0)" ::"i"(G) : "memory");
+        __builtin_amdgcn_s_barrier();
+        compute_stage(i % 3);
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    compute_stage((n_stages - 1) % 3);
+
+    // --- epilogue: bf16 store (+relu), optional fused BN partials --------
+    const int64_t out_row0 = m0 + wave_m * 128 + (lane >> 4) * 4;
+    const int out_col0 = col0 + wave_n * (BN / 4) + (lane & 15);
+#pragma unroll
+    for (int mf = 0; mf < MF; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < NF; ++nf)
+#pragma unroll
+            for (int rr = 0; rr < 4; ++rr) {
+                const int64_t row = out_row0 + mf * 16 + rr;
+                if (row < M) {
+                    float v = acc[mf][nf][rr];
+                    if (RELU) v = fmaxf(v, 0.f);
+                    y[row * d.K + out_col0 + nf * 16] = f32_to_bf16(v);
+                }
+            }
+
+    if (bn_ws != nullptr) {
+        __syncthreads();                 // main loop fully done: reuse lds
+        float* sred = reinterpret_cast<float*>(lds);   // [2*WM][BN]
+#pragma unroll
+        for (int nf = 0; nf < NF; ++nf) {
+            float s = 0.f, s2 = 0.f;
+#pragma unroll
+            for (int mf = 0; mf < MF; ++mf)
+#pragma unroll
+                for (int rr = 0; rr < 4; ++rr) {
+                    const int64_t row = out_row0 + mf * 16 + rr;
+                    float v = acc[mf][nf][rr];
+                    if (RELU) v = fmaxf(v, 0.f);
+                    if (row >= M) v = 0.f;
+                    s += v;
+                    s2 = fmaf(v, v, s2);
+                }
+            s += __shfl_xor(s, 16, 64);
+            s += __shfl_xor(s, 32, 64);
+            s2 += __shfl_xor(s2, 16, 64);
+            s2 += __shfl_xor(s2, 32, 64);
+            if (lane < 16) {
+                const int colL = wave_n * (BN / 4) + nf * 16 + lane;
+                sred[wave_m * BN + colL] = s;
+                sred[(2 + wave_m) * BN + colL] = s2;
+            }
+        }
+        __syncthreads();
+        if (tid < BN) {
+            const float s = sred[tid] + sred[BN + tid];
+            const float s2 = sred[2 * BN + tid] + sred[3 * BN + tid];
+            const int c = col0 + tid;
+            bn_ws[(int64_t)c * gridDim.x + blockIdx.x] = s;
+            bn_ws[((int64_t)d.K + c) * gridDim.x + blockIdx.x] = s2;
+        }
+    }
+}
+
+// --- dispatch ---------------------------------------------------------------
+// Eligibility + grid for the 8-wave kernel; returns grid.x (m-tiles) or 0.
+// Decided purely from the dims so conv_fwd_msplit (BN-partials sizing) and
+// the launcher always agree.
+#include <cstdlib>
+extern "C" int conv_fwd8_plan(ConvDims d, int* bn_out) {
+    static int disabled = [] {
+        const char* e = getenv("FLASHY_NO_FWD8");
+        return e && e[0] == '1';
+    }();
+    if (disabled) return 0;
+    const int64_t M = (int64_t)d.N * d.Ho * d.Wo;
+    const int rsc = d.R * d.S * d.C;
+    if (d.C % 64 || d.K % 64 || rsc % 64) return 0;
+    const int64_t x_elems = (int64_t)d.N * d.H * d.W * d.C;
+    if (x_elems * 2 >= (int64_t)OOB_SENTINEL) return 0;   // 32-bit voffset
+    const int mtiles = (int)((M + 255) / 256);
+    if (d.K % 128 == 0 && (int64_t)mtiles * (d.K / 128) >= 160) {
+        *bn_out = 128;
+        return mtiles;
+    }
+    if ((int64_t)mtiles * (d.K / 64) >= 160) {
+        *bn_out = 64;
+        return mtiles;
+    }
+    return 0;
+}
+
+extern "C" void launch_conv_fwd8(const void* x, const void* w, void* y,
+                                 ConvDims d, int relu, void* bn_ws, int bn,
+                                 int mtiles, hipStream_t stream) {
+    dim3 grid((unsigned)mtiles, (unsigned)(d.K / bn));
+    const unsigned xb = (unsigned)((int64_t)d.N * d.H * d.W * d.C * 2);
+    auto xx = (const uint16_t*)x;
+    auto ww = (const uint16_t*)w;
+    auto yy = (uint16_t*)y;
+    if (bn == 128) {
+        if (relu)
+            k_conv_fwd8<128, true><<<grid, 512, 0, stream>>>(
+                xx, ww, yy, (float*)bn_ws, d, xb);
+        else
+            k_conv_fwd8<128, false><<<grid, 512, 0, stream>>>(
+                xx, ww, yy, (float*)bn_ws, d, xb);
+    } else {
+        if (relu)
+            k_conv_fwd8<64, true><<<grid, 512, 0, stream>>>(
+                xx, ww, yy, (float*)bn_ws, d, xb);
+        else
+            k_conv_fwd8<64, false><<<grid, 512, 0, stream>>>(
+                xx, ww, yy, (float*)bn_ws, d, xb);
+    }
+}
