@@ -178,7 +178,8 @@ def conv_fwd(x: torch.Tensor, w: torch.Tensor, y: torch.Tensor, d: ConvDims,
         assert d.K % 64 == 0 and rsc % 32 == 0, d
         M = d.N * d.Ho * d.Wo
         stages = (rsc + 63) // 64
-        zn = _splitk_plan(M, d.K // 64, stages)
+        zn = 0 if ext.conv8_eligible(*d, False) else \
+            _splitk_plan(M, d.K // 64, stages)
         if zn:
             spz = (stages + zn - 1) // zn
             zeff = (stages + spz - 1) // spz
@@ -221,7 +222,8 @@ def conv_dgrad(dout: torch.Tensor, w_rsck: torch.Tensor, dx: torch.Tensor,
     M = d.N * d.H * d.W
     rsk = d.R * d.S * d.K
     stages = (rsk + 63) // 64
-    zn = _splitk_plan(M, d.C // 64, stages)
+    zn = 0 if ext.conv8_eligible(*d, True) else \
+        _splitk_plan(M, d.C // 64, stages)
     if zn:
         spz = (stages + zn - 1) // zn
         zeff = (stages + spz - 1) // spz

@@ -176,6 +176,90 @@ k_bn_apply(const uint16_t* __restrict__ x, const uint16_t* __restrict__ res,
     }
 }
 
+// Channel-resident variant for C % 64 == 0 (every trunk BN): thread owns a
+// fixed 8-channel group, coefficients load ONCE into registers, the loop is
+// pure 16B streaming over m — no per-iteration int64 div/mod (the generic
+// kernel's `(i*8) % C` is an emulated 64-bit divide and caps it ~3 TB/s).
+template <bool RELU, bool RES>
+__global__ void __launch_bounds__(256)
+k_bn_apply_c64(const uint16_t* __restrict__ x, const uint16_t* __restrict__ res,
+               uint16_t* __restrict__ y, const float* __restrict__ work,
+               int64_t M, int C, float slope, int m_per_block) {
+    const int c = blockIdx.x * 64 + (threadIdx.x & 7) * 8;
+    const int mlane = threadIdx.x >> 3;
+    const int64_t m0 = (int64_t)blockIdx.y * m_per_block;
+    const int64_t m1 = min(m0 + (int64_t)m_per_block, M);
+    float sc[8], sh[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+        sc[j] = work[2 * (int64_t)C + c + j];
+        sh[j] = work[3 * (int64_t)C + c + j];
+    }
+    for (int64_t m = m0 + mlane; m < m1; m += 32) {
+        const int64_t off = m * C + c;
+        short8 xv = *reinterpret_cast<const short8*>(x + off);
+        short8 rv = {};
+        if (RES) rv = *reinterpret_cast<const short8*>(res + off);
+        short8 out;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            float v = fmaf(bf16_to_f32(((const uint16_t*)&xv)[j]), sc[j], sh[j]);
+            if (RES) v += bf16_to_f32(((const uint16_t*)&rv)[j]);
+            if (RELU) v = v > 0.f ? v : slope * v;
+            ((uint16_t*)&out)[j] = f32_to_bf16(v);
+        }
+        *reinterpret_cast<short8*>(y + off) = out;
+    }
+}
+
+__global__ void __launch_bounds__(256)
+k_bn_bwd_apply_c64(const uint16_t* __restrict__ dz,
+                   const uint16_t* __restrict__ x,
+                   const float* __restrict__ work,
+                   const float* __restrict__ bsums,
+                   uint16_t* __restrict__ dx, int64_t M, int C,
+                   int m_per_block) {
+    const int c = blockIdx.x * 64 + (threadIdx.x & 7) * 8;
+    const int mlane = threadIdx.x >> 3;
+    const int64_t m0 = (int64_t)blockIdx.y * m_per_block;
+    const int64_t m1 = min(m0 + (int64_t)m_per_block, M);
+    const float invM = 1.f / (float)M;
+    float mn[8], is[8], sc[8], b0[8], b1[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+        mn[j] = work[c + j];
+        is[j] = work[C + c + j];
+        sc[j] = work[2 * (int64_t)C + c + j];
+        b0[j] = bsums[c + j] * invM;
+        b1[j] = bsums[C + c + j] * invM;
+    }
+    for (int64_t m = m0 + mlane; m < m1; m += 32) {
+        const int64_t off = m * C + c;
+        short8 gz = *reinterpret_cast<const short8*>(dz + off);
+        short8 xv = *reinterpret_cast<const short8*>(x + off);
+        short8 out;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+            const float xhat = (bf16_to_f32(((const uint16_t*)&xv)[j]) - mn[j])
+                               * is[j];
+            const float g = bf16_to_f32(((const uint16_t*)&gz)[j]);
+            ((uint16_t*)&out)[j] =
+                f32_to_bf16(sc[j] * (g - b0[j] - xhat * b1[j]));
+        }
+        *reinterpret_cast<short8*>(dx + off) = out;
+    }
+}
+
+// m-blocks for the c64 apply kernels: ~1024 blocks saturate the chip
+static inline int apply_mpb(int64_t M, int C) {
+    const int cblocks = C / 64;
+    int target = 1024 / cblocks;
+    if (target < 1) target = 1;
+    int64_t mblocks = (M + 31) / 32;
+    if (mblocks > target) mblocks = target;
+    return (int)((M + mblocks - 1) / mblocks);
+}
+
 // ---------------------------------------------------------------------------
 // bwd 1: dz = dy * (y > 0) [if relu]; per-channel partial sums of dz and
 // dz*xhat.  Same geometry as stats.
@@ -318,6 +402,27 @@ extern "C" void launch_bn_finalize(const void* partials, int msplit,
 extern "C" void launch_bn_apply(const void* x, const void* res, void* y,
                                 const void* work, int64_t M, int C, int relu,
                                 float slope, hipStream_t stream) {
+    if (C % 64 == 0) {
+        const int mpb = apply_mpb(M, C);
+        dim3 g((unsigned)(C / 64), (unsigned)((M + mpb - 1) / mpb));
+        if (relu && res)
+            k_bn_apply_c64<true, true><<<g, 256, 0, stream>>>(
+                (const uint16_t*)x, (const uint16_t*)res, (uint16_t*)y,
+                (const float*)work, M, C, slope, mpb);
+        else if (relu)
+            k_bn_apply_c64<true, false><<<g, 256, 0, stream>>>(
+                (const uint16_t*)x, nullptr, (uint16_t*)y,
+                (const float*)work, M, C, slope, mpb);
+        else if (res)
+            k_bn_apply_c64<false, true><<<g, 256, 0, stream>>>(
+                (const uint16_t*)x, (const uint16_t*)res, (uint16_t*)y,
+                (const float*)work, M, C, slope, mpb);
+        else
+            k_bn_apply_c64<false, false><<<g, 256, 0, stream>>>(
+                (const uint16_t*)x, nullptr, (uint16_t*)y,
+                (const float*)work, M, C, slope, mpb);
+        return;
+    }
     const int grid = ew_grid(M * C / 8, 256, 4);
     if (relu && res)
         k_bn_apply<true, true><<<grid, 256, 0, stream>>>(

@@ -207,6 +207,11 @@ k_conv_dgrad(const uint16_t* __restrict__ dout, const uint16_t* __restrict__ w_r
 extern "C" int launch_conv_dgrad_s2(const void* dout, const void* w_rsck,
                                     void* dx, ConvDims d, hipStream_t stream);
 
+extern "C" int conv_dgrad8_plan(ConvDims d, int* bn_out);
+extern "C" void launch_conv_dgrad8(const void* dout, const void* w_rsck,
+                                   void* dx, ConvDims d, int bn, int mtiles,
+                                   hipStream_t stream);
+
 extern "C" void launch_conv_dgrad(const void* dout, const void* w_rsck,
                                   void* dx, ConvDims d, hipStream_t stream) {
     const int64_t M = (int64_t)d.N * d.H * d.W;
@@ -216,6 +221,14 @@ extern "C" void launch_conv_dgrad(const void* dout, const void* w_rsck,
     if (d.stride == 2 &&
         launch_conv_dgrad_s2(dout, w_rsck, dx, d, stream))
         return;  // parity-class form (no zero-filled MFMA work)
+    {
+        int bn8;
+        const int mt8 = conv_dgrad8_plan(d, &bn8);
+        if (mt8) {
+            launch_conv_dgrad8(dout, w_rsck, dx, d, bn8, mt8, stream);
+            return;
+        }
+    }
     if (d.C % 128 == 0 && (M + 127) / 128 * (d.C / 128) >= 208) {
         dim3 g((unsigned)((M + 127) / 128), (unsigned)(d.C / 128));
         if (d.stride == 1)

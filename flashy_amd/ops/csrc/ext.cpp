@@ -24,6 +24,8 @@ extern "C" {
 void launch_conv_fwd(const void* x, const void* w, void* y, ConvDims d,
                      int relu, void* bn_ws, hipStream_t stream);
 int conv_fwd_msplit(ConvDims d);
+int conv_fwd8_plan(ConvDims d, int* bn_out);
+int conv_dgrad8_plan(ConvDims d, int* bn_out);
 void launch_conv_stem_fwd(const void* x, const void* w, void* y, ConvDims d,
                           hipStream_t stream);
 void launch_conv_stem_wgrad(const void* x, const void* dout, void* dw,
@@ -123,6 +125,15 @@ PYBIND11_MODULE(_hip_ops, m) {
                               relu ? 1 : 0, (void*)bn_ws, as_stream(stream));
               check_last();
           });
+    m.def("conv8_eligible",
+          [](int N, int H, int W, int C, int K, int R, int S, int Ho, int Wo,
+             int stride, int pad, bool dgrad) {
+              int bn = 0;
+              ConvDims d = make_dims(N, H, W, C, K, R, S, Ho, Wo, stride, pad);
+              return (dgrad ? conv_dgrad8_plan(d, &bn)
+                            : conv_fwd8_plan(d, &bn)) > 0;
+          });
+
     m.def("conv_fwd_msplit",
           [](int N, int H, int W, int C, int K, int R, int S, int Ho, int Wo,
              int stride, int pad) {
