@@ -261,10 +261,14 @@ def conv_wgrad(x: torch.Tensor, dout: torch.Tensor, dw: torch.Tensor,
     if d.C % 8 == 0:
         assert d.K % 64 == 0 and rsc % 64 == 0, d
         if n_splits is None:
-            tiles = (d.K // 64) * (rsc // 64)
-            n_splits = max(1, min(1024 // tiles if tiles else 1, 128))
-            M = d.N * d.Ho * d.Wo
-            n_splits = max(1, min(n_splits, M // 32 or 1))
+            forced = os.environ.get("FLASHY_WGRAD_SPLITS")
+            if forced:  # e.g. =1 -> deterministic wgrad (no atomic splits)
+                n_splits = int(forced)
+            else:
+                tiles = (d.K // 64) * (rsc // 64)
+                n_splits = max(1, min(1024 // tiles if tiles else 1, 128))
+                M = d.N * d.Ho * d.Wo
+                n_splits = max(1, min(n_splits, M // 32 or 1))
         ext.conv_wgrad(x.data_ptr(), dout.data_ptr(), dw.data_ptr(), *d,
                        n_splits, _stream())
     else:
