@@ -15,6 +15,7 @@ flashy_amd/functional.py).
 from __future__ import annotations
 
 import importlib.util
+import os
 import typing as tp
 from pathlib import Path
 

@@ -168,11 +168,13 @@ k_conv_dgrad8(const uint16_t* __restrict__ dout,
                     (const char*)bbase + byte);
             }
             __builtin_amdgcn_s_setprio(1);
+            // operands swapped: 4 consecutive C channels per lane in the
+            // D-fragment -> one packed 8 B store (see conv_fwd8.hip)
 #pragma unroll
             for (int mf = 0; mf < MF; ++mf)
 #pragma unroll
                 for (int nf = 0; nf < NF; ++nf)
-                    acc[mf][nf] = MFMA_BF16(a[mf], b[nf], acc[mf][nf]);
+                    acc[mf][nf] = MFMA_BF16(b[nf], a[mf], acc[mf][nf]);
             __builtin_amdgcn_s_setprio(0);
         }
     };
@@ -188,19 +190,23 @@ k_conv_dgrad8(const uint16_t* __restrict__ dout,
     __builtin_amdgcn_s_barrier();
     compute_stage((n_stages - 1) % 3);
 
-    const int64_t out_row0 = m0 + wave_m * 128 + (lane >> 4) * 4;
-    const int out_col0 = col0 + wave_n * (BN / 4) + (lane & 15);
+    const int64_t out_row0 = m0 + wave_m * 128 + (lane & 15);
+    const int out_col0 = col0 + wave_n * (BN / 4) + (lane >> 4) * 4;
 #pragma unroll
-    for (int mf = 0; mf < MF; ++mf)
+    for (int mf = 0; mf < MF; ++mf) {
+        const int64_t row = out_row0 + mf * 16;
+        if (row < M) {
 #pragma unroll
-        for (int nf = 0; nf < NF; ++nf)
+            for (int nf = 0; nf < NF; ++nf) {
+                ushort4 pk;
 #pragma unroll
-            for (int rr = 0; rr < 4; ++rr) {
-                const int64_t row = out_row0 + mf * 16 + rr;
-                if (row < M)
-                    dx[row * d.C + out_col0 + nf * 16] =
-                        f32_to_bf16(acc[mf][nf][rr]);
+                for (int rr = 0; rr < 4; ++rr)
+                    ((uint16_t*)&pk)[rr] = f32_to_bf16(acc[mf][nf][rr]);
+                *reinterpret_cast<ushort4*>(
+                    dx + row * d.C + out_col0 + nf * 16) = pk;
             }
+        }
+    }
 }
 
 #include <cstdlib>

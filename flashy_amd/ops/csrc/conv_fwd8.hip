@@ -166,11 +166,17 @@ k_conv_fwd8(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
                     (const char*)bbase + byte);
             }
             __builtin_amdgcn_s_setprio(1);
+            // operands SWAPPED (weights first): the D-fragment then holds
+            // 4 CONSECUTIVE output channels per lane ((l>>4)*4+rr) at one
+            // pixel (l&15) — the epilogue packs them into one 8 B store
+            // instead of four scalar 2 B stores (the big-K 1x1 layers are
+            // store-issue-bound).  A- and B-fragment lane maps are
+            // transposes of each other, so the same LDS reads serve both.
 #pragma unroll
             for (int mf = 0; mf < MF; ++mf)
 #pragma unroll
                 for (int nf = 0; nf < NF; ++nf)
-                    acc[mf][nf] = MFMA_BF16(a[mf], b[nf], acc[mf][nf]);
+                    acc[mf][nf] = MFMA_BF16(b[nf], a[mf], acc[mf][nf]);
             __builtin_amdgcn_s_setprio(0);
         }
         // no tail barrier: wave skew is bounded by the next iteration's
@@ -191,48 +197,63 @@ k_conv_fwd8(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
     __builtin_amdgcn_s_barrier();
     compute_stage((n_stages - 1) % 3);
 
-    // --- epilogue: bf16 store (+relu), optional fused BN partials --------
-    const int64_t out_row0 = m0 + wave_m * 128 + (lane >> 4) * 4;
-    const int out_col0 = col0 + wave_n * (BN / 4) + (lane & 15);
+    // --- epilogue ---------------------------------------------------------
+    // Swapped-operand D layout: lane l of acc[mf][nf] holds pixel row
+    // (mf*16 + (l&15)) at output channels (nf*16 + (l>>4)*4 + rr) — four
+    // CONSECUTIVE channels per lane, packed into one 8 B store.
+    const int64_t out_row0 = m0 + wave_m * 128 + (lane & 15);
+    const int out_col0 = col0 + wave_n * (BN / 4) + (lane >> 4) * 4;
 #pragma unroll
-    for (int mf = 0; mf < MF; ++mf)
+    for (int mf = 0; mf < MF; ++mf) {
+        const int64_t row = out_row0 + mf * 16;
+        if (row < M) {
 #pragma unroll
-        for (int nf = 0; nf < NF; ++nf)
+            for (int nf = 0; nf < NF; ++nf) {
+                ushort4 pk;
 #pragma unroll
-            for (int rr = 0; rr < 4; ++rr) {
-                const int64_t row = out_row0 + mf * 16 + rr;
-                if (row < M) {
+                for (int rr = 0; rr < 4; ++rr) {
                     float v = acc[mf][nf][rr];
                     if (RELU) v = fmaxf(v, 0.f);
-                    y[row * d.K + out_col0 + nf * 16] = f32_to_bf16(v);
+                    ((uint16_t*)&pk)[rr] = f32_to_bf16(v);
                 }
+                *reinterpret_cast<ushort4*>(
+                    y + row * d.K + out_col0 + nf * 16) = pk;
             }
+        }
+    }
 
     if (bn_ws != nullptr) {
         __syncthreads();                 // main loop fully done: reuse lds
         float* sred = reinterpret_cast<float*>(lds);   // [2*WM][BN]
 #pragma unroll
         for (int nf = 0; nf < NF; ++nf) {
-            float s = 0.f, s2 = 0.f;
+            float s[4] = {}, s2[4] = {};
 #pragma unroll
-            for (int mf = 0; mf < MF; ++mf)
+            for (int mf = 0; mf < MF; ++mf) {
+                const int64_t row = out_row0 + mf * 16;
 #pragma unroll
                 for (int rr = 0; rr < 4; ++rr) {
-                    const int64_t row = out_row0 + mf * 16 + rr;
                     float v = acc[mf][nf][rr];
                     if (RELU) v = fmaxf(v, 0.f);
                     if (row >= M) v = 0.f;
-                    s += v;
-                    s2 = fmaf(v, v, s2);
+                    s[rr] += v;
+                    s2[rr] = fmaf(v, v, s2[rr]);
                 }
-            s += __shfl_xor(s, 16, 64);
-            s += __shfl_xor(s, 32, 64);
-            s2 += __shfl_xor(s2, 16, 64);
-            s2 += __shfl_xor(s2, 32, 64);
-            if (lane < 16) {
-                const int colL = wave_n * (BN / 4) + nf * 16 + lane;
-                sred[wave_m * BN + colL] = s;
-                sred[(2 + wave_m) * BN + colL] = s2;
+            }
+            // reduce over the 16 pixel lanes (low 4 lane bits)
+#pragma unroll
+            for (int off = 1; off < 16; off <<= 1)
+#pragma unroll
+                for (int rr = 0; rr < 4; ++rr) {
+                    s[rr] += __shfl_xor(s[rr], off, 64);
+                    s2[rr] += __shfl_xor(s2[rr], off, 64);
+                }
+            if ((lane & 15) == 0) {
+                const int colL = wave_n * (BN / 4) + nf * 16 + (lane >> 4) * 4;
+                *reinterpret_cast<float4*>(&sred[wave_m * BN + colL]) =
+                    make_float4(s[0], s[1], s[2], s[3]);
+                *reinterpret_cast<float4*>(&sred[(2 + wave_m) * BN + colL]) =
+                    make_float4(s2[0], s2[1], s2[2], s2[3]);
             }
         }
         __syncthreads();
