@@ -143,11 +143,14 @@ def save_state(state: tp.Any, path: tp.Union[str, Path]) -> None:
 class AsyncCheckpointer:
     """Overlap the pickle+disk half of a checkpoint with training.
 
-    EXPERIMENTAL: state-vs-file equality is verified
-    (scripts/async_ckpt_check.py, scripts/async_restore_check.py), but the
-    GPU cifar resume soak occasionally shows a first-epoch-after-resume
-    loss transient with this enabled that the sync path has not shown —
-    keep the default (sync) for production until that is root-caused.
+    Root-caused (round 2): the round-1 "first-epoch-after-resume loss
+    transient" was workload nondeterminism (wgrad atomic splits), not
+    checkpoint corruption — with deterministic wgrad
+    (FLASHY_WGRAD_SPLITS=1) a 6-round async vs 6-round sync GPU resume
+    soak shows zero bad resumes on either side
+    (scripts/soak_async_ab.py, profiles/r02c_soak_ab.json), on top of the
+    earlier state-vs-file equality checks (scripts/async_ckpt_check.py,
+    scripts/async_restore_check.py).
 
     ``save()`` blocks only for the device-to-host staging (tens of ms),
     then serializes and atomically renames on a background thread;

@@ -113,38 +113,61 @@ k_conv_wgrad(const uint16_t* __restrict__ x, const uint16_t* __restrict__ dout,
             stage_write(ndT, nxT, dv, xv);
             if (i + 2 < n_stages) load_pair(ms + (i + 2) * 2 * CONV_BK, dv, xv);
         }
+        // Issue BOTH subchunks' 16 transpose reads up front; a counted
+        // lgkmcnt(8) releases sc0's MFMAs while sc1's reads are still in
+        // flight — sc1's LDS traffic issues UNDER sc0's MFMA cluster
+        // (round-1 PMC: 27% issue-stall parked on the old per-sc
+        // lgkmcnt(0)).  offset:576 = +4 m rows; offset:4608 = +subchunk.
+        const unsigned a0 = (unsigned)(unsigned long long)(const void*)
+            &dT[tr_lane + wave_k * 32];
+        const unsigned a1 = a0 + 32;                    // +16 ch * 2 B
+        const unsigned b0 = (unsigned)(unsigned long long)(const void*)
+            &xTb[tr_lane + wave_j * 32];
+        const unsigned b1 = b0 + 32;
+        U64x2 af[2][2], bf[2][2];   // [sc][frag]
+        asm volatile(
+            "ds_read_b64_tr_b16 %0, %16\n\t"
+            "ds_read_b64_tr_b16 %1, %16 offset:576\n\t"
+            "ds_read_b64_tr_b16 %2, %17\n\t"
+            "ds_read_b64_tr_b16 %3, %17 offset:576\n\t"
+            "ds_read_b64_tr_b16 %4, %18\n\t"
+            "ds_read_b64_tr_b16 %5, %18 offset:576\n\t"
+            "ds_read_b64_tr_b16 %6, %19\n\t"
+            "ds_read_b64_tr_b16 %7, %19 offset:576\n\t"
+            "ds_read_b64_tr_b16 %8, %16 offset:4608\n\t"
+            "ds_read_b64_tr_b16 %9, %16 offset:5184\n\t"
+            "ds_read_b64_tr_b16 %10, %17 offset:4608\n\t"
+            "ds_read_b64_tr_b16 %11, %17 offset:5184\n\t"
+            "ds_read_b64_tr_b16 %12, %18 offset:4608\n\t"
+            "ds_read_b64_tr_b16 %13, %18 offset:5184\n\t"
+            "ds_read_b64_tr_b16 %14, %19 offset:4608\n\t"
+            "ds_read_b64_tr_b16 %15, %19 offset:5184\n\t"
+            "s_waitcnt lgkmcnt(8)"
+            : "=&v"(af[0][0].q.lo), "=&v"(af[0][0].q.hi),
+              "=&v"(af[0][1].q.lo), "=&v"(af[0][1].q.hi),
+              "=&v"(bf[0][0].q.lo), "=&v"(bf[0][0].q.hi),
+              "=&v"(bf[0][1].q.lo), "=&v"(bf[0][1].q.hi),
+              "=&v"(af[1][0].q.lo), "=&v"(af[1][0].q.hi),
+              "=&v"(af[1][1].q.lo), "=&v"(af[1][1].q.hi),
+              "=&v"(bf[1][0].q.lo), "=&v"(bf[1][0].q.hi),
+              "=&v"(bf[1][1].q.lo), "=&v"(bf[1][1].q.hi)
+            : "v"(a0), "v"(a1), "v"(b0), "v"(b1));
 #pragma unroll
-        for (int sc = 0; sc < 2; ++sc) {
-            const int base = sc * 32 * WG_P + tr_lane;
-            const unsigned a0 = (unsigned)(unsigned long long)(const void*)
-                &dT[base + wave_k * 32];
-            const unsigned a1 = a0 + 32;                    // +16 ch * 2 B
-            const unsigned b0 = (unsigned)(unsigned long long)(const void*)
-                &xTb[base + wave_j * 32];
-            const unsigned b1 = b0 + 32;
-            U64x2 af[2], bf[2];
-            // 8 transpose reads; offset:576 = +4 m rows (4 * WG_P * 2 B)
-            asm volatile(
-                "ds_read_b64_tr_b16 %0, %8\n\t"
-                "ds_read_b64_tr_b16 %1, %8 offset:576\n\t"
-                "ds_read_b64_tr_b16 %2, %9\n\t"
-                "ds_read_b64_tr_b16 %3, %9 offset:576\n\t"
-                "ds_read_b64_tr_b16 %4, %10\n\t"
-                "ds_read_b64_tr_b16 %5, %10 offset:576\n\t"
-                "ds_read_b64_tr_b16 %6, %11\n\t"
-                "ds_read_b64_tr_b16 %7, %11 offset:576\n\t"
-                "s_waitcnt lgkmcnt(0)"
-                : "=v"(af[0].q.lo), "=v"(af[0].q.hi),
-                  "=v"(af[1].q.lo), "=v"(af[1].q.hi),
-                  "=v"(bf[0].q.lo), "=v"(bf[0].q.hi),
-                  "=v"(bf[1].q.lo), "=v"(bf[1].q.hi)
-                : "v"(a0), "v"(a1), "v"(b0), "v"(b1));
+        for (int kf = 0; kf < 2; ++kf)
 #pragma unroll
-            for (int kf = 0; kf < 2; ++kf)
+            for (int jf = 0; jf < 2; ++jf)
+                acc[kf][jf] = MFMA_BF16(af[0][kf].v, bf[0][jf].v, acc[kf][jf]);
+        asm volatile("s_waitcnt lgkmcnt(0)"
+                     : "+v"(af[1][0].q.lo), "+v"(af[1][0].q.hi),
+                       "+v"(af[1][1].q.lo), "+v"(af[1][1].q.hi),
+                       "+v"(bf[1][0].q.lo), "+v"(bf[1][0].q.hi),
+                       "+v"(bf[1][1].q.lo), "+v"(bf[1][1].q.hi));
+        __builtin_amdgcn_sched_barrier(0);
 #pragma unroll
-                for (int jf = 0; jf < 2; ++jf)
-                    acc[kf][jf] = MFMA_BF16(af[kf].v, bf[jf].v, acc[kf][jf]);
-        }
+        for (int kf = 0; kf < 2; ++kf)
+#pragma unroll
+            for (int jf = 0; jf < 2; ++jf)
+                acc[kf][jf] = MFMA_BF16(af[1][kf].v, bf[1][jf].v, acc[kf][jf]);
         __syncthreads();
     };
     for (int64_t i = 0; i < n_stages;) {
