@@ -471,6 +471,14 @@ extern "C" void launch_bn_bwd_apply(const void* dz, const void* x,
                                     const void* work, const void* bsums,
                                     void* dx, int64_t M, int C,
                                     hipStream_t stream) {
+    if (C % 64 == 0) {
+        const int mpb = apply_mpb(M, C);
+        dim3 g((unsigned)(C / 64), (unsigned)((M + mpb - 1) / mpb));
+        k_bn_bwd_apply_c64<<<g, 256, 0, stream>>>(
+            (const uint16_t*)dz, (const uint16_t*)x, (const float*)work,
+            (const float*)bsums, (uint16_t*)dx, M, C, mpb);
+        return;
+    }
     const int grid = ew_grid(M * C / 8, 256, 4);
     k_bn_bwd_apply<<<grid, 256, 0, stream>>>(
         (const uint16_t*)dz, (const uint16_t*)x, (const float*)work,

@@ -17,7 +17,11 @@
 
 #define OOB_SENTINEL 0xF0000000u
 
-template <int BN, bool S1>
+// SCAT2: 1x1 stride-2 dgrad as a quarter-size GEMM — dx is nonzero only at
+// even (hi, wi), so A rows are the OUTPUT pixels read linearly from dy and
+// the epilogue scatters each result to (2ho, 2wo) while writing the three
+// odd-position siblings as zeros (no zero-filled MFMA work, no memset pass).
+template <int BN, bool S1, bool SCAT2 = false>
 __global__ void __launch_bounds__(512, 2)
 k_conv_dgrad8(const uint16_t* __restrict__ dout,
               const uint16_t* __restrict__ w_rsck,
@@ -32,7 +36,8 @@ k_conv_dgrad8(const uint16_t* __restrict__ dout,
     constexpr int G = 4 + BN / 64;
 
     const int rsk = d.R * d.S * d.K;
-    const int64_t M = (int64_t)d.N * d.H * d.W;
+    const int64_t M = SCAT2 ? (int64_t)d.N * d.Ho * d.Wo
+                            : (int64_t)d.N * d.H * d.W;
     const int tid = threadIdx.x;
     const int lane = tid & 63;
     const int wid_u = __builtin_amdgcn_readfirstlane(tid >> 6);
@@ -59,7 +64,9 @@ k_conv_dgrad8(const uint16_t* __restrict__ dout,
         const int row = chunk >> 3;
         const int kc_s = (chunk & 7) ^ (row & 7);
         const int64_t m = m0 + row;
-        if (m < M) {
+        if (SCAT2) {
+            a_n[g] = m < M ? m : -1;   // linear dy row (1x1 over out grid)
+        } else if (m < M) {
             const int hw = d.H * d.W;
             a_n[g] = m / hw;
             const int rem = (int)(m % hw);
@@ -94,7 +101,11 @@ k_conv_dgrad8(const uint16_t* __restrict__ dout,
 #pragma unroll
         for (int g = 0; g < 4; ++g) {
             unsigned voff = OOB_SENTINEL;
-            if (a_n[g] >= 0 && a_r[g] < d.R) {
+            if (SCAT2) {
+                if (a_n[g] >= 0)
+                    voff = (unsigned)((a_n[g] * d.K + a_k[g]) * 2);
+                a_k[g] += BK;   // 1x1: k never wraps within rsk
+            } else if (a_n[g] >= 0 && a_r[g] < d.R) {
                 const int hnum = a_hi[g] - a_r[g];   // = ho * stride
                 const int wnum = a_wi[g] - a_s[g];
                 const int ho = S1 ? hnum : hnum / d.stride;
@@ -105,12 +116,14 @@ k_conv_dgrad8(const uint16_t* __restrict__ dout,
                     voff = (unsigned)((((a_n[g] * d.Ho + ho) * d.Wo + wo) *
                                        (int64_t)d.K + a_k[g]) * 2);
             }
-            int k = a_k[g] + BK;
-            while (k >= d.K) {
-                k -= d.K;
-                if (++a_s[g] == d.S) { a_s[g] = 0; ++a_r[g]; }
+            if (!SCAT2) {
+                int k = a_k[g] + BK;
+                while (k >= d.K) {
+                    k -= d.K;
+                    if (++a_s[g] == d.S) { a_s[g] = 0; ++a_r[g]; }
+                }
+                a_k[g] = k;
             }
-            a_k[g] = k;
             __builtin_amdgcn_raw_ptr_buffer_load_lds(
                 arsrc,
                 (__attribute__((address_space(3))) void*)
@@ -196,14 +209,30 @@ k_conv_dgrad8(const uint16_t* __restrict__ dout,
     for (int mf = 0; mf < MF; ++mf) {
         const int64_t row = out_row0 + mf * 16;
         if (row < M) {
+            int64_t obase;
+            if (SCAT2) {
+                const int wo = (int)(row % d.Wo);
+                const int ho = (int)((row / d.Wo) % d.Ho);
+                const int64_t n = row / ((int64_t)d.Ho * d.Wo);
+                obase = ((n * d.H + 2 * ho) * d.W + 2 * wo) * (int64_t)d.C;
+            } else {
+                obase = row * d.C;
+            }
 #pragma unroll
             for (int nf = 0; nf < NF; ++nf) {
                 ushort4 pk;
 #pragma unroll
                 for (int rr = 0; rr < 4; ++rr)
                     ((uint16_t*)&pk)[rr] = f32_to_bf16(acc[mf][nf][rr]);
-                *reinterpret_cast<ushort4*>(
-                    dx + row * d.C + out_col0 + nf * 16) = pk;
+                const int64_t o = obase + out_col0 + nf * 16;
+                *reinterpret_cast<ushort4*>(dx + o) = pk;
+                if (SCAT2) {   // zero the three odd-position siblings
+                    const ushort4 z = {};
+                    *reinterpret_cast<ushort4*>(dx + o + d.C) = z;
+                    *reinterpret_cast<ushort4*>(dx + o + (int64_t)d.W * d.C) = z;
+                    *reinterpret_cast<ushort4*>(
+                        dx + o + (int64_t)d.W * d.C + d.C) = z;
+                }
             }
         }
     }

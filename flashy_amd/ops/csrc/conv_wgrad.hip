@@ -192,10 +192,208 @@ k_conv_wgrad(const uint16_t* __restrict__ x, const uint16_t* __restrict__ dout,
             }
 }
 
+// ---------------------------------------------------------------------------
+// 8-wave 128(K) x 128(rsc) wgrad: double the tile edge of the kernel above,
+// HALVING both operands' re-read traffic (dout re-read rsc/128 times and x
+// re-read K/128 times instead of /64) — round-2 profiling put wgrad at 21%
+// of the ResNet-50/224 step, bound by this on-chip traffic.  Same
+// tr_b16-transpose-read scheme; pitch 144 elems (72 dwords ≡ 8 mod 64:
+// conflict-free row banks); 512 threads = 8 waves as 2(K) x 4(rsc); both
+// subchunks' reads issued ahead of the MFMA clusters behind counted
+// lgkmcnt.  Requires K % 128 == 0, rsc % 128 == 0.
+// ---------------------------------------------------------------------------
+
+#define WG_P8 144  // bf16 elems per LDS row (128 data + 16 pad)
+
+__global__ void __launch_bounds__(512, 2)
+k_conv_wgrad8(const uint16_t* __restrict__ x, const uint16_t* __restrict__ dout,
+              float* __restrict__ dw, ConvDims d, int m_per_split) {
+    const int rsc = d.R * d.S * d.C;
+    const int64_t M = (int64_t)d.N * d.Ho * d.Wo;
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6;
+    const int wave_k = wid >> 2;   // 0..1: 64-K half
+    const int wave_j = wid & 3;    // 0..3: 32-rsc quarter
+    const int k0 = blockIdx.x * 128;
+    const int j0 = blockIdx.y * 128;
+    const int64_t ms = (int64_t)blockIdx.z * m_per_split;
+    const int64_t me = min(ms + (int64_t)m_per_split, M);
+
+    // [buffer][sc 2][m 32][WG_P8] rows as loaded
+    __shared__ __attribute__((aligned(16))) uint16_t doutT[2][2 * 32 * WG_P8];
+    __shared__ __attribute__((aligned(16))) uint16_t xT[2][2 * 32 * WG_P8];
+
+    floatx4 acc[4][2] = {};
+
+    const int m_r = tid >> 3;            // 0..63 (both subchunks)
+    const int k8 = (tid & 7) * 8;        // channel octet within 64
+    // two fixed taps per thread: j0 + k8 and j0 + 64 + k8
+    int t_hoff[2], t_woff[2], t_c[2];
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+        const int jj = j0 + h * 64 + k8;
+        const int r = jj / (d.S * d.C);
+        const int scc = jj - r * d.S * d.C;
+        const int s = scc / d.C;
+        t_c[h] = scc - s * d.C;
+        t_hoff[h] = r - d.pad;
+        t_woff[h] = s - d.pad;
+    }
+    auto load_pair = [&](int64_t mc, short8* dv, short8* xv) {
+        const int64_t m = mc + m_r;
+        short8 z = {};
+        dv[0] = dv[1] = xv[0] = xv[1] = z;
+        if (m < me) {
+            dv[0] = *reinterpret_cast<const short8*>(dout + m * d.K + k0 + k8);
+            dv[1] = *reinterpret_cast<const short8*>(
+                dout + m * d.K + k0 + 64 + k8);
+            const int wo = (int)(m % d.Wo);
+            const int ho = (int)((m / d.Wo) % d.Ho);
+            const int64_t n = m / ((int64_t)d.Ho * d.Wo);
+#pragma unroll
+            for (int h = 0; h < 2; ++h) {
+                const int hi = ho * d.stride + t_hoff[h];
+                const int wi = wo * d.stride + t_woff[h];
+                if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W)
+                    xv[h] = *reinterpret_cast<const short8*>(
+                        x + (((n * d.H + hi) * d.W + wi) * (int64_t)d.C +
+                             t_c[h]));
+            }
+        }
+    };
+    auto stage_write = [&](uint16_t (&dT)[2 * 32 * WG_P8],
+                           uint16_t (&xTb)[2 * 32 * WG_P8],
+                           const short8* dv, const short8* xv) {
+        const int at = m_r * WG_P8 + k8;   // [sc(m_r>>5)][m_r&31] row-major
+#pragma unroll
+        for (int h = 0; h < 2; ++h) {
+            *reinterpret_cast<short8*>(&dT[at + h * 64]) = dv[h];
+            *reinterpret_cast<short8*>(&xTb[at + h * 64]) = xv[h];
+        }
+    };
+
+    const int tr_lane = ((lane & 15) >> 2) * WG_P8 + 4 * (lane & 3) +
+                        (lane >> 4) * 8 * WG_P8;
+    union U64x2 { struct { unsigned long long lo, hi; } q; short8 v; };
+
+    const int64_t n_stages = (me - ms + 2 * CONV_BK - 1) / (2 * CONV_BK);
+    short8 dv[2], xv[2];
+    load_pair(ms, dv, xv);
+    stage_write(doutT[0], xT[0], dv, xv);
+    if (n_stages > 1) load_pair(ms + 2 * CONV_BK, dv, xv);
+    __syncthreads();
+
+    // per-sc fragment reads: dout 4 frags (wave_k*64 + kf*16), x 2 frags
+    // (wave_j*32 + jf*16); offsets: +4 m rows = 1152 B, +subchunk = 9216 B
+    auto step = [&](int64_t i, const uint16_t (&dT)[2 * 32 * WG_P8],
+                    const uint16_t (&xTb)[2 * 32 * WG_P8],
+                    uint16_t (&ndT)[2 * 32 * WG_P8],
+                    uint16_t (&nxT)[2 * 32 * WG_P8]) {
+        if (i + 1 < n_stages) {
+            stage_write(ndT, nxT, dv, xv);
+            if (i + 2 < n_stages) load_pair(ms + (i + 2) * 2 * CONV_BK, dv, xv);
+        }
+        const unsigned a0 = (unsigned)(unsigned long long)(const void*)
+            &dT[tr_lane + wave_k * 64];
+        const unsigned b0 = (unsigned)(unsigned long long)(const void*)
+            &xTb[tr_lane + wave_j * 32];
+        U64x2 af[2][4], bf[2][2];   // [sc][frag]
+#pragma unroll
+        for (int sc = 0; sc < 2; ++sc) {
+            // 12 reads per subchunk in one asm block (24 outs + 2 addr ins)
+            asm volatile(
+                "ds_read_b64_tr_b16 %0, %12\n\t"
+                "ds_read_b64_tr_b16 %1, %12 offset:1152\n\t"
+                "ds_read_b64_tr_b16 %2, %12 offset:32\n\t"
+                "ds_read_b64_tr_b16 %3, %12 offset:1184\n\t"
+                "ds_read_b64_tr_b16 %4, %12 offset:64\n\t"
+                "ds_read_b64_tr_b16 %5, %12 offset:1216\n\t"
+                "ds_read_b64_tr_b16 %6, %12 offset:96\n\t"
+                "ds_read_b64_tr_b16 %7, %12 offset:1248\n\t"
+                "ds_read_b64_tr_b16 %8, %13\n\t"
+                "ds_read_b64_tr_b16 %9, %13 offset:1152\n\t"
+                "ds_read_b64_tr_b16 %10, %13 offset:32\n\t"
+                "ds_read_b64_tr_b16 %11, %13 offset:1184\n\t"
+                : "=&v"(af[sc][0].q.lo), "=&v"(af[sc][0].q.hi),
+                  "=&v"(af[sc][1].q.lo), "=&v"(af[sc][1].q.hi),
+                  "=&v"(af[sc][2].q.lo), "=&v"(af[sc][2].q.hi),
+                  "=&v"(af[sc][3].q.lo), "=&v"(af[sc][3].q.hi),
+                  "=&v"(bf[sc][0].q.lo), "=&v"(bf[sc][0].q.hi),
+                  "=&v"(bf[sc][1].q.lo), "=&v"(bf[sc][1].q.hi)
+                : "v"(a0 + sc * 9216), "v"(b0 + sc * 9216));
+        }
+#pragma unroll
+        for (int sc = 0; sc < 2; ++sc) {
+            if (sc == 0)
+                asm volatile("s_waitcnt lgkmcnt(12)"
+                             : "+v"(af[0][0].q.lo), "+v"(af[0][0].q.hi),
+                               "+v"(af[0][1].q.lo), "+v"(af[0][1].q.hi),
+                               "+v"(af[0][2].q.lo), "+v"(af[0][2].q.hi),
+                               "+v"(af[0][3].q.lo), "+v"(af[0][3].q.hi),
+                               "+v"(bf[0][0].q.lo), "+v"(bf[0][0].q.hi),
+                               "+v"(bf[0][1].q.lo), "+v"(bf[0][1].q.hi));
+            else
+                asm volatile("s_waitcnt lgkmcnt(0)"
+                             : "+v"(af[1][0].q.lo), "+v"(af[1][0].q.hi),
+                               "+v"(af[1][1].q.lo), "+v"(af[1][1].q.hi),
+                               "+v"(af[1][2].q.lo), "+v"(af[1][2].q.hi),
+                               "+v"(af[1][3].q.lo), "+v"(af[1][3].q.hi),
+                               "+v"(bf[1][0].q.lo), "+v"(bf[1][0].q.hi),
+                               "+v"(bf[1][1].q.lo), "+v"(bf[1][1].q.hi));
+            __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+            for (int kf = 0; kf < 4; ++kf)
+#pragma unroll
+                for (int jf = 0; jf < 2; ++jf)
+                    acc[kf][jf] =
+                        MFMA_BF16(af[sc][kf].v, bf[sc][jf].v, acc[kf][jf]);
+        }
+        __syncthreads();
+    };
+    for (int64_t i = 0; i < n_stages;) {
+        step(i, doutT[0], xT[0], doutT[1], xT[1]);
+        if (++i >= n_stages) break;
+        step(i, doutT[1], xT[1], doutT[0], xT[0]);
+        ++i;
+    }
+
+    const int out_k0 = k0 + wave_k * 64 + (lane >> 4) * 4;
+    const int out_j0 = j0 + wave_j * 32 + (lane & 15);
+#pragma unroll
+    for (int kf = 0; kf < 4; ++kf)
+#pragma unroll
+        for (int jf = 0; jf < 2; ++jf)
+#pragma unroll
+            for (int rr = 0; rr < 4; ++rr) {
+                const int k = out_k0 + kf * 16 + rr;
+                const int j = out_j0 + jf * 16;
+                atomicAdd(&dw[(int64_t)k * rsc + j], acc[kf][jf][rr]);
+            }
+}
+
 extern "C" void launch_conv_wgrad(const void* x, const void* dout, void* dw,
                                   ConvDims d, int n_splits, hipStream_t stream) {
     const int rsc = d.R * d.S * d.C;
     const int64_t M = (int64_t)d.N * d.Ho * d.Wo;
+    if (d.K % 128 == 0 && rsc % 128 == 0) {
+        const int tiles = (d.K / 128) * (rsc / 128);
+        int ns = n_splits;
+        if (ns > 1) {   // re-target for 512-thread blocks (one per CU)
+            ns = 512 / tiles;
+            if (ns < 1) ns = 1;
+            if (ns > 128) ns = 128;
+            const int64_t mcap = M / 64;
+            if (ns > mcap) ns = (int)(mcap ? mcap : 1);
+        }
+        int mps = (int)((M + ns - 1) / ns);
+        mps = (mps + 2 * CONV_BK - 1) / (2 * CONV_BK) * (2 * CONV_BK);
+        const int zn = (int)((M + mps - 1) / mps);
+        dim3 grid((unsigned)(d.K / 128), (unsigned)(rsc / 128), (unsigned)zn);
+        k_conv_wgrad8<<<grid, 512, 0, stream>>>(
+            (const uint16_t*)x, (const uint16_t*)dout, (float*)dw, d, mps);
+        return;
+    }
     int m_per_split = (int)((M + n_splits - 1) / n_splits);
     m_per_split = (m_per_split + 2 * CONV_BK - 1) / (2 * CONV_BK) * (2 * CONV_BK);
     const int zn = (int)((M + m_per_split - 1) / m_per_split);
