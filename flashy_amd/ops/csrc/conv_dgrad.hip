@@ -211,6 +211,10 @@ extern "C" int conv_dgrad8_plan(ConvDims d, int* bn_out);
 extern "C" void launch_conv_dgrad8(const void* dout, const void* w_rsck,
                                    void* dx, ConvDims d, int bn, int mtiles,
                                    hipStream_t stream);
+extern "C" int conv_dgrad8_s2_plan(ConvDims d, int* bn_out);
+extern "C" void launch_conv_dgrad8_s2(const void* dout, const void* w_rsck,
+                                      void* dx, ConvDims d, int bn,
+                                      int mtiles, hipStream_t stream);
 
 extern "C" void launch_conv_dgrad(const void* dout, const void* w_rsck,
                                   void* dx, ConvDims d, hipStream_t stream) {
@@ -218,6 +222,14 @@ extern "C" void launch_conv_dgrad(const void* dout, const void* w_rsck,
     auto dd = (const uint16_t*)dout;
     auto ww = (const uint16_t*)w_rsck;
     auto xx = (uint16_t*)dx;
+    if (d.stride == 2) {
+        int bn8;   // 1x1 s2: quarter-size GEMM + sibling zero scatter
+        const int mt8 = conv_dgrad8_s2_plan(d, &bn8);
+        if (mt8) {
+            launch_conv_dgrad8_s2(dout, w_rsck, dx, d, bn8, mt8, stream);
+            return;
+        }
+    }
     if (d.stride == 2 &&
         launch_conv_dgrad_s2(dout, w_rsck, dx, d, stream))
         return;  // parity-class form (no zero-filled MFMA work)

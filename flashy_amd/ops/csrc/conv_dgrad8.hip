@@ -261,6 +261,41 @@ extern "C" int conv_dgrad8_plan(ConvDims d, int* bn_out) {
     return 0;
 }
 
+// 1x1 stride-2 plan: quarter-size GEMM over the output grid (SCAT2)
+extern "C" int conv_dgrad8_s2_plan(ConvDims d, int* bn_out) {
+    if (d.R != 1 || d.S != 1 || d.stride != 2) return 0;
+    if (d.C % 64 || d.K % 64) return 0;
+    const int64_t Mq = (int64_t)d.N * d.Ho * d.Wo;
+    const int64_t dout_elems = Mq * d.K;
+    if (dout_elems * 2 >= (int64_t)OOB_SENTINEL) return 0;
+    if (d.W % 2 || d.H % 2) return 0;   // sibling zero-stores assume even
+    const int mtiles = (int)((Mq + 255) / 256);
+    if (d.C % 128 == 0 && (int64_t)mtiles * (d.C / 128) >= 104) {
+        *bn_out = 128;
+        return mtiles;
+    }
+    if ((int64_t)mtiles * (d.C / 64) >= 104) {
+        *bn_out = 64;
+        return mtiles;
+    }
+    return 0;
+}
+
+extern "C" void launch_conv_dgrad8_s2(const void* dout, const void* w_rsck,
+                                      void* dx, ConvDims d, int bn,
+                                      int mtiles, hipStream_t stream) {
+    dim3 grid((unsigned)mtiles, (unsigned)(d.C / bn));
+    const unsigned db =
+        (unsigned)((int64_t)d.N * d.Ho * d.Wo * d.K * 2);
+    auto dd = (const uint16_t*)dout;
+    auto ww = (const uint16_t*)w_rsck;
+    auto xx = (uint16_t*)dx;
+    if (bn == 128)
+        k_conv_dgrad8<128, false, true><<<grid, 512, 0, stream>>>(dd, ww, xx, d, db);
+    else
+        k_conv_dgrad8<64, false, true><<<grid, 512, 0, stream>>>(dd, ww, xx, d, db);
+}
+
 extern "C" void launch_conv_dgrad8(const void* dout, const void* w_rsck,
                                    void* dx, ConvDims d, int bn, int mtiles,
                                    hipStream_t stream) {

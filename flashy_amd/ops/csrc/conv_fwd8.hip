@@ -267,6 +267,85 @@ k_conv_fwd8(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
     }
 }
 
+// --- stem padding helpers ---------------------------------------------------
+// The C=3 7x7 stem is run through the 8-wave kernel by padding to C'=4,
+// R'=S'=8 (taps beyond 7x7 carry zero weights; borders are zero pixels), so
+// every 16 B glds chunk is two contiguous (w, w+1) pixels and rsc' = 256.
+// Useful/total MFMA work = 147/256, vs the direct stem kernel's 47 TF/s.
+
+__global__ void __launch_bounds__(256)
+k_stem_pad_x(const uint16_t* __restrict__ x, uint16_t* __restrict__ xp,
+             int64_t N, int H, int W, int Hp, int Wp, int pad) {
+    const int64_t total = N * Hp * Wp;
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+         i += stride) {
+        const int wp_ = (int)(i % Wp);
+        const int hp_ = (int)((i / Wp) % Hp);
+        const int64_t n = i / ((int64_t)Hp * Wp);
+        const int h = hp_ - pad, w = wp_ - pad;
+        ushort4 v = {};
+        if (h >= 0 && h < H && w >= 0 && w < W) {
+            const uint16_t* src = x + ((n * H + h) * (int64_t)W + w) * 3;
+            v.x = src[0];
+            v.y = src[1];
+            v.z = src[2];
+        }
+        *reinterpret_cast<ushort4*>(xp + i * 4) = v;
+    }
+}
+
+__global__ void __launch_bounds__(256)
+k_stem_pad_w(const uint16_t* __restrict__ w, uint16_t* __restrict__ wp,
+             int K, int R, int S) {
+    // wp[k][8][8][4] from w[k][R][S][3]
+    const int total = K * 64;
+    for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+         i += gridDim.x * blockDim.x) {
+        const int s = i % 8, r = (i / 8) % 8, k = i / 64;
+        ushort4 v = {};
+        if (r < R && s < S) {
+            const uint16_t* src = w + ((k * R + r) * S + s) * 3;
+            v.x = src[0];
+            v.y = src[1];
+            v.z = src[2];
+        }
+        *reinterpret_cast<ushort4*>(wp + i * 4) = v;
+    }
+}
+
+__global__ void __launch_bounds__(256)
+k_stem_unpad_dw(const float* __restrict__ dwp, float* __restrict__ dw,
+                int K, int R, int S) {
+    // dw[k][R][S][3] += dwp[k][8][8][4]
+    const int total = K * R * S * 3;
+    for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+         i += gridDim.x * blockDim.x) {
+        const int c = i % 3, s = (i / 3) % S, r = (i / (3 * S)) % R,
+                  k = i / (3 * S * R);
+        dw[i] += dwp[((k * 8 + r) * 8 + s) * 4 + c];
+    }
+}
+
+extern "C" void launch_stem_pad_x(const void* x, void* xp, int64_t N, int H,
+                                  int W, int Hp, int Wp, int pad,
+                                  hipStream_t stream) {
+    k_stem_pad_x<<<ew_grid(N * Hp * Wp, 256, 2), 256, 0, stream>>>(
+        (const uint16_t*)x, (uint16_t*)xp, N, H, W, Hp, Wp, pad);
+}
+
+extern "C" void launch_stem_pad_w(const void* w, void* wp, int K, int R,
+                                  int S, hipStream_t stream) {
+    k_stem_pad_w<<<(K * 64 + 255) / 256, 256, 0, stream>>>(
+        (const uint16_t*)w, (uint16_t*)wp, K, R, S);
+}
+
+extern "C" void launch_stem_unpad_dw(const void* dwp, void* dw, int K, int R,
+                                     int S, hipStream_t stream) {
+    k_stem_unpad_dw<<<(K * R * S * 3 + 255) / 256, 256, 0, stream>>>(
+        (const float*)dwp, (float*)dw, K, R, S);
+}
+
 // --- dispatch ---------------------------------------------------------------
 // Eligibility + grid for the 8-wave kernel; returns grid.x (m-tiles) or 0.
 // Decided purely from the dims so conv_fwd_msplit (BN-partials sizing) and

@@ -376,7 +376,11 @@ extern "C" void launch_conv_wgrad(const void* x, const void* dout, void* dw,
                                   ConvDims d, int n_splits, hipStream_t stream) {
     const int rsc = d.R * d.S * d.C;
     const int64_t M = (int64_t)d.N * d.Ho * d.Wo;
-    if (d.K % 128 == 0 && rsc % 128 == 0) {
+    // 128x128 tile only where re-read traffic dominates (R*S > 1): the 1x1
+    // wgrads already run at the HBM roof on the 64x64 kernel's 4-blocks/CU
+    // occupancy, and the 1-block/CU 8-wave kernel loses TLP there (measured:
+    // 1x1 196->156 TF, 3x3 186->244 TF).
+    if (d.R * d.S > 1 && d.K % 128 == 0 && rsc % 128 == 0) {
         const int tiles = (d.K / 128) * (rsc / 128);
         int ns = n_splits;
         if (ns > 1) {   // re-target for 512-thread blocks (one per CU)

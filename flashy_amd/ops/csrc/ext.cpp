@@ -26,6 +26,15 @@ void launch_conv_fwd(const void* x, const void* w, void* y, ConvDims d,
 int conv_fwd_msplit(ConvDims d);
 int conv_fwd8_plan(ConvDims d, int* bn_out);
 int conv_dgrad8_plan(ConvDims d, int* bn_out);
+void launch_conv_fwd8(const void* x, const void* w, void* y, ConvDims d,
+                      int relu, void* bn_ws, int bn, int mtiles,
+                      hipStream_t stream);
+void launch_stem_pad_x(const void* x, void* xp, int64_t N, int H, int W,
+                       int Hp, int Wp, int pad, hipStream_t stream);
+void launch_stem_pad_w(const void* w, void* wp, int K, int R, int S,
+                       hipStream_t stream);
+void launch_stem_unpad_dw(const void* dwp, void* dw, int K, int R, int S,
+                          hipStream_t stream);
 void launch_conv_stem_fwd(const void* x, const void* w, void* y, ConvDims d,
                           hipStream_t stream);
 void launch_conv_stem_wgrad(const void* x, const void* dout, void* dw,

@@ -21,6 +21,10 @@ SHAPES = [
     (3, 7, 5, 64, 64, 3, 1, 1),       # non-pow2 spatial, odd M tail
     (2, 16, 16, 64, 128, 4, 2, 1),    # even kernel, stride 2 (GAN shapes)
     (2, 32, 32, 3, 64, 7, 2, 3),      # ImageNet-style stem (C=3, chunked taps)
+    # large-M shapes that engage the 8-wave deep-pipeline kernels
+    (64, 28, 28, 128, 128, 3, 1, 1),  # fwd8/dgrad8 BN=128 + wgrad8
+    (64, 29, 28, 128, 128, 3, 1, 1),  # same with an M tail (odd rows)
+    (16, 56, 56, 256, 512, 1, 2, 0),  # 1x1 stride-2: dgrad8 SCAT2 scatter
 ]
 
 
