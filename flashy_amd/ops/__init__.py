@@ -203,9 +203,47 @@ def conv_fwd(x: torch.Tensor, w: torch.Tensor, y: torch.Tensor, d: ConvDims,
             return stats
     else:
         assert not relu
+        M = d.N * d.Ho * d.Wo
+        if d.C == 3 and d.K % 64 == 0 and d.R <= 8 and d.S <= 8 \
+                and M >= 100_000:
+            return _stem_fwd8(ext, x, w, y, d, want_stats)
         assert d.R * d.S * d.C <= 160 and d.K == 64, d
         ext.conv_stem_fwd(x.data_ptr(), w.data_ptr(), y.data_ptr(), *d, _stream())
     return None
+
+
+def _stem_pad_x(ext, x: torch.Tensor, d: ConvDims) -> tp.Tuple[torch.Tensor, int, int]:
+    """Zero-padded C'=4 / 8x8-tap copy of the C=3 stem input (see
+    csrc/conv_fwd8.hip stem helpers)."""
+    Hp = (d.Ho - 1) * d.stride + 8
+    Wp = (d.Wo - 1) * d.stride + 8
+    xp = torch.empty(d.N, Hp, Wp, 4, dtype=torch.bfloat16, device=x.device)
+    ext.stem_pad_x(x.data_ptr(), xp.data_ptr(), d.N, d.H, d.W, Hp, Wp, d.pad,
+                   _stream())
+    return xp, Hp, Wp
+
+
+def _stem_fwd8(ext, x, w, y, d: ConvDims, want_stats: bool):
+    """Large-M C=3 stem through the 8-wave kernel: pad channels to 4 and
+    taps to 8x8 (zero weights), rsc' = 256 — the direct small-C kernel runs
+    at ~47 TF/s on the 224px stem; this path reaches the implicit-GEMM rate
+    at 147/256 useful work."""
+    xp, Hp, Wp = _stem_pad_x(ext, x, d)
+    wp = torch.empty(d.K, 8, 8, 4, dtype=torch.bfloat16, device=x.device)
+    ext.stem_pad_w(w.data_ptr(), wp.data_ptr(), d.K, d.R, d.S, _stream())
+    M = d.N * d.Ho * d.Wo
+    mtiles = (M + 255) // 256
+    stats = None
+    bn_ptr = 0
+    if want_stats:
+        partials = torch.empty(2 * d.K * mtiles, dtype=torch.float32,
+                               device=x.device)
+        stats = (partials, mtiles)
+        bn_ptr = partials.data_ptr()
+    ext.conv_fwd8_direct(xp.data_ptr(), wp.data_ptr(), y.data_ptr(),
+                         d.N, Hp, Wp, 4, d.K, 8, 8, d.Ho, d.Wo, d.stride, 0,
+                         False, bn_ptr, 64, mtiles, _stream())
+    return stats
 
 
 def conv_stem_dgrad(dout: torch.Tensor, w_krsc: torch.Tensor,
@@ -273,6 +311,24 @@ def conv_wgrad(x: torch.Tensor, dout: torch.Tensor, dw: torch.Tensor,
         ext.conv_wgrad(x.data_ptr(), dout.data_ptr(), dw.data_ptr(), *d,
                        n_splits, _stream())
     else:
+        M = d.N * d.Ho * d.Wo
+        if d.C == 3 and d.K % 64 == 0 and d.R <= 8 and d.S <= 8 \
+                and M >= 100_000:
+            # padded-stem wgrad: pad x to C'=4 / 8x8 taps, run the regular
+            # implicit-GEMM wgrad on rsc'=256, unpad-accumulate into dw
+            xp, Hp, Wp = _stem_pad_x(ext, x, d)
+            dwp = torch.zeros(d.K * 256, dtype=torch.float32, device=x.device)
+            dp = ConvDims(d.N, Hp, Wp, 4, d.K, 8, 8, d.Ho, d.Wo, d.stride, 0)
+            tiles = (d.K // 64) * 4
+            ns = max(1, min(1024 // tiles, 128, M // 32 or 1))
+            forced = os.environ.get("FLASHY_WGRAD_SPLITS")
+            if forced:
+                ns = int(forced)
+            ext.conv_wgrad(xp.data_ptr(), dout.data_ptr(), dwp.data_ptr(),
+                           *dp, ns, _stream())
+            ext.stem_unpad_dw(dwp.data_ptr(), dw.data_ptr(), d.K, d.R, d.S,
+                              _stream())
+            return
         assert rsc <= 160, d  # small-C stem kernels stage taps/weights in LDS
         ext.conv_stem_wgrad(x.data_ptr(), dout.data_ptr(), dw.data_ptr(), *d,
                             _stream())

@@ -134,6 +134,43 @@ PYBIND11_MODULE(_hip_ops, m) {
                               relu ? 1 : 0, (void*)bn_ws, as_stream(stream));
               check_last();
           });
+    m.def("conv_fwd8_direct",
+          [](uintptr_t x, uintptr_t w, uintptr_t y, int N, int H, int W,
+             int C, int K, int R, int S, int Ho, int Wo, int stride, int pad,
+             bool relu, uintptr_t bn_ws, int bn, int mtiles,
+             uintptr_t stream) {
+              launch_conv_fwd8((const void*)x, (const void*)w, (void*)y,
+                               make_dims(N, H, W, C, K, R, S, Ho, Wo, stride,
+                                         pad),
+                               relu, (void*)bn_ws, bn, mtiles,
+                               as_stream(stream));
+              check_last();
+          });
+
+    m.def("stem_pad_x",
+          [](uintptr_t x, uintptr_t xp, int64_t N, int H, int W, int Hp,
+             int Wp, int pad, uintptr_t stream) {
+              launch_stem_pad_x((const void*)x, (void*)xp, N, H, W, Hp, Wp,
+                                pad, as_stream(stream));
+              check_last();
+          });
+
+    m.def("stem_pad_w",
+          [](uintptr_t w, uintptr_t wp, int K, int R, int S,
+             uintptr_t stream) {
+              launch_stem_pad_w((const void*)w, (void*)wp, K, R, S,
+                                as_stream(stream));
+              check_last();
+          });
+
+    m.def("stem_unpad_dw",
+          [](uintptr_t dwp, uintptr_t dw, int K, int R, int S,
+             uintptr_t stream) {
+              launch_stem_unpad_dw((const void*)dwp, (void*)dw, K, R, S,
+                                   as_stream(stream));
+              check_last();
+          });
+
     m.def("conv8_eligible",
           [](int N, int H, int W, int C, int K, int R, int S, int Ho, int Wo,
              int stride, int pad, bool dgrad) {
