@@ -94,6 +94,32 @@ def measure_checkpoint(model, optim, folder: str):
     return save_s, restore_s
 
 
+def measure_large_checkpoint(folder: str, gib: float = 2.0):
+    """Multi-GB checkpoint round-trip (the 288 GB-HBM sizing direction of
+    BASELINE's north star): GB/s for the pinned-staged save and the CPU
+    restore, on a synthetic CUDA state.  The SECOND save reuses the pinned
+    pool (zero pinned allocation), which is the steady-state number."""
+    if not torch.cuda.is_available():
+        return None
+    n = int(gib * (1 << 30) // 4 // 16)
+    state = {f"t{i}": torch.randn(n, device="cuda") for i in range(16)}
+    path = os.path.join(folder, "bench_large_ckpt.th")
+    torch.cuda.synchronize()
+    out = {}
+    for label in ("cold", "warm"):   # warm = pinned pool reused
+        t0 = time.perf_counter()
+        fckpt.save_state(state, path)
+        out[f"save_{label}_s"] = round(time.perf_counter() - t0, 3)
+    t0 = time.perf_counter()
+    loaded = fckpt.load_state(path)
+    out["restore_s"] = round(time.perf_counter() - t0, 3)
+    del loaded
+    out["gib"] = round(n * 16 * 4 / (1 << 30), 2)
+    out["save_warm_gbps"] = round(out["gib"] / out["save_warm_s"], 2)
+    os.unlink(path)
+    return out
+
+
 def main():
     parser = argparse.ArgumentParser("flashy_amd bench")
     parser.add_argument("--gpus", type=int, default=1)
@@ -247,9 +273,11 @@ def main():
         torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
     elapsed = float(t.item())
 
-    ckpt_save_s = ckpt_restore_s = None
+    ckpt_save_s = ckpt_restore_s = large_ckpt = None
     if not args.no_ckpt and rank == 0:
         ckpt_save_s, ckpt_restore_s = measure_checkpoint(model, optim, ".")
+        if ws == 1:
+            large_ckpt = measure_large_checkpoint(".")
 
     if rank == 0:
         total_imgs = ws * args.batch * args.steps
@@ -279,6 +307,7 @@ def main():
                 "channels_last": args.channels_last,
                 "checkpoint_save_s": ckpt_save_s,
                 "checkpoint_restore_s": ckpt_restore_s,
+                "large_checkpoint": large_ckpt,
             },
         }
         print(json.dumps(result))

@@ -33,7 +33,11 @@ k_conv_dgrad8(const uint16_t* __restrict__ dout,
     constexpr int A_ELEMS = BM * BK;
     constexpr int B_ELEMS = BN * BK;
     constexpr int BUF_ELEMS = A_ELEMS + B_ELEMS;
-    constexpr int G = 4 + BN / 64;
+    constexpr int G = 4 + BN / 64;       // glds per wave per stage (A + B)
+    // BN=64 buffers fit 4x in the 160 KiB LDS -> prefetch DEPTH 2 (two
+    // stages' DMA in flight across barriers); BN=128 fits 3 -> depth 1
+    constexpr int BUFS = BN == 64 ? 4 : 3;
+    constexpr int DEPTH = BUFS - 2;
 
     const int rsk = d.R * d.S * d.K;
     const int64_t M = SCAT2 ? (int64_t)d.N * d.Ho * d.Wo
@@ -50,7 +54,7 @@ k_conv_dgrad8(const uint16_t* __restrict__ dout,
     const int64_t m0 = (int64_t)bx * BM;
     const int col0 = blockIdx.y * BN;
 
-    __shared__ uint16_t lds[3 * BUF_ELEMS];
+    __shared__ uint16_t lds[BUFS * BUF_ELEMS];
 
     const auto arsrc = __builtin_amdgcn_make_buffer_rsrc(
         (void*)dout, 0, dout_nbytes, 0x00020000);
@@ -192,16 +196,22 @@ k_conv_dgrad8(const uint16_t* __restrict__ dout,
         }
     };
 
-    issue_stage(0);
-    for (int i = 0; i + 1 < n_stages; ++i) {
-        issue_stage((i + 1) % 3);
+    for (int p = 0; p < DEPTH && p < n_stages; ++p)
+        issue_stage(p % BUFS);
+    for (int i = 0; i + DEPTH < n_stages; ++i) {
+        issue_stage((i + DEPTH) % BUFS);
+        asm volatile("s_waitcnt vmcnt(%0)" ::"i"(DEPTH * G) : "memory");
+        __builtin_amdgcn_s_barrier();
+        compute_stage(i % BUFS);
+    }
+    if (DEPTH == 2 && n_stages >= 2) {   // tail with one stage in flight
         asm volatile("s_waitcnt vmcnt(%0)" ::"i"(G) : "memory");
         __builtin_amdgcn_s_barrier();
-        compute_stage(i % 3);
+        compute_stage((n_stages - 2) % BUFS);
     }
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
-    compute_stage((n_stages - 1) % 3);
+    compute_stage((n_stages - 1) % BUFS);
 
     const int64_t out_row0 = m0 + wave_m * 128 + (lane & 15);
     const int out_col0 = col0 + wave_n * (BN / 4) + (lane >> 4) * 4;

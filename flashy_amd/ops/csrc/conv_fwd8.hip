@@ -42,6 +42,10 @@ k_conv_fwd8(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
     constexpr int B_ELEMS = BN * BK;     // 4096/8192 bf16 = 8/16 KiB
     constexpr int BUF_ELEMS = A_ELEMS + B_ELEMS;
     constexpr int G = 4 + BN / 64;       // glds per wave per stage (A + B)
+    // BN=64 buffers fit 4x in the 160 KiB LDS -> prefetch DEPTH 2 (two
+    // stages' DMA in flight across barriers); BN=128 fits 3 -> depth 1
+    constexpr int BUFS = BN == 64 ? 4 : 3;
+    constexpr int DEPTH = BUFS - 2;
 
     const int rsc = d.R * d.S * d.C;
     const int64_t M = (int64_t)d.N * d.Ho * d.Wo;
@@ -57,7 +61,7 @@ k_conv_fwd8(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
     const int64_t m0 = (int64_t)bx * BM;
     const int col0 = blockIdx.y * BN;
 
-    __shared__ uint16_t lds[3 * BUF_ELEMS];
+    __shared__ uint16_t lds[BUFS * BUF_ELEMS];
 
     const auto xrsrc = __builtin_amdgcn_make_buffer_rsrc(
         (void*)x, 0 /*stride*/, x_nbytes, 0x00020000 /*flags*/);
@@ -184,18 +188,24 @@ k_conv_fwd8(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
         // laggards read b[i%3] — never the same buffer.
     };
 
-    issue_stage(0, 0);
-    for (int i = 0; i + 1 < n_stages; ++i) {
-        issue_stage(i + 1, (i + 1) % 3);
-        // stage i landed; own G loads of stage i+1 stay in flight across
-        // the barrier (counted wait — never vmcnt(0) in the hot loop)
+    for (int p = 0; p < DEPTH && p < n_stages; ++p)
+        issue_stage(p, p % BUFS);
+    for (int i = 0; i + DEPTH < n_stages; ++i) {
+        issue_stage(i + DEPTH, (i + DEPTH) % BUFS);
+        // stage i landed; own DEPTH*G loads stay in flight across the
+        // barrier (counted wait — never vmcnt(0) in the hot loop)
+        asm volatile("s_waitcnt vmcnt(%0)" ::"i"(DEPTH * G) : "memory");
+        __builtin_amdgcn_s_barrier();
+        compute_stage(i % BUFS);
+    }
+    if (DEPTH == 2 && n_stages >= 2) {   // tail with one stage in flight
         asm volatile("s_waitcnt vmcnt(%0)" ::"i"(G) : "memory");
         __builtin_amdgcn_s_barrier();
-        compute_stage(i % 3);
+        compute_stage((n_stages - 2) % BUFS);
     }
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
-    compute_stage((n_stages - 1) % 3);
+    compute_stage((n_stages - 1) % BUFS);
 
     // --- epilogue ---------------------------------------------------------
     // Swapped-operand D layout: lane l of acc[mf][nf] holds pixel row
