@@ -277,6 +277,249 @@ k_conv_fwd8(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
     }
 }
 
+// --- persistent-B m-loop 1x1 kernel ----------------------------------------
+// For single-stage 1x1 convs (C == 64, stride 1): the generic kernel's
+// whole life is ONE stage, so every block pays a full un-pipelined
+// HBM->LDS->MFMA round trip (r1_1x1b measured 116 TF).  Here each block
+// keeps the weight tile resident in LDS and loops over MANY 256-row
+// m-tiles, pipelining the A DMA two tiles ahead — the round-trip latency
+// amortizes across the whole m walk.  BN-stats partials are written
+// per-(tile, wave_m) slice (msplit = 2 * mtiles), so the epilogue needs no
+// LDS and no barrier that would drain the in-flight DMA.
+
+template <int BN, bool RELU>
+__global__ void __launch_bounds__(512, 2)
+k_conv1x1_mloop(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
+                uint16_t* __restrict__ y, float* __restrict__ bn_ws,
+                ConvDims d, unsigned x_nbytes, int mtiles) {
+    constexpr int BK = 64;               // = C
+    constexpr int NF = BN / 64;
+    constexpr int MF = 8;
+    constexpr int A_ELEMS = 256 * BK;
+    constexpr int B_ELEMS = BN * BK;
+
+    const int64_t M = (int64_t)d.N * d.Ho * d.Wo;
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid_u = __builtin_amdgcn_readfirstlane(tid >> 6);
+    const int wave_m = wid_u >> 2;
+    const int wave_n = wid_u & 3;
+
+    unsigned bx = blockIdx.x;
+    if ((gridDim.x & 7) == 0)
+        bx = (bx & 7) * (gridDim.x >> 3) + (bx >> 3);
+    const int col0 = blockIdx.y * BN;
+    const int gT = gridDim.x;
+
+    __shared__ uint16_t lds[3 * A_ELEMS + B_ELEMS];
+    uint16_t* const bbase = lds + 3 * A_ELEMS;
+
+    const auto xrsrc = __builtin_amdgcn_make_buffer_rsrc(
+        (void*)x, 0, x_nbytes, 0x00020000);
+
+    // B once: dest chunk t -> source chunk (col, kc ^ (col&7))
+#pragma unroll
+    for (int g = 0; g < NF; ++g) {
+        const int chunk = g * 512 + tid;
+        const int col = chunk >> 3;
+        const int kc_s = (chunk & 7) ^ (col & 7);
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int*)
+                (w + (int64_t)(col0 + col) * BK + kc_s * 8),
+            (__attribute__((address_space(3))) unsigned int*)
+                (bbase + (g * 512 + wid_u * 64) * 8),
+            16, 0, 0);
+    }
+
+    // per-thread A chunk geometry (fixed): row + swizzled k-chunk
+    int a_row[4], a_koff[4];
+#pragma unroll
+    for (int g = 0; g < 4; ++g) {
+        const int chunk = g * 512 + tid;
+        const int row = chunk >> 3;
+        a_row[g] = row;
+        a_koff[g] = ((chunk & 7) ^ (row & 7)) * 8;
+    }
+    auto issue_a = [&](int64_t tile, int slot) {
+        uint16_t* base = lds + slot * A_ELEMS;
+#pragma unroll
+        for (int g = 0; g < 4; ++g) {
+            const int64_t m = tile * 256 + a_row[g];
+            const unsigned voff = m < M
+                ? (unsigned)((m * BK + a_koff[g]) * 2) : OOB_SENTINEL;
+            __builtin_amdgcn_raw_ptr_buffer_load_lds(
+                xrsrc,
+                (__attribute__((address_space(3))) void*)
+                    (base + (g * 512 + wid_u * 64) * 8),
+                16, voff, 0, 0, 0);
+        }
+    };
+
+    const int a_row_l = wave_m * 128 + (lane & 15);
+    const int frag_kb = (lane >> 4) * 16;
+    const int b_col_l = wave_n * (BN / 4) + (lane & 15);
+
+    short8 bfrag[2][NF];
+#pragma unroll
+    for (int sub = 0; sub < 2; ++sub)
+#pragma unroll
+        for (int nf = 0; nf < NF; ++nf) {
+            const int col = b_col_l + nf * 16;
+            const int byte = (col * 128 + sub * 64 + frag_kb) ^
+                             ((col & 7) << 4);
+            // read after the first barrier below (placed into registers
+            // lazily by the compiler behind its own lgkm counting)
+            bfrag[sub][nf] = *reinterpret_cast<const short8*>(
+                (const char*)bbase + byte);
+        }
+
+    const int64_t ntl = (mtiles - (int64_t)bx + gT - 1) / gT;  // my tiles
+    if (ntl <= 0) return;
+    issue_a(bx, 0);
+    if (ntl > 1) issue_a(bx + gT, 1);
+
+    for (int64_t k = 0; k < ntl; ++k) {
+        const int64_t tile = bx + k * gT;
+        if (k + 2 < ntl) {
+            issue_a(tile + 2 * gT, (int)((k + 2) % 3));
+            asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+        } else if (k + 1 < ntl) {
+            asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        } else {
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        }
+        __builtin_amdgcn_s_barrier();
+        const uint16_t* base = lds + (k % 3) * A_ELEMS;
+        floatx4 acc[MF][NF] = {};
+#pragma unroll
+        for (int sub = 0; sub < 2; ++sub) {
+            short8 a[MF];
+#pragma unroll
+            for (int mf = 0; mf < MF; ++mf) {
+                const int row = a_row_l + mf * 16;
+                const int byte = (row * 128 + sub * 64 + frag_kb) ^
+                                 ((row & 7) << 4);
+                a[mf] = *reinterpret_cast<const short8*>(
+                    (const char*)base + byte);
+            }
+            __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+            for (int mf = 0; mf < MF; ++mf)
+#pragma unroll
+                for (int nf = 0; nf < NF; ++nf)
+                    acc[mf][nf] =
+                        MFMA_BF16(bfrag[sub][nf], a[mf], acc[mf][nf]);
+            __builtin_amdgcn_s_setprio(0);
+        }
+        // epilogue for this tile (packed 8 B stores, swapped layout)
+        const int64_t out_row0 = tile * 256 + wave_m * 128 + (lane & 15);
+        const int out_col0 = col0 + wave_n * (BN / 4) + (lane >> 4) * 4;
+#pragma unroll
+        for (int mf = 0; mf < MF; ++mf) {
+            const int64_t row = out_row0 + mf * 16;
+            if (row < M) {
+#pragma unroll
+                for (int nf = 0; nf < NF; ++nf) {
+                    ushort4 pk;
+#pragma unroll
+                    for (int rr = 0; rr < 4; ++rr) {
+                        float v = acc[mf][nf][rr];
+                        if (RELU) v = fmaxf(v, 0.f);
+                        ((uint16_t*)&pk)[rr] = f32_to_bf16(v);
+                    }
+                    *reinterpret_cast<ushort4*>(
+                        y + row * d.K + out_col0 + nf * 16) = pk;
+                }
+            }
+        }
+        if (bn_ws != nullptr) {   // per-(tile, wave_m) slice: LDS-free
+            const int64_t msplit = 2 * (int64_t)mtiles;
+            const int64_t slice = tile * 2 + wave_m;
+#pragma unroll
+            for (int nf = 0; nf < NF; ++nf) {
+                float s[4] = {}, s2[4] = {};
+#pragma unroll
+                for (int mf = 0; mf < MF; ++mf) {
+                    const int64_t row = out_row0 + mf * 16;
+#pragma unroll
+                    for (int rr = 0; rr < 4; ++rr) {
+                        float v = acc[mf][nf][rr];
+                        if (RELU) v = fmaxf(v, 0.f);
+                        if (row >= M) v = 0.f;
+                        s[rr] += v;
+                        s2[rr] = fmaf(v, v, s2[rr]);
+                    }
+                }
+#pragma unroll
+                for (int off = 1; off < 16; off <<= 1)
+#pragma unroll
+                    for (int rr = 0; rr < 4; ++rr) {
+                        s[rr] += __shfl_xor(s[rr], off, 64);
+                        s2[rr] += __shfl_xor(s2[rr], off, 64);
+                    }
+                if ((lane & 15) == 0) {
+                    const int c = col0 + wave_n * (BN / 4) + nf * 16 +
+                                  (lane >> 4) * 4;
+#pragma unroll
+                    for (int rr = 0; rr < 4; ++rr) {
+                        bn_ws[(int64_t)(c + rr) * msplit + slice] = s[rr];
+                        bn_ws[((int64_t)d.K + c + rr) * msplit + slice] =
+                            s2[rr];
+                    }
+                }
+            }
+        }
+    }
+}
+
+extern "C" int conv1x1_mloop_plan(ConvDims d, int* bn_out, int* gridx_out) {
+    static int disabled = [] {
+        const char* e = getenv("FLASHY_NO_FWD8");
+        return e && e[0] == '1';
+    }();
+    if (disabled) return 0;
+    if (d.R != 1 || d.S != 1 || d.stride != 1 || d.pad != 0) return 0;
+    if (d.C != 64 || d.K % 64) return 0;
+    const int64_t M = (int64_t)d.N * d.Ho * d.Wo;
+    const int64_t x_elems = (int64_t)d.N * d.H * d.W * d.C;
+    if (x_elems * 2 >= (int64_t)OOB_SENTINEL) return 0;
+    const int mtiles = (int)((M + 255) / 256);
+    if (mtiles < 128) return 0;
+    const int bn = d.K % 128 == 0 ? 128 : 64;
+    *bn_out = bn;
+    int gx = 256 / (d.K / bn);
+    if (gx < 8) gx = 8;
+    if (gx > mtiles) gx = mtiles;
+    *gridx_out = gx;
+    return mtiles;
+}
+
+extern "C" void launch_conv1x1_mloop(const void* x, const void* w, void* y,
+                                     ConvDims d, int relu, void* bn_ws,
+                                     int bn, int gridx, int mtiles,
+                                     hipStream_t stream) {
+    dim3 grid((unsigned)gridx, (unsigned)(d.K / bn));
+    const unsigned xb = (unsigned)((int64_t)d.N * d.H * d.W * d.C * 2);
+    auto xx = (const uint16_t*)x;
+    auto ww = (const uint16_t*)w;
+    auto yy = (uint16_t*)y;
+    if (bn == 128) {
+        if (relu)
+            k_conv1x1_mloop<128, true><<<grid, 512, 0, stream>>>(
+                xx, ww, yy, (float*)bn_ws, d, xb, mtiles);
+        else
+            k_conv1x1_mloop<128, false><<<grid, 512, 0, stream>>>(
+                xx, ww, yy, (float*)bn_ws, d, xb, mtiles);
+    } else {
+        if (relu)
+            k_conv1x1_mloop<64, true><<<grid, 512, 0, stream>>>(
+                xx, ww, yy, (float*)bn_ws, d, xb, mtiles);
+        else
+            k_conv1x1_mloop<64, false><<<grid, 512, 0, stream>>>(
+                xx, ww, yy, (float*)bn_ws, d, xb, mtiles);
+    }
+}
+
 // --- stem padding helpers ---------------------------------------------------
 // The C=3 7x7 stem is run through the 8-wave kernel by padding to C'=4,
 // R'=S'=8 (taps beyond 7x7 carry zero weights; borders are zero pixels), so

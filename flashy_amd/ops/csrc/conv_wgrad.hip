@@ -379,8 +379,14 @@ extern "C" void launch_conv_wgrad(const void* x, const void* dout, void* dw,
     // 128x128 tile only where re-read traffic dominates (R*S > 1): the 1x1
     // wgrads already run at the HBM roof on the 64x64 kernel's 4-blocks/CU
     // occupancy, and the 1-block/CU 8-wave kernel loses TLP there (measured:
-    // 1x1 196->156 TF, 3x3 186->244 TF).
-    if (d.R * d.S > 1 && d.K % 128 == 0 && rsc % 128 == 0) {
+    // 1x1 196->156 TF, 3x3 186->244 TF).  Within 3x3, the measured win/loss
+    // split over the ResNet-50/224 + CIFAR shapes: big M always wins, small
+    // M only with enough (K,rsc) tiles to fill the chip.
+    const int tiles8 = (d.K / 128) * (rsc / 128);
+    const bool wg8 = d.R * d.S > 1 && d.K % 128 == 0 && rsc % 128 == 0 &&
+                     (M >= 32768 || (M >= 8192 && tiles8 >= 18) ||
+                      tiles8 >= 100);
+    if (wg8) {
         const int tiles = (d.K / 128) * (rsc / 128);
         int ns = n_splits;
         if (ns > 1) {   // re-target for 512-thread blocks (one per CU)
