@@ -306,11 +306,18 @@ extern "C" int conv_fwd8_plan(ConvDims d, int* bn_out);
 extern "C" void launch_conv_fwd8(const void* x, const void* w, void* y,
                                  ConvDims d, int relu, void* bn_ws, int bn,
                                  int mtiles, hipStream_t stream);
+extern "C" int conv1x1_mloop_plan(ConvDims d, int* bn_out, int* gridx_out);
+extern "C" void launch_conv1x1_mloop(const void* x, const void* w, void* y,
+                                     ConvDims d, int relu, void* bn_ws,
+                                     int bn, int gridx, int mtiles,
+                                     hipStream_t stream);
 
 // grid.x the fwd launcher will use for these dims (= the msplit of the
 // fused BN-stats partials); Python sizes the partials buffer with this.
 extern "C" int conv_fwd_msplit(ConvDims d) {
-    int bn8;
+    int bn8, gx;
+    const int mtl = conv1x1_mloop_plan(d, &bn8, &gx);
+    if (mtl) return 2 * mtl;   // per-(tile, wave_m) slices
     const int mt8 = conv_fwd8_plan(d, &bn8);
     if (mt8) return mt8;
     const int64_t M = (int64_t)d.N * d.Ho * d.Wo;
@@ -322,7 +329,12 @@ extern "C" int conv_fwd_msplit(ConvDims d) {
 extern "C" void launch_conv_fwd(const void* x, const void* w, void* y,
                                 ConvDims d, int relu, void* bn_ws,
                                 hipStream_t stream) {
-    int bn8;
+    int bn8, gx;
+    const int mtl = conv1x1_mloop_plan(d, &bn8, &gx);
+    if (mtl) {
+        launch_conv1x1_mloop(x, w, y, d, relu, bn_ws, bn8, gx, mtl, stream);
+        return;
+    }
     const int mt8 = conv_fwd8_plan(d, &bn8);
     if (mt8) {
         launch_conv_fwd8(x, w, y, d, relu, bn_ws, bn8, mt8, stream);

@@ -360,19 +360,6 @@ k_conv1x1_mloop(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
     const int b_col_l = wave_n * (BN / 4) + (lane & 15);
 
     short8 bfrag[2][NF];
-#pragma unroll
-    for (int sub = 0; sub < 2; ++sub)
-#pragma unroll
-        for (int nf = 0; nf < NF; ++nf) {
-            const int col = b_col_l + nf * 16;
-            const int byte = (col * 128 + sub * 64 + frag_kb) ^
-                             ((col & 7) << 4);
-            // read after the first barrier below (placed into registers
-            // lazily by the compiler behind its own lgkm counting)
-            bfrag[sub][nf] = *reinterpret_cast<const short8*>(
-                (const char*)bbase + byte);
-        }
-
     const int64_t ntl = (mtiles - (int64_t)bx + gT - 1) / gT;  // my tiles
     if (ntl <= 0) return;
     issue_a(bx, 0);
@@ -389,6 +376,18 @@ k_conv1x1_mloop(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
             asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         }
         __builtin_amdgcn_s_barrier();
+        if (k == 0) {   // weights landed (oldest DMA): load the B fragments
+#pragma unroll
+            for (int sub = 0; sub < 2; ++sub)
+#pragma unroll
+                for (int nf = 0; nf < NF; ++nf) {
+                    const int col = b_col_l + nf * 16;
+                    const int byte = (col * 128 + sub * 64 + frag_kb) ^
+                                     ((col & 7) << 4);
+                    bfrag[sub][nf] = *reinterpret_cast<const short8*>(
+                        (const char*)bbase + byte);
+                }
+        }
         const uint16_t* base = lds + (k % 3) * A_ELEMS;
         floatx4 acc[MF][NF] = {};
 #pragma unroll
