@@ -215,6 +215,11 @@ extern "C" int conv_dgrad8_s2_plan(ConvDims d, int* bn_out);
 extern "C" void launch_conv_dgrad8_s2(const void* dout, const void* w_rsck,
                                       void* dx, ConvDims d, int bn,
                                       int mtiles, hipStream_t stream);
+extern "C" int conv1x1_mloop_plan(ConvDims d, int* bn_out, int* gridx_out);
+extern "C" void launch_conv1x1_mloop(const void* x, const void* w, void* y,
+                                     ConvDims d, int relu, void* bn_ws,
+                                     int bn, int gridx, int mtiles,
+                                     hipStream_t stream);
 
 extern "C" void launch_conv_dgrad(const void* dout, const void* w_rsck,
                                   void* dx, ConvDims d, hipStream_t stream) {
@@ -227,6 +232,21 @@ extern "C" void launch_conv_dgrad(const void* dout, const void* w_rsck,
         const int mt8 = conv_dgrad8_s2_plan(d, &bn8);
         if (mt8) {
             launch_conv_dgrad8_s2(dout, w_rsck, dx, d, bn8, mt8, stream);
+            return;
+        }
+    }
+    if (d.R == 1 && d.S == 1 && d.stride == 1 && d.pad == 0 && d.K == 64) {
+        // single-stage 1x1 dgrad = the fwd m-loop GEMM with remapped dims:
+        // A = dy [M][64], B = w_rsck [C][64] (same [col][red] addressing
+        // as the fwd weight), out = dx [M][C]
+        ConvDims dd = d;
+        dd.C = 64;
+        dd.K = d.C;
+        int bn8, gx;
+        const int mtl = conv1x1_mloop_plan(dd, &bn8, &gx);
+        if (mtl) {
+            launch_conv1x1_mloop(dout, w_rsck, dx, dd, 0, nullptr, bn8, gx,
+                                 mtl, stream);
             return;
         }
     }

@@ -12,6 +12,9 @@ SHAPES = {
     "l1conv": (32, 32, 64, 64, 3, 1),
     "l2conv": (16, 16, 128, 128, 3, 1),
     "l4conv": (4, 4, 512, 512, 3, 1),
+    "r2_3x3": (28, 28, 128, 128, 3, 1),
+    "r1_3x3": (56, 56, 64, 64, 3, 1),
+    "r1_1x1b": (56, 56, 64, 256, 1, 1),
 }
 H, W, C, K, R, stride = SHAPES[name]
 N, pad = 64, R // 2
