@@ -25,14 +25,6 @@ struct ConvDims {
 
 #define MFMA_BF16(a, b, c) __builtin_amdgcn_mfma_f32_16x16x32_bf16((a), (b), (c), 0, 0, 0)
 
-// v_mfma_f32_32x32x16_bf16: same A/B lane maps with 32-row/col tiles
-// (A: lane l row=l&31, k=(l>>5)*8+j); D: lane l holds D[(reg&3)+8*(reg>>2)
-// +4*(l>>5)][l&31] over 16 f32 regs.  4x the flops per instruction of the
-// 16x16 form at ~2x the issue cost — used by the BN=64 compute paths where
-// the 16x16 form is LDS-read-bound (NF=1).
-typedef float floatx16 __attribute__((ext_vector_type(16)));
-#define MFMA_BF16_32(a, b, c) __builtin_amdgcn_mfma_f32_32x32x16_bf16((a), (b), (c), 0, 0, 0)
-
 // Block geometry shared by conv fwd and dgrad kernels.
 #define CONV_BM 128      // GEMM rows (output pixels) per block
 #define CONV_BN 64       // GEMM cols (channels) per block

@@ -158,39 +158,13 @@ k_conv_dgrad8(const uint16_t* __restrict__ dout,
     const int a_row_l = wave_m * 128 + (lane & 15);
     const int frag_kb = (lane >> 4) * 16;
     const int b_col_l = wave_n * (BN / 4) + (lane & 15);
-    const int wave_m32 = wid_u >> 1;   // MFMA32 decomposition (BN==64)
-    const int wave_n32 = wid_u & 1;
 
     floatx4 acc[MF][NF] = {};
-    floatx16 acc32[2] = {};
     const int n_stages = rsk / BK;
 
     auto compute_stage = [&](int buf) {
         const uint16_t* base = lds + buf * BUF_ELEMS;
         const uint16_t* bbase = base + A_ELEMS;
-        if constexpr (BN == 64) {   // see conv_fwd8.hip MFMA32 notes
-#pragma unroll
-            for (int kc = 0; kc < 4; ++kc) {
-                short8 a[2], b;
-                const int kb = kc * 32 + (lane >> 5) * 16;
-#pragma unroll
-                for (int mt = 0; mt < 2; ++mt) {
-                    const int row = wave_m32 * 64 + mt * 32 + (lane & 31);
-                    const int byte = (row * 128 + kb) ^ ((row & 7) << 4);
-                    a[mt] = *reinterpret_cast<const short8*>(
-                        (const char*)base + byte);
-                }
-                const int col = wave_n32 * 32 + (lane & 31);
-                const int bbyte = (col * 128 + kb) ^ ((col & 7) << 4);
-                b = *reinterpret_cast<const short8*>(
-                    (const char*)bbase + bbyte);
-                __builtin_amdgcn_s_setprio(1);
-                acc32[0] = MFMA_BF16_32(b, a[0], acc32[0]);
-                acc32[1] = MFMA_BF16_32(b, a[1], acc32[1]);
-                __builtin_amdgcn_s_setprio(0);
-            }
-            return;
-        }
 #pragma unroll
         for (int sub = 0; sub < 2; ++sub) {
             short8 a[MF], b[NF];
@@ -239,45 +213,6 @@ k_conv_dgrad8(const uint16_t* __restrict__ dout,
     __builtin_amdgcn_s_barrier();
     compute_stage((n_stages - 1) % BUFS);
 
-    if constexpr (BN == 64) {   // MFMA32 epilogue (packed 8 B stores)
-        const int64_t prow0 = m0 + wave_m32 * 64 + (lane & 31);
-        const int col0w = col0 + wave_n32 * 32 + 4 * (int)(lane >> 5);
-#pragma unroll
-        for (int mt = 0; mt < 2; ++mt) {
-            const int64_t row = prow0 + mt * 32;
-            if (row < M) {
-                int64_t obase;
-                if (SCAT2) {
-                    const int wo = (int)(row % d.Wo);
-                    const int ho = (int)((row / d.Wo) % d.Ho);
-                    const int64_t n = row / ((int64_t)d.Ho * d.Wo);
-                    obase = ((n * d.H + 2 * ho) * d.W + 2 * wo) *
-                            (int64_t)d.C;
-                } else {
-                    obase = row * d.C;
-                }
-#pragma unroll
-                for (int q = 0; q < 4; ++q) {
-                    ushort4 pk;
-#pragma unroll
-                    for (int rr = 0; rr < 4; ++rr)
-                        ((uint16_t*)&pk)[rr] =
-                            f32_to_bf16(acc32[mt][q * 4 + rr]);
-                    const int64_t o = obase + col0w + q * 8;
-                    *reinterpret_cast<ushort4*>(dx + o) = pk;
-                    if (SCAT2) {
-                        const ushort4 z = {};
-                        *reinterpret_cast<ushort4*>(dx + o + d.C) = z;
-                        *reinterpret_cast<ushort4*>(
-                            dx + o + (int64_t)d.W * d.C) = z;
-                        *reinterpret_cast<ushort4*>(
-                            dx + o + (int64_t)d.W * d.C + d.C) = z;
-                    }
-                }
-            }
-        }
-        return;
-    }
     const int64_t out_row0 = m0 + wave_m * 128 + (lane & 15);
     const int out_col0 = col0 + wave_n * (BN / 4) + (lane >> 4) * 4;
 #pragma unroll

@@ -143,42 +143,13 @@ k_conv_fwd8(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
     const int a_row_l = wave_m * 128 + (lane & 15);   // + mf*16
     const int frag_kb = (lane >> 4) * 16;             // byte within 64B half
     const int b_col_l = wave_n * (BN / 4) + (lane & 15);  // + nf*16
-    // MFMA32 decomposition (BN==64 only): 4 M-waves x 2 N-waves of 32x32
-    // tiles — with NF=1 the 16x16 form is LDS-read-bound (9 b128 reads per
-    // 8 MFMA); 32x32x16 halves reads per flop and quarters MFMA issue.
-    const int wave_m32 = wid_u >> 1;   // 0..3
-    const int wave_n32 = wid_u & 1;    // 0..1
 
     floatx4 acc[MF][NF] = {};
-    floatx16 acc32[2] = {};
     const int n_stages = rsc / BK;
 
     auto compute_stage = [&](int buf) {
         const uint16_t* base = lds + buf * BUF_ELEMS;
         const uint16_t* bbase = base + A_ELEMS;
-        if constexpr (BN == 64) {
-#pragma unroll
-            for (int kc = 0; kc < 4; ++kc) {   // 16-deep MFMA chunks
-                short8 a[2], b;
-                const int kb = kc * 32 + (lane >> 5) * 16;
-#pragma unroll
-                for (int mt = 0; mt < 2; ++mt) {
-                    const int row = wave_m32 * 64 + mt * 32 + (lane & 31);
-                    const int byte = (row * 128 + kb) ^ ((row & 7) << 4);
-                    a[mt] = *reinterpret_cast<const short8*>(
-                        (const char*)base + byte);
-                }
-                const int col = wave_n32 * 32 + (lane & 31);
-                const int bbyte = (col * 128 + kb) ^ ((col & 7) << 4);
-                b = *reinterpret_cast<const short8*>(
-                    (const char*)bbase + bbyte);
-                __builtin_amdgcn_s_setprio(1);
-                acc32[0] = MFMA_BF16_32(b, a[0], acc32[0]);
-                acc32[1] = MFMA_BF16_32(b, a[1], acc32[1]);
-                __builtin_amdgcn_s_setprio(0);
-            }
-            return;
-        }
 #pragma unroll
         for (int sub = 0; sub < 2; ++sub) {
             short8 a[MF], b[NF];
@@ -237,78 +208,6 @@ k_conv_fwd8(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
     compute_stage((n_stages - 1) % BUFS);
 
     // --- epilogue ---------------------------------------------------------
-    if constexpr (BN == 64) {
-        // MFMA32 D layout: lane l of acc32[mt] holds pixel (mt*32 + (l&31))
-        // at channels (q*8 + 4*(l>>5) + rr) for reg quads q = reg>>2.
-        const int64_t prow0 = m0 + wave_m32 * 64 + (lane & 31);
-        const int col0w = col0 + wave_n32 * 32 + 4 * (int)(lane >> 5);
-#pragma unroll
-        for (int mt = 0; mt < 2; ++mt) {
-            const int64_t row = prow0 + mt * 32;
-            if (row < M) {
-#pragma unroll
-                for (int q = 0; q < 4; ++q) {
-                    ushort4 pk;
-#pragma unroll
-                    for (int rr = 0; rr < 4; ++rr) {
-                        float v = acc32[mt][q * 4 + rr];
-                        if (RELU) v = fmaxf(v, 0.f);
-                        ((uint16_t*)&pk)[rr] = f32_to_bf16(v);
-                    }
-                    *reinterpret_cast<ushort4*>(
-                        y + row * d.K + col0w + q * 8) = pk;
-                }
-            }
-        }
-        if (bn_ws != nullptr) {
-            __syncthreads();
-            float* sred = reinterpret_cast<float*>(lds);   // [4 wm][64] x2
-#pragma unroll
-            for (int q = 0; q < 4; ++q) {
-                float s[4] = {}, s2[4] = {};
-#pragma unroll
-                for (int mt = 0; mt < 2; ++mt) {
-                    const int64_t row = prow0 + mt * 32;
-#pragma unroll
-                    for (int rr = 0; rr < 4; ++rr) {
-                        float v = acc32[mt][q * 4 + rr];
-                        if (RELU) v = fmaxf(v, 0.f);
-                        if (row >= M) v = 0.f;
-                        s[rr] += v;
-                        s2[rr] = fmaf(v, v, s2[rr]);
-                    }
-                }
-#pragma unroll
-                for (int off = 1; off < 32; off <<= 1)
-#pragma unroll
-                    for (int rr = 0; rr < 4; ++rr) {
-                        s[rr] += __shfl_xor(s[rr], off, 64);
-                        s2[rr] += __shfl_xor(s2[rr], off, 64);
-                    }
-                if ((lane & 31) == 0) {
-                    const int ch = wave_n32 * 32 + q * 8 + 4 * (int)(lane >> 5);
-                    *reinterpret_cast<float4*>(&sred[wave_m32 * 64 + ch]) =
-                        make_float4(s[0], s[1], s[2], s[3]);
-                    *reinterpret_cast<float4*>(
-                        &sred[(4 + wave_m32) * 64 + ch]) =
-                        make_float4(s2[0], s2[1], s2[2], s2[3]);
-                }
-            }
-            __syncthreads();
-            if (tid < 64) {
-                float s = 0.f, s2 = 0.f;
-#pragma unroll
-                for (int wm = 0; wm < 4; ++wm) {
-                    s += sred[wm * 64 + tid];
-                    s2 += sred[(4 + wm) * 64 + tid];
-                }
-                const int c = col0 + tid;
-                bn_ws[(int64_t)c * gridDim.x + blockIdx.x] = s;
-                bn_ws[((int64_t)d.K + c) * gridDim.x + blockIdx.x] = s2;
-            }
-        }
-        return;
-    }
     // Swapped-operand D layout: lane l of acc[mf][nf] holds pixel row
     // (mf*16 + (l&15)) at output channels (nf*16 + (l>>4)*4 + rr) — four
     // CONSECUTIVE channels per lane, packed into one 8 B store.
