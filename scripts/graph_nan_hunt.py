@@ -47,6 +47,17 @@ def finite_report(model, opt):
 def trial(i: int) -> bool:
     torch.manual_seed(3)
     model = resnet18(num_classes=10, small_input=True).cuda()
+    if "no-fc" in FLAGS:   # isolate the autocast bf16 GEMM (hipBLASLt)
+        model.fc = torch.nn.Identity()
+    if "fp32-fc" in FLAGS:  # keep autocast but run the fc GEMM in fp32
+        fc = model.fc
+
+        class F32FC(torch.nn.Module):
+            def forward(self, x):
+                with torch.autocast("cuda", enabled=False):
+                    return fc(x.float())
+
+        model.fc = F32FC()
     if "torch-bn-eval" in FLAGS:
         for m in model.modules():
             if isinstance(m, torch.nn.BatchNorm2d):
@@ -63,6 +74,8 @@ def trial(i: int) -> bool:
         opt.zero_grad(set_to_none=False)
         with torch.autocast("cuda", torch.bfloat16, enabled=use_ac):
             logits = model(static_x)
+        if "no-fc" in FLAGS:
+            logits = logits[:, :10]   # pooled features stand in for logits
         if "torch-ce" in FLAGS:
             loss = torch.nn.functional.cross_entropy(logits.float(), static_y)
         else:
