@@ -105,7 +105,13 @@ def init(backend: tp.Optional[str] = None) -> None:
         backend = "nccl" if torch.cuda.is_available() else "gloo"
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     os.environ.setdefault("MASTER_PORT", "29500")
-    dist.init_process_group(backend=backend, init_method="env://")
+    # explicit watchdog: a hung collective (rank divergence, a capture gone
+    # wrong) must abort the job quickly, never wedge the node
+    import datetime
+    timeout = datetime.timedelta(
+        seconds=int(os.environ.get("FLASHY_AMD_PG_TIMEOUT", "180")))
+    dist.init_process_group(backend=backend, init_method="env://",
+                            timeout=timeout)
 
 
 def device() -> torch.device:
