@@ -211,6 +211,57 @@ def test_graph_captured_step():
 
 
 @requires_gpu
+def test_overlapped_sync_in_graph_ws1(monkeypatch):
+    """The 8-GPU SCALE path in miniature: a real RCCL communicator at
+    world_size=1, the flagship step captured as ONE graph with the chunked
+    flat all-reduces recorded in-graph (ws=1 collectives are identities,
+    so numerics must match a no-comm run exactly up to wgrad atomics)."""
+    import torch.distributed as dist
+    from flashy_amd import distrib
+    from flashy_amd.graph import CapturedStep
+    from flashy_amd.models import native_resnet18
+    from flashy_amd.optim import FusedSGD
+    from flashy_amd.functional import cross_entropy
+
+    monkeypatch.setenv("MASTER_ADDR", "127.0.0.1")
+    monkeypatch.setenv("MASTER_PORT", "29619")
+    monkeypatch.setenv("RANK", "0")
+    monkeypatch.setenv("WORLD_SIZE", "1")
+    created = not dist.is_initialized()
+    if created:
+        dist.init_process_group("nccl", init_method="env://")
+    monkeypatch.setattr(distrib, "is_distributed", lambda: True)
+    try:
+        torch.manual_seed(11)
+        model = native_resnet18(num_classes=10, imagenet_stem=False).cuda()
+        opt = FusedSGD(model.parameters(), lr=0.05, momentum=0.9,
+                       bf16_mirror=True)
+        model.enable_wt_cache()
+        sync = distrib.OverlappedFlatSync(opt, chunk_bytes=4 << 20)
+        assert sync.n_chunks >= 2
+        x = torch.randn(32, 3, 32, 32, device="cuda")
+        y = torch.randint(10, (32,), device="cuda")
+
+        def step():
+            opt.zero_grad(set_to_none=False)
+            loss = cross_entropy(model(x), y)
+            loss.backward()
+            sync.finish()
+            opt.step()
+            return loss
+
+        graphed = CapturedStep(step, warmup=2).capture()
+        losses = [float(graphed().item()) for _ in range(5)]
+        torch.cuda.synchronize()
+        assert all(torch.isfinite(torch.tensor(losses))), losses
+        assert losses[-1] < losses[0] + 0.5   # trains, no blow-up
+        sync.remove()
+    finally:
+        if created:
+            dist.destroy_process_group()
+
+
+@requires_gpu
 def test_solver_end_to_end_gpu(tmp_path, monkeypatch):
     """One epoch of the cifar example solver on GPU + restore round-trip."""
     monkeypatch.setenv("_FLASHY_AMD_DIR", str(tmp_path))
