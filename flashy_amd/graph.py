@@ -36,6 +36,7 @@ Notes:
 """
 from __future__ import annotations
 
+import os
 import typing as tp
 
 import torch
@@ -100,7 +101,17 @@ class CapturedStep:
     def capture(self) -> "CapturedStep":
         if not self.enabled:
             return self
-        _warm_capture_machinery()
+        # MIOpen's implicit-GEMM conv solvers mis-execute under hipGraph
+        # REPLAY on this stack (ROCm 7.0/7.2, bf16): a torch-module model
+        # captured with autocast replays NaN after 2-4 replays on ~50-70%
+        # of processes, entering at the smallest conv.  Solver-class bisect
+        # (scripts/graph_nan_hunt.py, 12-process matrix per class):
+        # base 8/12 NaN, MIOPEN_DEBUG_CONV_IMPLICIT_GEMM=0 -> 0/12,
+        # CONV_GEMM=0 (more shapes onto implicit-GEMM) -> 12/12.
+        # Only affects torch-module (MIOpen) models — the native NHWC
+        # kernels never touch MIOpen.  Set before the warmup below so the
+        # find cache never selects the broken class for captured shapes.
+        os.environ.setdefault("MIOPEN_DEBUG_CONV_IMPLICIT_GEMM", "0")
         # The autocast weight cache is incompatible with graph capture
         # (same rule as torch.cuda.make_graphed_callables): cached casts
         # allocated during capture are freed into the graph's private pool
