@@ -49,13 +49,14 @@ def _warm_capture_machinery() -> None:
     """Throwaway first capture (once per process): a tiny elementwise +
     GEMM + conv graph, captured and replayed, so lazy per-stream library
     state (BLAS workspaces, MIOpen descriptors, allocator graph-pool
-    machinery) initializes OUTSIDE the first real graph.
+    machinery) initializes OUTSIDE the first real graph's private pool.
 
-    Empirically load-bearing: the intermittent fresh-box failure where the
-    first captured training step replays NaN after 2 replays (GPUTEST_r01,
-    and reproduced solo in round 2: 2/2 failing runs when the training
-    step is the process's first capture vs 12/12 clean with any prior
-    capture in the process — scripts/graph_nan_hunt.py)."""
+    Defensive hygiene, not the NaN fix: the captured-step NaN chase
+    (scripts/graph_nan_hunt.py) initially pointed here, but the
+    per-process-stochastic culprit was MIOpen's implicit-GEMM solver class
+    under replay (see capture() below).  Kept because first-capture lazy
+    allocations landing in a training graph's pool remain a real hazard
+    class and the cost is one tiny capture per process."""
     global _capture_warmed
     if _capture_warmed or not torch.cuda.is_available():
         return
