@@ -145,6 +145,12 @@ def main():
                         help="gan = DCGAN-style G/D on 64x64 synthetic images"
                              " (BASELINE config 4)")
     args = parser.parse_args()
+    if args.ref:
+        # reference mode must represent stock torch-ROCm fairly: restore
+        # MIOpen's full solver choice (flashy_amd import disables the
+        # implicit-GEMM class because it mis-executes under graph REPLAY;
+        # --ref runs eager and never captures)
+        os.environ["MIOPEN_DEBUG_CONV_IMPLICIT_GEMM"] = "1"
     if args.workload == "gan":
         return main_gan(args)
 
