@@ -193,184 +193,6 @@ k_conv_wgrad(const uint16_t* __restrict__ x, const uint16_t* __restrict__ dout,
 }
 
 // ---------------------------------------------------------------------------
-// j-resident 64(K) x 192(rsc) wgrad: the dout stage is loaded ONCE per
-// 64-m stage and shared by THREE 64-wide rsc subtiles (dout re-read
-// rsc/192 times instead of rsc/64 — a 3x cut on the dominant stream for
-// the K=64 / rsc%128!=0 shapes the 128x128 kernel cannot take, e.g. the
-// CIFAR layer-1 family at rsc=576).  Same tr_b16 layout as above.
-// ---------------------------------------------------------------------------
-
-__global__ void __launch_bounds__(CONV_THREADS, 3)
-k_conv_wgrad_j3(const uint16_t* __restrict__ x,
-                const uint16_t* __restrict__ dout, float* __restrict__ dw,
-                ConvDims d, int m_per_split) {
-    const int rsc = d.R * d.S * d.C;
-    const int64_t M = (int64_t)d.N * d.Ho * d.Wo;
-    const int tid = threadIdx.x;
-    const int lane = tid & 63;
-    const int wid = tid >> 6;
-    const int wave_k = wid >> 1;
-    const int wave_j = wid & 1;
-    const int k0 = blockIdx.x * 64;
-    const int j0 = blockIdx.y * 192;
-    const int64_t ms = (int64_t)blockIdx.z * m_per_split;
-    const int64_t me = min(ms + (int64_t)m_per_split, M);
-
-    __shared__ __attribute__((aligned(16))) uint16_t doutT[2][2 * 32 * WG_P];
-    __shared__ __attribute__((aligned(16))) uint16_t xT[2][3][2 * 32 * WG_P];
-
-    floatx4 acc[3][2][2] = {};
-
-    const int m_r = tid >> 3;
-    const int k8 = (tid & 7) * 8;
-    int t_hoff[3], t_woff[3], t_c[3];
-#pragma unroll
-    for (int jt = 0; jt < 3; ++jt) {
-        const int jj = j0 + jt * 64 + k8;
-        const int r = jj / (d.S * d.C);
-        const int scc = jj - r * d.S * d.C;
-        const int s = scc / d.C;
-        t_c[jt] = scc - s * d.C;
-        t_hoff[jt] = r - d.pad;
-        t_woff[jt] = s - d.pad;
-    }
-    auto load_pair = [&](int64_t mc, short8* dv, short8 (*xv)[2]) {
-#pragma unroll
-        for (int sc = 0; sc < 2; ++sc) {
-            const int64_t m = mc + sc * CONV_BK + m_r;
-            short8 v = {};
-            if (m < me)
-                v = *reinterpret_cast<const short8*>(dout + m * d.K + k0 + k8);
-            dv[sc] = v;
-            int wo = 0, ho = 0;
-            int64_t n = 0;
-            if (m < me) {
-                wo = (int)(m % d.Wo);
-                ho = (int)((m / d.Wo) % d.Ho);
-                n = m / ((int64_t)d.Ho * d.Wo);
-            }
-#pragma unroll
-            for (int jt = 0; jt < 3; ++jt) {
-                short8 u = {};
-                if (m < me) {
-                    const int hi = ho * d.stride + t_hoff[jt];
-                    const int wi = wo * d.stride + t_woff[jt];
-                    if (hi >= 0 && hi < d.H && wi >= 0 && wi < d.W)
-                        u = *reinterpret_cast<const short8*>(
-                            x + (((n * d.H + hi) * d.W + wi) * (int64_t)d.C +
-                                 t_c[jt]));
-                }
-                xv[jt][sc] = u;
-            }
-        }
-    };
-    auto stage_write = [&](uint16_t (&dT)[2 * 32 * WG_P],
-                           uint16_t (&xTb)[3][2 * 32 * WG_P],
-                           const short8* dv, const short8 (*xv)[2]) {
-#pragma unroll
-        for (int sc = 0; sc < 2; ++sc) {
-            const int at = (sc * 32 + m_r) * WG_P + k8;
-            *reinterpret_cast<short8*>(&dT[at]) = dv[sc];
-#pragma unroll
-            for (int jt = 0; jt < 3; ++jt)
-                *reinterpret_cast<short8*>(&xTb[jt][at]) = xv[jt][sc];
-        }
-    };
-
-    const int tr_lane = ((lane & 15) >> 2) * WG_P + 4 * (lane & 3) +
-                        (lane >> 4) * 8 * WG_P;
-    union U64x2 { struct { unsigned long long lo, hi; } q; short8 v; };
-
-    const int64_t n_stages = (me - ms + 2 * CONV_BK - 1) / (2 * CONV_BK);
-    short8 dv[2], xv[3][2];
-    load_pair(ms, dv, xv);
-    stage_write(doutT[0], xT[0], dv, xv);
-    if (n_stages > 1) load_pair(ms + 2 * CONV_BK, dv, xv);
-    __syncthreads();
-
-    auto step = [&](int64_t i, const uint16_t (&dT)[2 * 32 * WG_P],
-                    const uint16_t (&xTb)[3][2 * 32 * WG_P],
-                    uint16_t (&ndT)[2 * 32 * WG_P],
-                    uint16_t (&nxT)[3][2 * 32 * WG_P]) {
-        if (i + 1 < n_stages) {
-            stage_write(ndT, nxT, dv, xv);
-            if (i + 2 < n_stages) load_pair(ms + (i + 2) * 2 * CONV_BK, dv, xv);
-        }
-#pragma unroll
-        for (int sc = 0; sc < 2; ++sc) {
-            const int base = sc * 32 * WG_P + tr_lane;
-            const unsigned a0 = (unsigned)(unsigned long long)(const void*)
-                &dT[base + wave_k * 32];
-            U64x2 af[2];
-            asm volatile(
-                "ds_read_b64_tr_b16 %0, %4\n\t"
-                "ds_read_b64_tr_b16 %1, %4 offset:576\n\t"
-                "ds_read_b64_tr_b16 %2, %4 offset:32\n\t"
-                "ds_read_b64_tr_b16 %3, %4 offset:608"
-                : "=&v"(af[0].q.lo), "=&v"(af[0].q.hi),
-                  "=&v"(af[1].q.lo), "=&v"(af[1].q.hi)
-                : "v"(a0));
-            U64x2 bf[3][2];
-            const unsigned b0 = (unsigned)(unsigned long long)(const void*)
-                &xTb[0][base + wave_j * 32];
-            constexpr unsigned JSTRIDE = 2 * 32 * WG_P * 2;  // bytes per jt
-            asm volatile(
-                "ds_read_b64_tr_b16 %0, %12\n\t"
-                "ds_read_b64_tr_b16 %1, %12 offset:576\n\t"
-                "ds_read_b64_tr_b16 %2, %12 offset:32\n\t"
-                "ds_read_b64_tr_b16 %3, %12 offset:608\n\t"
-                "ds_read_b64_tr_b16 %4, %13\n\t"
-                "ds_read_b64_tr_b16 %5, %13 offset:576\n\t"
-                "ds_read_b64_tr_b16 %6, %13 offset:32\n\t"
-                "ds_read_b64_tr_b16 %7, %13 offset:608\n\t"
-                "ds_read_b64_tr_b16 %8, %14\n\t"
-                "ds_read_b64_tr_b16 %9, %14 offset:576\n\t"
-                "ds_read_b64_tr_b16 %10, %14 offset:32\n\t"
-                "ds_read_b64_tr_b16 %11, %14 offset:608\n\t"
-                "s_waitcnt lgkmcnt(0)"
-                : "=&v"(bf[0][0].q.lo), "=&v"(bf[0][0].q.hi),
-                  "=&v"(bf[0][1].q.lo), "=&v"(bf[0][1].q.hi),
-                  "=&v"(bf[1][0].q.lo), "=&v"(bf[1][0].q.hi),
-                  "=&v"(bf[1][1].q.lo), "=&v"(bf[1][1].q.hi),
-                  "=&v"(bf[2][0].q.lo), "=&v"(bf[2][0].q.hi),
-                  "=&v"(bf[2][1].q.lo), "=&v"(bf[2][1].q.hi)
-                : "v"(b0), "v"(b0 + JSTRIDE), "v"(b0 + 2 * JSTRIDE));
-            __builtin_amdgcn_sched_barrier(0);
-#pragma unroll
-            for (int jt = 0; jt < 3; ++jt)
-#pragma unroll
-                for (int kf = 0; kf < 2; ++kf)
-#pragma unroll
-                    for (int jf = 0; jf < 2; ++jf)
-                        acc[jt][kf][jf] = MFMA_BF16(af[kf].v, bf[jt][jf].v,
-                                                    acc[jt][kf][jf]);
-        }
-        __syncthreads();
-    };
-    for (int64_t i = 0; i < n_stages;) {
-        step(i, doutT[0], xT[0], doutT[1], xT[1]);
-        if (++i >= n_stages) break;
-        step(i, doutT[1], xT[1], doutT[0], xT[0]);
-        ++i;
-    }
-
-    const int out_k0 = k0 + wave_k * 32 + (lane >> 4) * 4;
-    const int out_j0 = j0 + wave_j * 32 + (lane & 15);
-#pragma unroll
-    for (int jt = 0; jt < 3; ++jt)
-#pragma unroll
-        for (int kf = 0; kf < 2; ++kf)
-#pragma unroll
-            for (int jf = 0; jf < 2; ++jf)
-#pragma unroll
-                for (int rr = 0; rr < 4; ++rr) {
-                    const int k = out_k0 + kf * 16 + rr;
-                    const int j = out_j0 + jt * 64 + jf * 16;
-                    atomicAdd(&dw[(int64_t)k * rsc + j], acc[jt][kf][jf][rr]);
-                }
-}
-
-// ---------------------------------------------------------------------------
 // 8-wave 128(K) x 128(rsc) wgrad: double the tile edge of the kernel above,
 // HALVING both operands' re-read traffic (dout re-read rsc/128 times and x
 // re-read K/128 times instead of /64) — round-2 profiling put wgrad at 21%
@@ -579,25 +401,6 @@ extern "C" void launch_conv_wgrad(const void* x, const void* dout, void* dw,
         const int zn = (int)((M + mps - 1) / mps);
         dim3 grid((unsigned)(d.K / 128), (unsigned)(rsc / 128), (unsigned)zn);
         k_conv_wgrad8<<<grid, 512, 0, stream>>>(
-            (const uint16_t*)x, (const uint16_t*)dout, (float*)dw, d, mps);
-        return;
-    }
-    if (d.R * d.S > 1 && rsc % 192 == 0) {
-        // j-resident 64x192 tile (3 blocks/CU occupancy cap)
-        const int tiles = (d.K / 64) * (rsc / 192);
-        int ns = n_splits;
-        if (ns > 1) {
-            ns = 768 / tiles;
-            if (ns < 1) ns = 1;
-            if (ns > 128) ns = 128;
-            const int64_t mcap = M / 64;
-            if (ns > mcap) ns = (int)(mcap ? mcap : 1);
-        }
-        int mps = (int)((M + ns - 1) / ns);
-        mps = (mps + 2 * CONV_BK - 1) / (2 * CONV_BK) * (2 * CONV_BK);
-        const int zn = (int)((M + mps - 1) / mps);
-        dim3 grid((unsigned)(d.K / 64), (unsigned)(rsc / 192), (unsigned)zn);
-        k_conv_wgrad_j3<<<grid, CONV_THREADS, 0, stream>>>(
             (const uint16_t*)x, (const uint16_t*)dout, (float*)dw, d, mps);
         return;
     }
