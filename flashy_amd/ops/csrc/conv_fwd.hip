@@ -307,6 +307,12 @@ extern "C" void launch_conv_fwd8(const void* x, const void* w, void* y,
                                  ConvDims d, int relu, void* bn_ws, int bn,
                                  int mtiles, hipStream_t stream);
 extern "C" int conv1x1_mloop_plan(ConvDims d, int* bn_out, int* gridx_out);
+extern "C" int conv_dedup_plan(ConvDims d, int* bn_out, int* tpi_out,
+                               int* rows_out, int* elems_out);
+extern "C" void launch_conv_dedup(const void* x, const void* w, void* y,
+                                  ConvDims d, int relu, void* bn_ws, int bn,
+                                  int tpi, int rows, int lds_elems,
+                                  hipStream_t stream);
 extern "C" void launch_conv1x1_mloop(const void* x, const void* w, void* y,
                                      ConvDims d, int relu, void* bn_ws,
                                      int bn, int gridx, int mtiles,
@@ -316,6 +322,11 @@ extern "C" void launch_conv1x1_mloop(const void* x, const void* w, void* y,
 // fused BN-stats partials); Python sizes the partials buffer with this.
 extern "C" int conv_fwd_msplit(ConvDims d) {
     int bn8, gx;
+    {
+        int tpi, rows, elems;
+        const int gd = conv_dedup_plan(d, &bn8, &tpi, &rows, &elems);
+        if (gd) return gd;
+    }
     const int mtl = conv1x1_mloop_plan(d, &bn8, &gx);
     if (mtl) return 2 * mtl;   // per-(tile, wave_m) slices
     const int mt8 = conv_fwd8_plan(d, &bn8);
@@ -330,6 +341,15 @@ extern "C" void launch_conv_fwd(const void* x, const void* w, void* y,
                                 ConvDims d, int relu, void* bn_ws,
                                 hipStream_t stream) {
     int bn8, gx;
+    {
+        int tpi, rows, elems;
+        const int gd = conv_dedup_plan(d, &bn8, &tpi, &rows, &elems);
+        if (gd) {
+            launch_conv_dedup(x, w, y, d, relu, bn_ws, bn8, tpi, rows,
+                              elems, stream);
+            return;
+        }
+    }
     const int mtl = conv1x1_mloop_plan(d, &bn8, &gx);
     if (mtl) {
         launch_conv1x1_mloop(x, w, y, d, relu, bn_ws, bn8, gx, mtl, stream);
