@@ -155,12 +155,13 @@ k_conv_dedup(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
                 c -= d.C;
                 if (++s == d.S) { s = 0; ++r; }
             }
-            const int tap_off = (r * Wp + s) * d.C + c;
+            const int sp_off = r * Wp + s;   // slab-pixel offset of the tap
             short8 a[MF], b[NF];
 #pragma unroll
             for (int mf = 0; mf < MF; ++mf) {
-                const int elem = a_hobase[mf] * d.C + tap_off;
-                const int byte = (elem * 2) ^ (((elem / d.C) & 7) << 4);
+                const int slabpix = a_hobase[mf] + sp_off;
+                const int byte = ((slabpix * d.C + c) * 2) ^
+                                 ((slabpix & 7) << 4);
                 a[mf] = *reinterpret_cast<const short8*>(
                     (const char*)lds + byte);
             }
