@@ -15,7 +15,7 @@ Extras beyond the reference: ``xp`` (experiment runtime), ``graph``
 (HIP-graph capture), ``models``, ``ops`` (CDNA4 kernels).
 """
 
-__version__ = "0.1.0a1"
+__version__ = "0.2.0a1"
 
 import os as _os
 
