@@ -291,6 +291,12 @@ extern "C" void launch_conv_dgrad(const void* dout, const void* w_rsck,
     }
 }
 
+extern "C" void launch_conv_dgrad8_splitk(const void* dout,
+                                          const void* w_rsck, void* ws,
+                                          ConvDims d, int bn, int mtiles,
+                                          int spz, int zeff,
+                                          hipStream_t stream);
+
 extern "C" void launch_conv_dgrad_splitk(const void* dout, const void* w_rsck,
                                          void* ws, ConvDims d, int spz,
                                          hipStream_t stream) {
@@ -298,6 +304,17 @@ extern "C" void launch_conv_dgrad_splitk(const void* dout, const void* w_rsck,
     const int rsk = d.R * d.S * d.K;
     const int all_stages = (rsk + 63) / 64;
     const int zeff = (all_stages + spz - 1) / spz;
+    // 8-wave split-K for the small-M long-reduction layers (r4-class)
+    if (d.C % 64 == 0 && d.K % 64 == 0 && rsk % 64 == 0 &&
+        (int64_t)d.N * d.Ho * d.Wo * d.K * 2 < (int64_t)0xF0000000u) {
+        const int mtiles = (int)((M + 255) / 256);
+        const int bn8 = d.C % 128 == 0 ? 128 : 64;
+        if ((int64_t)mtiles * (d.C / bn8) * zeff >= 120) {
+            launch_conv_dgrad8_splitk(dout, w_rsck, ws, d, bn8, mtiles, spz,
+                                      zeff, stream);
+            return;
+        }
+    }
     dim3 grid((unsigned)((M + 63) / 64), (unsigned)(d.C / CONV_BN), (unsigned)zeff);
     if (d.stride == 1)
         k_conv_dgrad<64, true, true><<<grid, CONV_THREADS, 0, stream>>>(

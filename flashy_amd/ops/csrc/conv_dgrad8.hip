@@ -21,11 +21,12 @@
 // even (hi, wi), so A rows are the OUTPUT pixels read linearly from dy and
 // the epilogue scatters each result to (2ho, 2wo) while writing the three
 // odd-position siblings as zeros (no zero-filled MFMA work, no memset pass).
-template <int BN, bool S1, bool SCAT2 = false>
+template <int BN, bool S1, bool SCAT2 = false, bool SPLITK = false>
 __global__ void __launch_bounds__(512, 2)
 k_conv_dgrad8(const uint16_t* __restrict__ dout,
               const uint16_t* __restrict__ w_rsck,
-              uint16_t* __restrict__ dx, ConvDims d, unsigned dout_nbytes) {
+              uint16_t* __restrict__ dx, ConvDims d, unsigned dout_nbytes,
+              int spz = 0) {
     constexpr int BM = 256;
     constexpr int BK = 64;
     constexpr int NF = BN / 64;
@@ -79,7 +80,7 @@ k_conv_dgrad8(const uint16_t* __restrict__ dout,
         } else {
             a_n[g] = -1;
         }
-        const int kk = kc_s * 8;
+        const int kk = (SPLITK ? blockIdx.z * spz * 64 : 0) + kc_s * 8;
         a_r[g] = kk / (d.S * d.K);
         const int sk = kk - a_r[g] * d.S * d.K;
         a_s[g] = sk / d.K;
@@ -93,7 +94,7 @@ k_conv_dgrad8(const uint16_t* __restrict__ dout,
         const int col = chunk >> 3;
         const int kc_s = (chunk & 7) ^ (col & 7);
         b_col[g] = col0 + col;
-        const int kk = kc_s * 8;
+        const int kk = (SPLITK ? blockIdx.z * spz * 64 : 0) + kc_s * 8;
         b_r[g] = kk / (d.S * d.K);
         const int sk = kk - b_r[g] * d.S * d.K;
         b_s[g] = sk / d.K;
@@ -160,7 +161,12 @@ k_conv_dgrad8(const uint16_t* __restrict__ dout,
     const int b_col_l = wave_n * (BN / 4) + (lane & 15);
 
     floatx4 acc[MF][NF] = {};
-    const int n_stages = rsk / BK;
+    int n_stages = rsk / BK;
+    if (SPLITK) {
+        const int remain = n_stages - blockIdx.z * spz;
+        n_stages = remain < spz ? remain : spz;
+        if (n_stages <= 0) return;
+    }
 
     auto compute_stage = [&](int buf) {
         const uint16_t* base = lds + buf * BUF_ELEMS;
@@ -215,6 +221,23 @@ k_conv_dgrad8(const uint16_t* __restrict__ dout,
 
     const int64_t out_row0 = m0 + wave_m * 128 + (lane & 15);
     const int out_col0 = col0 + wave_n * (BN / 4) + (lane >> 4) * 4;
+    if (SPLITK) {   // fp32 partials slab [z][M][C]
+        float* ws = reinterpret_cast<float*>(dx);
+#pragma unroll
+        for (int mf = 0; mf < MF; ++mf) {
+            const int64_t row = out_row0 + mf * 16;
+            if (row < M) {
+#pragma unroll
+                for (int nf = 0; nf < NF; ++nf)
+                    *reinterpret_cast<float4*>(
+                        ws + ((int64_t)blockIdx.z * M + row) * d.C +
+                        out_col0 + nf * 16) =
+                        make_float4(acc[mf][nf][0], acc[mf][nf][1],
+                                    acc[mf][nf][2], acc[mf][nf][3]);
+            }
+        }
+        return;
+    }
 #pragma unroll
     for (int mf = 0; mf < MF; ++mf) {
         const int64_t row = out_row0 + mf * 16;
@@ -324,5 +347,31 @@ extern "C" void launch_conv_dgrad8(const void* dout, const void* w_rsck,
             k_conv_dgrad8<64, true><<<grid, 512, 0, stream>>>(dd, ww, xx, d, db);
         else
             k_conv_dgrad8<64, false><<<grid, 512, 0, stream>>>(dd, ww, xx, d, db);
+    }
+}
+
+extern "C" void launch_conv_dgrad8_splitk(const void* dout,
+                                          const void* w_rsck, void* ws,
+                                          ConvDims d, int bn, int mtiles,
+                                          int spz, int zeff,
+                                          hipStream_t stream) {
+    dim3 grid((unsigned)mtiles, (unsigned)(d.C / bn), (unsigned)zeff);
+    const unsigned db = (unsigned)((int64_t)d.N * d.Ho * d.Wo * d.K * 2);
+    auto dd = (const uint16_t*)dout;
+    auto ww = (const uint16_t*)w_rsck;
+    if (bn == 128) {
+        if (d.stride == 1)
+            k_conv_dgrad8<128, true, false, true><<<grid, 512, 0, stream>>>(
+                dd, ww, (uint16_t*)ws, d, db, spz);
+        else
+            k_conv_dgrad8<128, false, false, true><<<grid, 512, 0, stream>>>(
+                dd, ww, (uint16_t*)ws, d, db, spz);
+    } else {
+        if (d.stride == 1)
+            k_conv_dgrad8<64, true, false, true><<<grid, 512, 0, stream>>>(
+                dd, ww, (uint16_t*)ws, d, db, spz);
+        else
+            k_conv_dgrad8<64, false, false, true><<<grid, 512, 0, stream>>>(
+                dd, ww, (uint16_t*)ws, d, db, spz);
     }
 }

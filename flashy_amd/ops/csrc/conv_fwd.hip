@@ -402,12 +402,29 @@ extern "C" void launch_conv_fwd(const void* x, const void* w, void* y,
 
 // split-K path: BM=64 tiles, grid.z over stage ranges, fp32 workspace
 // [zn][M][K], then k_splitk_combine.
+extern "C" void launch_conv_fwd8_splitk(const void* x, const void* w,
+                                        void* ws, ConvDims d, int bn,
+                                        int mtiles, int spz, int zeff,
+                                        hipStream_t stream);
+
 extern "C" void launch_conv_fwd_splitk(const void* x, const void* w, void* ws,
                                        ConvDims d, int spz, hipStream_t stream) {
     const int64_t M = (int64_t)d.N * d.Ho * d.Wo;
     const int rsc = d.R * d.S * d.C;
     const int all_stages = (rsc + 63) / 64;
     const int zeff = (all_stages + spz - 1) / spz;
+    // 8-wave split-K for the small-M long-reduction layers (r4-class):
+    // same Python-decided spz/zeff and ws layout, deeper pipeline
+    if (d.C % 64 == 0 && d.K % 64 == 0 && rsc % 64 == 0 &&
+        (int64_t)d.N * d.H * d.W * d.C * 2 < (int64_t)0xF0000000u) {
+        const int mtiles = (int)((M + 255) / 256);
+        const int bn8 = d.K % 128 == 0 ? 128 : 64;
+        if ((int64_t)mtiles * (d.K / bn8) * zeff >= 120) {
+            launch_conv_fwd8_splitk(x, w, ws, d, bn8, mtiles, spz, zeff,
+                                    stream);
+            return;
+        }
+    }
     dim3 grid((unsigned)((M + 63) / 64), (unsigned)(d.K / CONV_BN), (unsigned)zeff);
     k_conv_fwd<64, false, true><<<grid, CONV_THREADS, 0, stream>>>(
         (const uint16_t*)x, (const uint16_t*)w, nullptr, (float*)ws, nullptr,

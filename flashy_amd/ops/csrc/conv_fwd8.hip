@@ -29,11 +29,11 @@
 // here are < 2^31 - 2^28 bytes); the buffer bounds check returns zeros.
 #define OOB_SENTINEL 0xF0000000u
 
-template <int BN, bool RELU>
+template <int BN, bool RELU, bool SPLITK = false>
 __global__ void __launch_bounds__(512, 2)
 k_conv_fwd8(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
             uint16_t* __restrict__ y, float* __restrict__ bn_ws,
-            ConvDims d, unsigned x_nbytes) {
+            ConvDims d, unsigned x_nbytes, int spz = 0) {
     constexpr int BM = 256;
     constexpr int BK = 64;
     constexpr int NF = BN / 64;          // n fragments per wave (4 N-waves)
@@ -85,7 +85,7 @@ k_conv_fwd8(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
         } else {
             a_n[g] = -1;
         }
-        const int kk = kc_s * 8;                  // tap offset at stage 0
+        const int kk = (SPLITK ? blockIdx.z * spz * 64 : 0) + kc_s * 8;
         a_r[g] = kk / (d.S * d.C);
         const int sc = kk - a_r[g] * d.S * d.C;
         a_s[g] = sc / d.C;
@@ -98,7 +98,9 @@ k_conv_fwd8(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
         const int chunk = g * 512 + tid;          // dest chunk 0..BN*8-1
         const int col = chunk >> 3;
         const int kc_s = (chunk & 7) ^ (col & 7);
-        b_src[g] = (unsigned)(((int64_t)(col0 + col) * rsc + kc_s * 8) * 2);
+        b_src[g] = (unsigned)(((int64_t)(col0 + col) * rsc +
+                              (SPLITK ? blockIdx.z * spz * 64 : 0) +
+                              kc_s * 8) * 2);
     }
 
     auto issue_stage = [&](int stage, int buf) {
@@ -145,7 +147,12 @@ k_conv_fwd8(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
     const int b_col_l = wave_n * (BN / 4) + (lane & 15);  // + nf*16
 
     floatx4 acc[MF][NF] = {};
-    const int n_stages = rsc / BK;
+    int n_stages = rsc / BK;
+    if (SPLITK) {
+        const int remain = n_stages - blockIdx.z * spz;
+        n_stages = remain < spz ? remain : spz;
+        if (n_stages <= 0) return;
+    }
 
     auto compute_stage = [&](int buf) {
         const uint16_t* base = lds + buf * BUF_ELEMS;
@@ -213,6 +220,23 @@ k_conv_fwd8(const uint16_t* __restrict__ x, const uint16_t* __restrict__ w,
     // CONSECUTIVE channels per lane, packed into one 8 B store.
     const int64_t out_row0 = m0 + wave_m * 128 + (lane & 15);
     const int out_col0 = col0 + wave_n * (BN / 4) + (lane >> 4) * 4;
+    if (SPLITK) {   // fp32 partials slab [z][M][K], float4 per lane
+        float* ws = reinterpret_cast<float*>(bn_ws);   // reused arg
+#pragma unroll
+        for (int mf = 0; mf < MF; ++mf) {
+            const int64_t row = out_row0 + mf * 16;
+            if (row < M) {
+#pragma unroll
+                for (int nf = 0; nf < NF; ++nf)
+                    *reinterpret_cast<float4*>(
+                        ws + ((int64_t)blockIdx.z * M + row) * d.K +
+                        out_col0 + nf * 16) =
+                        make_float4(acc[mf][nf][0], acc[mf][nf][1],
+                                    acc[mf][nf][2], acc[mf][nf][3]);
+            }
+        }
+        return;
+    }
 #pragma unroll
     for (int mf = 0; mf < MF; ++mf) {
         const int64_t row = out_row0 + mf * 16;
@@ -517,6 +541,22 @@ extern "C" void launch_conv1x1_mloop(const void* x, const void* w, void* y,
             k_conv1x1_mloop<64, false><<<grid, 512, 0, stream>>>(
                 xx, ww, yy, (float*)bn_ws, d, xb, mtiles);
     }
+}
+
+extern "C" void launch_conv_fwd8_splitk(const void* x, const void* w,
+                                        void* ws, ConvDims d, int bn,
+                                        int mtiles, int spz, int zeff,
+                                        hipStream_t stream) {
+    dim3 grid((unsigned)mtiles, (unsigned)(d.K / bn), (unsigned)zeff);
+    const unsigned xb = (unsigned)((int64_t)d.N * d.H * d.W * d.C * 2);
+    auto xx = (const uint16_t*)x;
+    auto wv = (const uint16_t*)w;
+    if (bn == 128)
+        k_conv_fwd8<128, false, true><<<grid, 512, 0, stream>>>(
+            xx, wv, nullptr, (float*)ws, d, xb, spz);
+    else
+        k_conv_fwd8<64, false, true><<<grid, 512, 0, stream>>>(
+            xx, wv, nullptr, (float*)ws, d, xb, spz);
 }
 
 // --- stem padding helpers ---------------------------------------------------

@@ -26,6 +26,7 @@ SHAPES = [
     (64, 29, 28, 128, 128, 3, 1, 1),  # same with an M tail (odd rows)
     (16, 56, 56, 256, 512, 1, 2, 0),  # 1x1 stride-2: dgrad8 SCAT2 scatter
     (64, 112, 112, 3, 64, 7, 2, 3),   # 224-class stem: padded fwd8 + wgrad
+    (64, 7, 7, 512, 512, 3, 1, 1),    # r4_3x3: 8-wave split-K fwd/dgrad
 ]
 
 
