@@ -181,6 +181,16 @@ def conv_fwd(x: torch.Tensor, w: torch.Tensor, y: torch.Tensor, d: ConvDims,
         stages = (rsc + 63) // 64
         zn = 0 if ext.conv8_eligible(*d, False) else \
             _splitk_plan(M, d.K // 64, stages)
+        if zn == 0 and stages >= 16 and not ext.conv8_eligible(*d, False):
+            # small-M long-reduction (r4-class 1x1): the 8-wave grid
+            # underfills without a K split — slice it so the 8-wave
+            # split-K launcher engages
+            mt8 = (M + 255) // 256
+            bn8 = 128 if d.K % 128 == 0 else 64
+            tiles8 = mt8 * (d.K // bn8)
+            if tiles8 and tiles8 < 160:
+                zn = min(max(2, -(-160 // tiles8)), 8, stages // 8)
+                zn = 0 if zn < 2 else zn
         if zn:
             spz = (stages + zn - 1) // zn
             zeff = (stages + spz - 1) // spz
@@ -263,6 +273,14 @@ def conv_dgrad(dout: torch.Tensor, w_rsck: torch.Tensor, dx: torch.Tensor,
     stages = (rsk + 63) // 64
     zn = 0 if ext.conv8_eligible(*d, True) else \
         _splitk_plan(M, d.C // 64, stages)
+    if zn == 0 and stages >= 16 and d.stride == 1 \
+            and not ext.conv8_eligible(*d, True):
+        mt8 = (M + 255) // 256
+        bn8 = 128 if d.C % 128 == 0 else 64
+        tiles8 = mt8 * (d.C // bn8)
+        if tiles8 and tiles8 < 160:
+            zn = min(max(2, -(-160 // tiles8)), 8, stages // 8)
+            zn = 0 if zn < 2 else zn
     if zn:
         spz = (stages + zn - 1) // zn
         zeff = (stages + spz - 1) // spz
