@@ -306,16 +306,31 @@ class ConvTranspose2d(nn.Module):
                                       self.input_grad and x.requires_grad)
 
 
+# Above this many FLOPs per product the fc goes to rocBLAS (the guide's
+# rule for PLAIN library GEMMs — no fusion opportunity here); below it the
+# one-kernel-per-product native path wins on launch overhead.  The ResNet-50
+# head (64x2048 -> 1000, 262 MFLOP) sits far above; the example MLPs far
+# below.  Measured: the naive fc kernels cost 435 us/step on the R50 head
+# vs ~15 us through rocBLAS (profiles/r02h).
+_LINEAR_GEMM_CUTOFF = 1 << 23
+
+
 class _LinearFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x: torch.Tensor, w: torch.Tensor,
                 b: tp.Optional[torch.Tensor]):
         B, I = x.shape
         O = w.shape[0]
-        y = x.new_empty((B, O))
-        ops.linear_fwd(x.contiguous(), w, b, y)
+        big = 2 * B * I * O >= _LINEAR_GEMM_CUTOFF and x.dtype == torch.float32
+        if big:
+            x = x.contiguous()
+            y = torch.addmm(b, x, w.t()) if b is not None else x.mm(w.t())
+        else:
+            y = x.new_empty((B, O))
+            ops.linear_fwd(x.contiguous(), w, b, y)
         ctx.save_for_backward(x)
         ctx.refs = (w, b)
+        ctx.big = big
         return y
 
     @staticmethod
@@ -326,7 +341,16 @@ class _LinearFn(torch.autograd.Function):
         dw = db = None
         need_w = ctx.needs_input_grad[1]
         need_b = b is not None and ctx.needs_input_grad[2]
-        if need_w or need_b:
+        if ctx.big:
+            if need_w:
+                dw, w_direct = _grad_target(w)
+                dw.addmm_(dy.t(), x)  # += dy^T @ x (rocBLAS, fp32)
+                dw = None if w_direct else dw
+            if need_b:
+                db, b_direct = _grad_target(b)
+                db.add_(dy.sum(0))
+                db = None if b_direct else db
+        elif need_w or need_b:
             # one kernel produces both; frozen side goes to a discarded temp
             dw, w_direct = _grad_target(w) if need_w \
                 else (torch.empty_like(w), True)
@@ -336,15 +360,21 @@ class _LinearFn(torch.autograd.Function):
             db = db if need_b and not b_direct else None
         dx = None
         if ctx.needs_input_grad[0]:
-            dx = x.new_empty(x.shape)
-            ops.linear_dx(dy, w, dx)
+            if ctx.big:
+                dx = dy.mm(w)
+            else:
+                dx = x.new_empty(x.shape)
+                ops.linear_dx(dy, w, dx)
         return dx, dw, db
 
 
 class Linear(nn.Module):
-    """fp32 fully-connected layer on the native fc kernels (the ResNet head
-    and the tiny example MLPs — launch-bound shapes, one kernel per product).
-    Weight [O, I], bias [O]; CPU falls back to torch.nn.functional.linear."""
+    """fp32 fully-connected layer.  Tiny (launch-bound) shapes run the
+    native fc kernels, one kernel per product; large heads (e.g. ResNet-50's
+    2048->1000) go through rocBLAS — a plain dense GEMM with no fusion
+    opportunity belongs on the library path (guide rule; 435 us -> ~15 us
+    per step measured).  Weight [O, I], bias [O]; CPU falls back to
+    torch.nn.functional.linear."""
 
     def __init__(self, in_features: int, out_features: int, bias: bool = True):
         super().__init__()

@@ -135,7 +135,9 @@ def test_accuracy_gpu(dtype, B, C):
 
 @requires_gpu
 @pytest.mark.parametrize("B,I,O,bias", [
-    (64, 512, 10, True), (7, 32, 1, True), (16, 33, 5, False)])
+    (64, 512, 10, True), (7, 32, 1, True), (16, 33, 5, False),
+    # crosses _LINEAR_GEMM_CUTOFF -> the rocBLAS branch (R50-head shape)
+    (64, 2048, 1000, True), (64, 2048, 1000, False)])
 def test_linear_gpu(B, I, O, bias):
     from flashy_amd.nn import Linear
     torch.manual_seed(6)
