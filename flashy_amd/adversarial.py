@@ -26,6 +26,8 @@ from . import distrib
 from .utils import readonly
 
 LossFn = tp.Callable[[torch.Tensor, torch.Tensor], torch.Tensor]
+# drop-in alias kept from the reference API (flashy/adversarial.py)
+LossType = LossFn
 
 
 def _bce_logits(logits: torch.Tensor, target_is_fake: float) -> torch.Tensor:

@@ -39,6 +39,10 @@ from .formatter import Formatter
 from .logging import LogProgressBar, ResultLogger
 from .state import AttributeWrapper, StateManager, WriteOnlyWrapper
 
+# drop-in alias kept from the reference API (flashy/solver.py):
+# a stage method as passed to run_stage.
+StageCallable = tp.Callable[..., tp.Optional[tp.Dict[str, tp.Any]]]
+
 logger = logging.getLogger(__name__)
 
 

@@ -17,6 +17,9 @@ from __future__ import annotations
 
 import typing as tp
 
+# drop-in alias kept from the reference API (flashy/state.py)
+StateDict = tp.Dict[str, tp.Any]
+
 
 @tp.runtime_checkable
 class StateDictSource(tp.Protocol):

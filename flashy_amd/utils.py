@@ -14,6 +14,9 @@ import typing as tp
 from contextlib import contextmanager
 from pathlib import Path
 
+# drop-in alias kept from the reference API (flashy/utils.py)
+AnyPath = tp.Union[str, Path]
+
 import torch
 
 
