@@ -278,3 +278,64 @@ def test_conv_fused_bn_stats():
     tol2 = 0.02 * M ** 0.5 * (yf * yf).max().item()
     assert (s2 - (yf * yf).sum(0)).abs().max().item() < tol2, \
         ((s2 - (yf * yf).sum(0)).abs().max(), tol2)
+
+
+def _fuzz_shapes(n=24, seed=1234):
+    """Deterministic random sample of the supported conv envelope
+    (C,K multiples of 64, R in {1,3,5}, stride in {1,2}) — regression net
+    for the dispatch chain (dedup / mloop / fwd8 / split-K under-fill
+    forcing / SCAT2 / s2-parity / old-tile fallbacks)."""
+    import random
+    rng = random.Random(seed)
+    shapes = []
+    while len(shapes) < n:
+        C = 64 * rng.choice([1, 1, 2, 3, 4])
+        K = 64 * rng.choice([1, 1, 2, 3, 4])
+        R = rng.choice([1, 1, 3, 3, 5])
+        stride = rng.choice([1, 1, 2])
+        pad = R // 2 if R > 1 else 0
+        N = rng.choice([1, 2, 3, 8])
+        H = rng.randint(4, 36)
+        W = rng.randint(4, 36)
+        Ho = (H + 2 * pad - R) // stride + 1
+        Wo = (W + 2 * pad - R) // stride + 1
+        if Ho < 1 or Wo < 1:
+            continue
+        shapes.append((N, H, W, C, K, R, stride, pad))
+    # two large-M entries to engage the 8-wave family + under-fill split-K
+    shapes.append((64, 14, 14, 256, 256, 3, 1, 1))
+    shapes.append((64, 14, 14, 1024, 256, 1, 1, 0))
+    return shapes
+
+
+@requires_gpu
+@pytest.mark.parametrize("shape", _fuzz_shapes())
+def test_conv_fuzz(shape):
+    from flashy_amd import ops
+    N, H, W, C, K, R, stride, pad = shape
+    x, w = _mk(N, H, W, C, K, R, seed=hash(shape) & 0xffff)
+    d = ops.ConvDims.infer(x, w, stride, pad)
+    g = torch.Generator(device="cuda").manual_seed(5)
+    dy = torch.randn(N, d.Ho, d.Wo, K, device="cuda", generator=g).to(torch.bfloat16)
+
+    y = x.new_empty((d.N, d.Ho, d.Wo, d.K))
+    ops.conv_fwd(x, w, y, d)
+    dx = x.new_empty(x.shape)
+    wt = w.new_empty((d.R, d.S, d.C, d.K))
+    ops.weight_transpose(w, wt)
+    ops.conv_dgrad(dy, wt, dx, d)
+    dw = torch.zeros(K, R, R, C, device="cuda", dtype=torch.float32)
+    ops.conv_wgrad(x, dy, dw, d)
+    torch.cuda.synchronize()
+
+    xr = x.float().permute(0, 3, 1, 2).requires_grad_(True)
+    wr = w.float().permute(0, 3, 1, 2).requires_grad_(True)
+    ref = F.conv2d(xr, wr, stride=stride, padding=pad)
+    ref.backward(dy.float().permute(0, 3, 1, 2))
+    for got, want, tag in [
+            (y.float(), ref.detach().permute(0, 2, 3, 1), "fwd"),
+            (dx.float(), xr.grad.permute(0, 2, 3, 1), "dgrad"),
+            (dw, wr.grad.permute(0, 2, 3, 1), "wgrad")]:
+        err = (got - want).abs().max().item()
+        scale = want.abs().max().item() + 1e-6
+        assert err / scale < 2e-2, (tag, err, scale, shape)
