@@ -15,6 +15,8 @@ SHAPES = {
     "r2_3x3": (28, 28, 128, 128, 3, 1),
     "r1_3x3": (56, 56, 64, 64, 3, 1),
     "r1_1x1b": (56, 56, 64, 256, 1, 1),
+    "r4_3x3": (7, 7, 512, 512, 3, 1),      # 8-wave split-K path
+    "r4_1x1a": (7, 7, 2048, 512, 1, 1),    # under-fill forced split-K
 }
 H, W, C, K, R, stride = SHAPES[name]
 N, pad = 64, R // 2
