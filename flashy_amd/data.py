@@ -47,6 +47,7 @@ class DevicePrefetcher:
         self._slots: tp.List[tp.Optional[tp.Tuple[torch.Tensor, ...]]] = \
             [None] * depth
         self._ready = [torch.cuda.Event() for _ in range(depth)]
+        self._bare = [False] * depth
         self._head = 0          # next slot to hand out
         self._primed = 0
         for _ in range(depth):
@@ -60,9 +61,11 @@ class DevicePrefetcher:
             host = next(self._it)
         except StopIteration:
             return False
-        if isinstance(host, torch.Tensor):
+        bare = isinstance(host, torch.Tensor)
+        if bare:
             host = (host,)
         slot = self._primed % self.depth
+        self._bare[slot] = bare
         with torch.cuda.stream(self._stream):
             staged = self._slots[slot]
             if staged is None:
@@ -86,7 +89,7 @@ class DevicePrefetcher:
             batch = self._slots[slot]
             assert batch is not None
             self._head += 1
-            yield batch if len(batch) > 1 else batch[0]
+            yield batch[0] if self._bare[slot] else batch
             # the caller has now ENQUEUED its use of `batch` on the current
             # stream; fence the copy stream behind it before the slot is
             # overwritten by the refill

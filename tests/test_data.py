@@ -50,7 +50,7 @@ def test_gpu_prefetch_slot_reuse_under_load():
     batches = [(torch.full((1 << 20,), float(i), pin_memory=True),) for i in range(n)]
     sums = []
     spin = torch.randn(2048, 2048, device="cuda")
-    for (bx,) in DevicePrefetcher(iter(batches), "cuda", depth=depth):
+    for (bx,) in DevicePrefetcher(iter(batches), "cuda", depth=depth):  # 1-tuple in -> 1-tuple out
         for _ in range(4):
             spin = spin @ spin.T / 2048  # queue depth on the main stream
         sums.append(bx.sum())

@@ -52,6 +52,11 @@ def test_local_ddp_two_workers(tmp_path):
 def test_bench_distributed_contract(tmp_path):
     """bench.py under torch.distributed.run exactly as the driver launches
     it (one JSON line from rank 0, value aggregated over the world)."""
+    import torch
+    if torch.cuda.is_available() and torch.cuda.device_count() < 2:
+        pytest.skip("2 ranks need 2 GPUs once cuda is visible (RCCL "
+                    "refuses two ranks on one device); the CPU/gloo "
+                    "container covers the contract")
     env = dict(os.environ)
     env.pop("RANK", None)
     env.pop("WORLD_SIZE", None)
