@@ -851,6 +851,11 @@ extern "C" void launch_conv_stem_wgrad(const void* x, const void* dout, void* dw
 // discriminator).  Computes the equivalent conv's dX for C <= 8, K == 64:
 //   dx[n,hi,wi,c] = sum over valid (r,s):  dout[n,ho,wo,k] * w[k,r,s,c]
 // with ho = (hi+pad-r)/stride when exact.  Weights staged to LDS as fp32.
+// CT = exact channel count when known (3 = RGB, 4), else 8 with runtime
+// masking.  The k loop runs as 8 short8 (16 B) vector loads per tap — the
+// original 64 scalar 2 B loads made the DCGAN edge-conv backward
+// issue-bound at ~198 us (13.8% of the GAN step, profiles/r02j).
+template <int CT>
 __global__ void __launch_bounds__(256)
 k_conv_stem_dgrad(const uint16_t* __restrict__ dout,
                   const uint16_t* __restrict__ w,
@@ -868,7 +873,7 @@ k_conv_stem_dgrad(const uint16_t* __restrict__ dout,
         const int wi = (int)(m % d.W);
         const int hi = (int)((m / d.W) % d.H);
         const int64_t n = m / ((int64_t)d.H * d.W);
-        float acc[8] = {};
+        float acc[CT] = {};
         for (int r = 0; r < d.R; ++r) {
             const int hnum = hi + d.pad - r;
             if (hnum < 0 || hnum % d.stride) continue;
@@ -882,20 +887,29 @@ k_conv_stem_dgrad(const uint16_t* __restrict__ dout,
                 const uint16_t* gp =
                     dout + ((n * d.Ho + ho) * d.Wo + wo) * (int64_t)d.K;
                 const int base = (r * d.S + s) * d.C;
-                for (int k = 0; k < 64; ++k) {
-                    const float g = bf16_to_f32(gp[k]);
-                    // compile-time trip count keeps acc[] in registers
-                    // (guide rule 20); C <= 8 by the dispatch condition.
 #pragma unroll
-                    for (int c = 0; c < 8; ++c)
-                        if (c < d.C)
-                            acc[c] = fmaf(g, w_lds[k * rsc + base + c], acc[c]);
+                for (int kv = 0; kv < 8; ++kv) {
+                    const short8 g8 =
+                        *reinterpret_cast<const short8*>(gp + kv * 8);
+#pragma unroll
+                    for (int j = 0; j < 8; ++j) {
+                        const float g =
+                            bf16_to_f32(((const uint16_t*)&g8)[j]);
+                        const int k = kv * 8 + j;
+                        // compile-time trip count keeps acc[] in
+                        // registers (guide rule 20)
+#pragma unroll
+                        for (int c = 0; c < CT; ++c)
+                            if (CT <= 4 || c < d.C)
+                                acc[c] = fmaf(g, w_lds[k * rsc + base + c],
+                                              acc[c]);
+                    }
                 }
             }
         }
 #pragma unroll
-        for (int c = 0; c < 8; ++c)
-            if (c < d.C)
+        for (int c = 0; c < CT; ++c)
+            if (CT <= 4 || c < d.C)
                 dx[m * d.C + c] = f32_to_bf16(acc[c]);
     }
 }
@@ -904,6 +918,14 @@ extern "C" void launch_conv_stem_dgrad(const void* dout, const void* w,
                                        void* dx, ConvDims d,
                                        hipStream_t stream) {
     const int64_t total = (int64_t)d.N * d.H * d.W;
-    k_conv_stem_dgrad<<<ew_grid(total, 256, 1), 256, 0, stream>>>(
-        (const uint16_t*)dout, (const uint16_t*)w, (uint16_t*)dx, d);
+    const int grid = ew_grid(total, 256, 1);
+    if (d.C == 3)
+        k_conv_stem_dgrad<3><<<grid, 256, 0, stream>>>(
+            (const uint16_t*)dout, (const uint16_t*)w, (uint16_t*)dx, d);
+    else if (d.C == 4)
+        k_conv_stem_dgrad<4><<<grid, 256, 0, stream>>>(
+            (const uint16_t*)dout, (const uint16_t*)w, (uint16_t*)dx, d);
+    else
+        k_conv_stem_dgrad<8><<<grid, 256, 0, stream>>>(
+            (const uint16_t*)dout, (const uint16_t*)w, (uint16_t*)dx, d);
 }
