@@ -58,7 +58,7 @@ class NativeDCGANDiscriminator(nn.Module):
         self.bn3 = fnn.BatchNorm2d(ndf * 4)
         self.conv4 = fnn.Conv2d(ndf * 4, ndf * 8, 4, 2, 1, feeds_bn=True)
         self.bn4 = fnn.BatchNorm2d(ndf * 8)
-        self.head = nn.Linear(ndf * 8 * 4 * 4, 1)  # = the final 4x4 conv
+        self.head = fnn.Linear(ndf * 8 * 4 * 4, 1)  # = the final 4x4 conv
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         if x.shape[1] == 3:  # NCHW -> logical NHWC

@@ -20,6 +20,7 @@ Replaces the ResNet conv+bn+relu torch chains of the reference workload
 from __future__ import annotations
 
 import math
+import os
 import typing as tp
 
 import torch
@@ -312,7 +313,7 @@ class ConvTranspose2d(nn.Module):
 # head (64x2048 -> 1000, 262 MFLOP) sits far above; the example MLPs far
 # below.  Measured: the naive fc kernels cost 435 us/step on the R50 head
 # vs ~15 us through rocBLAS (profiles/r02h).
-_LINEAR_GEMM_CUTOFF = 1 << 23
+_LINEAR_GEMM_CUTOFF = int(os.environ.get("FLASHY_LINEAR_CUTOFF", 1 << 23))
 
 
 class _LinearFn(torch.autograd.Function):
