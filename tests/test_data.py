@@ -57,3 +57,7 @@ def test_gpu_prefetch_slot_reuse_under_load():
     torch.cuda.synchronize()
     for i, s in enumerate(sums):
         assert s.item() == pytest.approx(float(i) * (1 << 20), rel=1e-6), i
+
+
+def test_empty_iterator_cpu():
+    assert list(DevicePrefetcher(iter([]), "cpu")) == []
